@@ -1,0 +1,342 @@
+"""Field codecs: encode/decode between in-memory numpy values and the
+Parquet-storable representation of a field.
+
+Behavioral parity with the reference /root/reference/petastorm/codecs.py:
+
+* ``DataframeColumnCodec`` protocol (codecs.py:36-55)
+* ``CompressedImageCodec`` png/jpeg with quality (codecs.py:58-130) —
+  implemented over Pillow instead of OpenCV (this environment has PIL, not
+  cv2; the observable contract — RGB/grayscale uint8/uint16 ndarray in,
+  compressed bytes out, lossless for png — is preserved).  JPEG encode emits
+  a restart marker at every MCU row (``restart_marker_rows=1``) so the
+  MI355X decoder can Huffman-decode restart segments in parallel; this only
+  adds ~2 bytes/MCU-row and is valid baseline JPEG.
+* ``NdarrayCodec`` .npy bytes (codecs.py:133-171)
+* ``CompressedNdarrayCodec`` zlib-compressed .npz (codecs.py:174-212)
+* ``ScalarCodec`` (codecs.py:215-271) — here it validates/casts scalars; the
+  storage type is derived from the field's numpy dtype since this framework
+  writes through pyarrow, not Spark.
+* shape compliance checks (codecs.py:274-294)
+
+Codecs are JSON-serializable (``codec_to_dict``/``codec_from_dict``) because
+the dataset schema itself is stored as JSON, not pickle.
+"""
+
+import io
+import zlib
+from decimal import Decimal
+
+import numpy as np
+
+from petastorm_amd.errors import DecodeFieldError
+
+
+class DataframeColumnCodec(object):
+    """Codec protocol (reference petastorm/codecs.py:36-55)."""
+
+    def encode(self, unischema_field, value):
+        raise NotImplementedError()
+
+    def decode(self, unischema_field, value):
+        raise NotImplementedError()
+
+    def arrow_type(self, unischema_field):
+        """pyarrow DataType of the *stored* column."""
+        raise NotImplementedError()
+
+    def to_dict(self):
+        return {'type': type(self).__name__}
+
+    def __eq__(self, other):
+        return isinstance(other, type(self)) and self.__dict__ == other.__dict__
+
+    def __hash__(self):
+        return hash(type(self).__name__)
+
+
+def _check_dtype(unischema_field, value):
+    expected = np.dtype(unischema_field.numpy_dtype)
+    if expected.kind in 'SU':  # flexible-width string/bytes dtypes
+        if value.dtype.kind != expected.kind:
+            raise ValueError('Field {!r}: expected dtype kind {!r}, got {}'
+                             .format(unischema_field.name, expected.kind,
+                                     value.dtype))
+        return
+    if value.dtype != expected:
+        raise ValueError('Field {!r}: expected dtype {}, got {}'
+                         .format(unischema_field.name, expected, value.dtype))
+
+
+def _check_shape_compliance(unischema_field, value):
+    """Verify ndarray rank/dims match the field's declared shape.
+
+    ``None`` dims match anything (reference petastorm/codecs.py:274-294).
+    """
+    expected = unischema_field.shape
+    if expected is None:
+        return
+    if len(value.shape) != len(expected):
+        raise ValueError('Field {!r}: expected rank {} (shape {}), got shape {}'
+                         .format(unischema_field.name, len(expected), expected,
+                                 value.shape))
+    for e, a in zip(expected, value.shape):
+        if e is not None and e != a:
+            raise ValueError('Field {!r}: value shape {} does not comply with '
+                             'declared shape {}'.format(unischema_field.name,
+                                                        value.shape, expected))
+
+
+_NUMPY_TO_ARROW_SCALAR = {
+    'int8': 'int8', 'uint8': 'uint8', 'int16': 'int16', 'uint16': 'uint16',
+    'int32': 'int32', 'uint32': 'uint32', 'int64': 'int64', 'uint64': 'uint64',
+    'float16': 'float16', 'float32': 'float32', 'float64': 'float64',
+    'bool': 'bool_',
+}
+
+
+def _arrow_scalar_type(np_dtype):
+    import pyarrow as pa
+    if np_dtype is Decimal:
+        # stored as string for exactness, like the reference stores decimals
+        # via spark DecimalType but reads back Decimal objects
+        # (petastorm/codecs.py:241-247 round-trips through str)
+        return pa.string()
+    if np_dtype in (np.str_,):
+        return pa.string()
+    if np_dtype in (np.bytes_,):
+        return pa.binary()
+    if np_dtype is np.datetime64:
+        return pa.timestamp('ns')
+    name = np.dtype(np_dtype).name
+    if name in _NUMPY_TO_ARROW_SCALAR:
+        return getattr(pa, _NUMPY_TO_ARROW_SCALAR[name])()
+    raise ValueError('No arrow mapping for numpy dtype {}'.format(np_dtype))
+
+
+class ScalarCodec(DataframeColumnCodec):
+    """Codec for scalar fields (reference petastorm/codecs.py:215-271).
+
+    The reference parameterizes ScalarCodec by a Spark type; this framework
+    derives the storage type from the field's numpy dtype, so the constructor
+    argument is optional and kept only for API familiarity.
+    """
+
+    def __init__(self, spark_type=None):
+        self._spark_type = None  # unused; retained for API-shape familiarity
+
+    def encode(self, unischema_field, value):
+        dt = unischema_field.numpy_dtype
+        if dt is Decimal:
+            return str(value if isinstance(value, Decimal) else Decimal(value))
+        if dt in (np.str_,):
+            return str(value)
+        if dt in (np.bytes_,):
+            return bytes(value)
+        if dt is np.datetime64:
+            return np.datetime64(value)
+        if isinstance(value, np.ndarray):
+            if value.shape != ():
+                raise ValueError('Field {!r} is scalar but got array of shape {}'
+                                 .format(unischema_field.name, value.shape))
+            value = value[()]
+        return np.dtype(dt).type(value).item()
+
+    def decode(self, unischema_field, value):
+        dt = unischema_field.numpy_dtype
+        if dt is Decimal:
+            return value if isinstance(value, Decimal) else Decimal(value)
+        if dt in (np.str_,):
+            return value if isinstance(value, str) else str(value)
+        if dt in (np.bytes_,):
+            return value
+        if dt is np.datetime64:
+            return np.datetime64(value)
+        return np.dtype(dt).type(value)
+
+    def arrow_type(self, unischema_field):
+        return _arrow_scalar_type(unischema_field.numpy_dtype)
+
+    def to_dict(self):
+        return {'type': 'ScalarCodec'}
+
+
+class NdarrayCodec(DataframeColumnCodec):
+    """Store an ndarray as .npy bytes (reference petastorm/codecs.py:133-171).
+
+    The .npy container keeps dtype+shape self-describing; the MI355X decode
+    path parses the 128-byte-aligned header on the host and bulk-copies /
+    casts the payload on-GPU (``npy_unpack`` kernel).
+    """
+
+    def encode(self, unischema_field, value):
+        _check_dtype(unischema_field, value)
+        _check_shape_compliance(unischema_field, value)
+        memfile = io.BytesIO()
+        np.save(memfile, value)
+        return memfile.getvalue()
+
+    def decode(self, unischema_field, value):
+        memfile = io.BytesIO(value)
+        return np.load(memfile, allow_pickle=False)
+
+    def arrow_type(self, unischema_field):
+        import pyarrow as pa
+        return pa.binary()
+
+    def to_dict(self):
+        return {'type': 'NdarrayCodec'}
+
+
+class CompressedNdarrayCodec(DataframeColumnCodec):
+    """Store an ndarray zlib-compressed (reference petastorm/codecs.py:174-212).
+
+    The reference uses ``np.savez_compressed`` (a zip container holding a
+    deflate-compressed .npy).  We store ``zlib.compress(npy_bytes)`` directly:
+    same codec family (DEFLATE), self-describing payload, and a simpler
+    framing for the GPU ``npz_inflate`` kernel.  Round-trip behavior is
+    identical.
+    """
+
+    def __init__(self, level=6):
+        self.level = level
+
+    def encode(self, unischema_field, value):
+        _check_dtype(unischema_field, value)
+        _check_shape_compliance(unischema_field, value)
+        memfile = io.BytesIO()
+        np.save(memfile, value)
+        return zlib.compress(memfile.getvalue(), self.level)
+
+    def decode(self, unischema_field, value):
+        raw = zlib.decompress(value)
+        return np.load(io.BytesIO(raw), allow_pickle=False)
+
+    def arrow_type(self, unischema_field):
+        import pyarrow as pa
+        return pa.binary()
+
+    def to_dict(self):
+        return {'type': 'CompressedNdarrayCodec', 'level': self.level}
+
+
+class CompressedImageCodec(DataframeColumnCodec):
+    """png/jpeg image codec (reference petastorm/codecs.py:58-130).
+
+    Accepts HxW (grayscale) or HxWx3 (RGB) uint8 arrays (png also uint16
+    grayscale).  Unlike the cv2-based reference there is no BGR<->RGB
+    swizzle: values are RGB end to end (the reference's swizzle at
+    codecs.py:92,112 exists only because OpenCV is BGR-native — net
+    behavior, RGB in == RGB out, is identical).
+    """
+
+    def __init__(self, image_codec='png', quality=80):
+        if image_codec not in ('png', 'jpeg', 'jpg'):
+            raise ValueError('Unsupported image codec: {}'.format(image_codec))
+        self._image_codec = 'jpeg' if image_codec in ('jpeg', 'jpg') else 'png'
+        self.quality = quality
+
+    @property
+    def image_codec(self):
+        return self._image_codec
+
+    def encode(self, unischema_field, value):
+        from PIL import Image
+        if unischema_field.numpy_dtype is not None and \
+                np.dtype(unischema_field.numpy_dtype) != value.dtype:
+            raise ValueError('Field {!r}: expected dtype {}, got {}'
+                             .format(unischema_field.name,
+                                     np.dtype(unischema_field.numpy_dtype),
+                                     value.dtype))
+        _check_shape_compliance(unischema_field, value)
+        if value.ndim == 3 and value.shape[2] == 1:
+            value = value[:, :, 0]
+        img = Image.fromarray(value)
+        buf = io.BytesIO()
+        if self._image_codec == 'jpeg':
+            if value.dtype != np.uint8:
+                raise ValueError('jpeg requires uint8 images')
+            # restart_marker_rows=1 -> one RSTn per MCU row: enables the
+            # restart-segment-parallel Huffman decode on the GPU.
+            img.save(buf, format='JPEG', quality=self.quality,
+                     restart_marker_rows=1)
+        else:
+            img.save(buf, format='PNG')
+        return buf.getvalue()
+
+    def decode(self, unischema_field, value):
+        from PIL import Image
+        img = Image.open(io.BytesIO(value))
+        arr = np.asarray(img)
+        if unischema_field.numpy_dtype is not None:
+            expected = np.dtype(unischema_field.numpy_dtype)
+            if arr.dtype != expected:
+                arr = arr.astype(expected)
+        return arr
+
+    def arrow_type(self, unischema_field):
+        import pyarrow as pa
+        return pa.binary()
+
+    def to_dict(self):
+        return {'type': 'CompressedImageCodec', 'image_codec': self._image_codec,
+                'quality': self.quality}
+
+
+# ---------------------------------------------------------------------------
+# codec <-> JSON descriptors, default-codec resolution
+# ---------------------------------------------------------------------------
+
+_CODEC_REGISTRY = {
+    'ScalarCodec': lambda d: ScalarCodec(),
+    'NdarrayCodec': lambda d: NdarrayCodec(),
+    'CompressedNdarrayCodec': lambda d: CompressedNdarrayCodec(d.get('level', 6)),
+    'CompressedImageCodec': lambda d: CompressedImageCodec(
+        d.get('image_codec', 'png'), d.get('quality', 80)),
+}
+
+
+def codec_to_dict(codec):
+    if codec is None:
+        return None
+    d = codec.to_dict()
+    if d['type'] not in _CODEC_REGISTRY:
+        raise ValueError('Unregistered codec type {!r}'.format(d['type']))
+    return d
+
+
+def codec_from_dict(d):
+    if d is None:
+        return None
+    return _CODEC_REGISTRY[d['type']](d)
+
+
+def effective_codec(unischema_field):
+    """Resolve the codec to use for a field: explicit, or dtype-default.
+
+    Reference semantics: a ``None`` codec defaults to scalar handling for
+    ``shape == ()`` and ndarray handling otherwise
+    (petastorm/unischema.py:69-76).
+    """
+    if unischema_field.codec is not None:
+        return unischema_field.codec
+    if unischema_field.shape == ():
+        return _DEFAULT_SCALAR
+    return _DEFAULT_NDARRAY
+
+
+_DEFAULT_SCALAR = ScalarCodec()
+_DEFAULT_NDARRAY = NdarrayCodec()
+
+
+def decode_value(unischema_field, value):
+    """Decode one stored value with error wrapping.
+
+    Reference: petastorm/utils.py:52-85 (decode_row's per-field behavior).
+    """
+    if value is None:
+        return None
+    try:
+        return effective_codec(unischema_field).decode(unischema_field, value)
+    except Exception as e:
+        raise DecodeFieldError('Unable to decode field {!r}: {}'
+                               .format(unischema_field.name, e)) from e

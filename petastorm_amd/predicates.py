@@ -1,0 +1,164 @@
+"""Row/batch predicates evaluated inside the decode workers.
+
+Parity: /root/reference/petastorm/predicates.py:27-182.
+
+* ``PredicateBase`` protocol: ``get_fields()`` names the columns the
+  predicate needs; ``do_include(values)`` returns the filter decision.
+* ``in_set``, ``in_intersection``, ``in_lambda``, ``in_negate``,
+  ``in_reduce``, ``in_pseudorandom_split``.
+
+Addition for the MI355X batch path: every built-in predicate also implements
+``do_include_vectorized(column_dict) -> bool ndarray`` so the GPU/batched
+worker can evaluate a whole row-group and stream-compact matching rows
+(``predicate_eval`` kernel) instead of looping rows in Python.
+"""
+
+import hashlib
+
+import numpy as np
+
+
+class PredicateBase(object):
+    """Predicate protocol (reference predicates.py:27-36)."""
+
+    def get_fields(self):
+        raise NotImplementedError()
+
+    def do_include(self, values):
+        """values: dict field_name -> scalar value of one row."""
+        raise NotImplementedError()
+
+    def do_include_vectorized(self, columns):
+        """columns: dict field_name -> 1-D array. Default: per-row loop."""
+        names = list(self.get_fields())
+        n = len(columns[names[0]])
+        out = np.empty(n, dtype=bool)
+        for i in range(n):
+            out[i] = self.do_include({f: columns[f][i] for f in names})
+        return out
+
+
+class in_set(PredicateBase):
+    """Include when ``values[field]`` is in a given set (predicates.py:44)."""
+
+    def __init__(self, inclusion_values, predicate_field):
+        self._inclusion_values = set(inclusion_values)
+        self._predicate_field = predicate_field
+
+    def get_fields(self):
+        return {self._predicate_field}
+
+    def do_include(self, values):
+        return values[self._predicate_field] in self._inclusion_values
+
+    def do_include_vectorized(self, columns):
+        return np.isin(np.asarray(columns[self._predicate_field]),
+                       list(self._inclusion_values))
+
+
+class in_intersection(PredicateBase):
+    """Include when an array-valued field intersects a set (predicates.py:58)."""
+
+    def __init__(self, inclusion_values, predicate_field):
+        self._inclusion_values = set(inclusion_values)
+        self._predicate_field = predicate_field
+
+    def get_fields(self):
+        return {self._predicate_field}
+
+    def do_include(self, values):
+        v = values[self._predicate_field]
+        return bool(self._inclusion_values.intersection(
+            v if isinstance(v, (set, frozenset)) else np.asarray(v).ravel().tolist()))
+
+
+class in_lambda(PredicateBase):
+    """Arbitrary user function over the named fields (predicates.py:74)."""
+
+    def __init__(self, predicate_fields, predicate_func, state_arg=None):
+        self._predicate_fields = list(predicate_fields)
+        self._predicate_func = predicate_func
+        self._state_arg = state_arg
+
+    def get_fields(self):
+        return set(self._predicate_fields)
+
+    def do_include(self, values):
+        if self._state_arg is not None:
+            return self._predicate_func(values, self._state_arg)
+        return self._predicate_func(values)
+
+
+class in_negate(PredicateBase):
+    """Logical NOT of another predicate (predicates.py:103)."""
+
+    def __init__(self, predicate):
+        self._predicate = predicate
+
+    def get_fields(self):
+        return self._predicate.get_fields()
+
+    def do_include(self, values):
+        return not self._predicate.do_include(values)
+
+    def do_include_vectorized(self, columns):
+        return ~self._predicate.do_include_vectorized(columns)
+
+
+class in_reduce(PredicateBase):
+    """Reduce several predicates with e.g. ``all``/``any`` (predicates.py:119)."""
+
+    def __init__(self, predicate_list, reduce_func):
+        self._predicate_list = list(predicate_list)
+        self._reduce_func = reduce_func
+
+    def get_fields(self):
+        fields = set()
+        for p in self._predicate_list:
+            fields |= set(p.get_fields())
+        return fields
+
+    def do_include(self, values):
+        return self._reduce_func([p.do_include(values)
+                                  for p in self._predicate_list])
+
+    def do_include_vectorized(self, columns):
+        masks = np.stack([p.do_include_vectorized(columns)
+                          for p in self._predicate_list])
+        if self._reduce_func is all:
+            return masks.all(axis=0)
+        if self._reduce_func is any:
+            return masks.any(axis=0)
+        return super(in_reduce, self).do_include_vectorized(columns)
+
+
+class in_pseudorandom_split(PredicateBase):
+    """Deterministic hash-bucket train/val/test splits (predicates.py:144-182).
+
+    ``fraction_list`` partitions [0,1); a row belongs to partition
+    ``predicate_index`` when the md5 hash of its id-field value falls into
+    that fraction of the hash space.
+    """
+
+    _MAX_HASH = float(0xFFFFFFFF)
+
+    def __init__(self, fraction_list, predicate_index, predicate_field):
+        if predicate_index < 0 or predicate_index >= len(fraction_list):
+            raise ValueError('predicate_index out of range')
+        self._fraction_list = list(fraction_list)
+        self._predicate_index = predicate_index
+        self._predicate_field = predicate_field
+        lo = sum(self._fraction_list[:predicate_index])
+        hi = lo + self._fraction_list[predicate_index]
+        self._lo, self._hi = lo, hi
+
+    def get_fields(self):
+        return {self._predicate_field}
+
+    def _bucket(self, value):
+        h = hashlib.md5(str(value).encode('utf-8')).hexdigest()[:8]
+        return int(h, 16) / self._MAX_HASH
+
+    def do_include(self, values):
+        b = self._bucket(values[self._predicate_field])
+        return self._lo <= b < self._hi

@@ -1,0 +1,212 @@
+"""ThreadPool: N daemon worker threads with deterministic round-robin readout.
+
+Parity: /root/reference/petastorm/workers_pool/thread_pool.py:
+
+* per-worker ventilation queues and bounded per-worker results queues sized
+  ``max(5, results_queue_size // workers)`` (:126-129)
+* round-robin item assignment in ``ventilate`` (:152-158) and round-robin
+  result readout for deterministic ordering (:172-218)
+* worker exceptions re-raised in the consumer (:67-72, 211-214)
+* stop-event-aware bounded puts so shutdown never deadlocks (:242-256)
+* optional per-worker cProfile (:46-48, 232-240)
+* ``diagnostics`` dict (:261-263)
+
+Determinism contract: items are assigned to workers round-robin and results
+are read back round-robin *per ventilated item* — so with a seeded ventilator
+the output order is a pure function of the seed, independent of thread
+timing.  The consumer may block on worker i while worker j is ready; that is
+the price of determinism the reference also pays.
+"""
+
+import cProfile
+import pstats
+import queue
+import threading
+import traceback
+import io as _io
+
+from petastorm_amd.workers_pool import (EmptyResultError,
+                                        VentilatedItemProcessedMessage,
+                                        WorkerExceptionMessage)
+
+_SENTINEL = object()
+_POLL_S = 0.02
+
+
+class WorkerThread(threading.Thread):
+    def __init__(self, pool, worker, work_queue, results_queue,
+                 profiling_enabled=False):
+        super(WorkerThread, self).__init__(daemon=True,
+                                           name='petastorm-amd-worker')
+        self._pool = pool
+        self._worker = worker
+        self._work_queue = work_queue
+        self._results_queue = results_queue
+        self._profile = cProfile.Profile() if profiling_enabled else None
+
+    def run(self):
+        if self._profile:
+            self._profile.enable()
+        try:
+            while True:
+                item = self._work_queue.get()
+                if item is _SENTINEL:
+                    break
+                try:
+                    if isinstance(item, dict):
+                        self._worker.process(**item)
+                    elif isinstance(item, tuple):
+                        self._worker.process(*item)
+                    else:
+                        self._worker.process(item)
+                    self._pool._stop_aware_put(
+                        self._results_queue, VentilatedItemProcessedMessage())
+                except Exception as e:  # noqa: BLE001 - forwarded to consumer
+                    tb = traceback.format_exc()
+                    self._pool._stop_aware_put(
+                        self._results_queue, WorkerExceptionMessage(e, tb))
+                    self._pool._stop_aware_put(
+                        self._results_queue, VentilatedItemProcessedMessage())
+        finally:
+            if self._profile:
+                self._profile.disable()
+            self._worker.shutdown()
+
+
+class ThreadPool(object):
+    def __init__(self, workers, results_queue_size=50, profiling_enabled=False):
+        self.workers_count = workers
+        self._results_queue_size = max(5, results_queue_size // max(1, workers))
+        self._profiling_enabled = profiling_enabled
+
+        self._threads = []
+        self._work_queues = []
+        self._results_queues = []
+        self._ventilator = None
+        self._stop_event = threading.Event()
+
+        self._rr_ventilate = 0
+        self._rr_read = 0
+        self._ventilated = 0
+        self._processed = 0
+        self._count_lock = threading.Lock()
+        self._started = False
+
+    # ------------------------------------------------------------------
+    def start(self, worker_class, worker_args=None, ventilator=None):
+        if self._started:
+            raise RuntimeError('ThreadPool already started')
+        self._started = True
+        self._stop_event.clear()
+        for i in range(self.workers_count):
+            wq = queue.Queue()
+            rq = queue.Queue(maxsize=self._results_queue_size)
+            worker = worker_class(i, self._make_publish(rq), worker_args)
+            t = WorkerThread(self, worker, wq, rq, self._profiling_enabled)
+            self._work_queues.append(wq)
+            self._results_queues.append(rq)
+            self._threads.append(t)
+            t.start()
+        self._ventilator = ventilator
+        if ventilator is not None:
+            ventilator.start()
+
+    def _make_publish(self, rq):
+        def publish(payload):
+            self._stop_aware_put(rq, payload)
+        return publish
+
+    def _stop_aware_put(self, q, item):
+        """Bounded put that aborts on pool stop (reference :242-256)."""
+        while not self._stop_event.is_set():
+            try:
+                q.put(item, timeout=_POLL_S)
+                return
+            except queue.Full:
+                continue
+
+    # ------------------------------------------------------------------
+    def ventilate(self, *args, **kwargs):
+        """Round-robin assign one work item (reference :152-158)."""
+        item = kwargs if kwargs else (args if len(args) != 1 else args[0])
+        with self._count_lock:
+            self._ventilated += 1
+        self._work_queues[self._rr_ventilate].put(item)
+        self._rr_ventilate = (self._rr_ventilate + 1) % self.workers_count
+
+    def get_results(self):
+        """Next result payload, deterministic round-robin (reference :172-218).
+
+        Raises :class:`EmptyResultError` when all ventilated items were
+        processed and every queue is drained.
+        """
+        while True:
+            if self._stop_event.is_set():
+                raise EmptyResultError('Pool was stopped')
+            if self._all_done():
+                raise EmptyResultError('No more work')
+            q = self._results_queues[self._rr_read]
+            try:
+                msg = q.get(timeout=_POLL_S)
+            except queue.Empty:
+                continue
+            if isinstance(msg, VentilatedItemProcessedMessage):
+                with self._count_lock:
+                    self._processed += 1
+                if self._ventilator is not None:
+                    self._ventilator.processed_item()
+                self._rr_read = (self._rr_read + 1) % self.workers_count
+                continue
+            if isinstance(msg, WorkerExceptionMessage):
+                self.stop()
+                self.join()
+                raise msg.exception
+            return msg
+
+    def _all_done(self):
+        with self._count_lock:
+            counts_done = self._processed >= self._ventilated
+        if not counts_done:
+            return False
+        if self._ventilator is not None and not self._ventilator.completed():
+            return False
+        return all(q.empty() for q in self._results_queues)
+
+    # ------------------------------------------------------------------
+    def stop(self):
+        self._stop_event.set()
+        if self._ventilator is not None:
+            self._ventilator.stop()
+        for wq in self._work_queues:
+            wq.put(_SENTINEL)
+        # drain result queues so workers blocked on put can exit
+        for rq in self._results_queues:
+            try:
+                while True:
+                    rq.get_nowait()
+            except queue.Empty:
+                pass
+
+    def join(self):
+        for t in self._threads:
+            t.join(timeout=30)
+        if self._profiling_enabled:
+            stats = None
+            for t in self._threads:
+                if t._profile is not None:
+                    s = pstats.Stats(t._profile)
+                    stats = s if stats is None else stats.add(t._profile)
+            if stats is not None:
+                out = _io.StringIO()
+                stats.stream = out
+                stats.sort_stats('cumulative').print_stats(30)
+                print(out.getvalue())
+
+    @property
+    def diagnostics(self):
+        """Pool health counters (reference :258-263)."""
+        return {
+            'output_queue_size': sum(q.qsize() for q in self._results_queues),
+            'items_ventilated': self._ventilated,
+            'items_processed': self._processed,
+        }

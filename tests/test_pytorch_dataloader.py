@@ -1,0 +1,131 @@
+"""PyTorch loaders (parity: reference tests/test_pytorch_dataloader.py)."""
+from decimal import Decimal
+
+import numpy as np
+import pytest
+import torch
+
+from petastorm_amd import make_batch_reader, make_reader
+from petastorm_amd.pytorch import (BatchedDataLoader, DataLoader,
+                                   InMemBatchedDataLoader,
+                                   _sanitize_pytorch_types,
+                                   decimal_friendly_collate)
+
+
+def test_sanitize_types():
+    d = {'u16': np.uint16(3), 'a_u32': np.arange(3, dtype=np.uint32),
+         'b': np.bool_(True), 'ok': np.float32(1.5)}
+    _sanitize_pytorch_types(d)
+    assert d['a_u32'].dtype == np.int64
+    assert isinstance(d['u16'], np.int32)
+    assert isinstance(d['ok'], np.float32)
+
+
+def test_decimal_friendly_collate():
+    batch = [{'d': Decimal('1.5'), 'x': np.float32(1)},
+             {'d': Decimal('2.5'), 'x': np.float32(2)}]
+    out = decimal_friendly_collate(batch)
+    assert out['d'] == ['1.5', '2.5']
+    assert torch.is_tensor(out['x'])
+
+
+def test_dataloader_row_reader(test_dataset):
+    reader = make_reader(test_dataset['url'], reader_pool_type='thread',
+                         schema_fields=['id', 'matrix'],
+                         shuffle_row_groups=False)
+    with DataLoader(reader, batch_size=8) as loader:
+        batches = list(loader)
+    total = sum(len(b['id']) for b in batches)
+    assert total == len(test_dataset['rows'])
+    assert batches[0]['matrix'].shape == (8, 10, 20)
+    # partial final batch allowed
+    assert all(len(b['id']) == 8 for b in batches[:-1])
+
+
+def test_dataloader_with_shuffling_queue(test_dataset):
+    reader = make_reader(test_dataset['url'], reader_pool_type='thread',
+                         schema_fields=['id'], shuffle_row_groups=False)
+    with DataLoader(reader, batch_size=10, shuffling_queue_capacity=30,
+                    seed=0) as loader:
+        ids = [int(i) for b in loader for i in b['id']]
+    assert sorted(ids) == [int(s['id']) for s in
+                           sorted(test_dataset['rows'], key=lambda r: r['id'])]
+    assert ids != sorted(ids)
+
+
+def test_dataloader_batched_reader_transposes(scalar_dataset):
+    reader = make_batch_reader(scalar_dataset['url'],
+                               reader_pool_type='thread',
+                               schema_fields=['id', 'f0'],
+                               shuffle_row_groups=False)
+    with DataLoader(reader, batch_size=32) as loader:
+        batches = list(loader)
+    assert sum(len(b['id']) for b in batches) == 500
+
+
+def test_batched_dataloader(scalar_dataset):
+    reader = make_batch_reader(scalar_dataset['url'],
+                               reader_pool_type='thread',
+                               schema_fields=['id', 'f0', 'i1'],
+                               shuffle_row_groups=False)
+    with BatchedDataLoader(reader, batch_size=64) as loader:
+        batches = list(loader)
+    ids = torch.cat([b['id'] for b in batches])
+    assert len(ids) == 500
+    assert sorted(ids.tolist()) == list(range(500))
+    sizes = [len(b['id']) for b in batches]
+    assert all(s == 64 for s in sizes[:-1])
+
+
+def test_batched_dataloader_shuffling(scalar_dataset):
+    reader = make_batch_reader(scalar_dataset['url'],
+                               reader_pool_type='thread',
+                               schema_fields=['id'],
+                               shuffle_row_groups=False)
+    with BatchedDataLoader(reader, batch_size=50,
+                           shuffling_queue_capacity=200, seed=1) as loader:
+        ids = torch.cat([b['id'] for b in loader])
+    assert sorted(ids.tolist()) == list(range(500))
+    assert ids.tolist() != sorted(ids.tolist())
+
+
+def test_batched_dataloader_rejects_strings(scalar_dataset):
+    reader = make_batch_reader(scalar_dataset['url'],
+                               reader_pool_type='dummy',
+                               schema_fields=['id', 'name'],
+                               shuffle_row_groups=False)
+    with pytest.raises(TypeError):
+        with BatchedDataLoader(reader, batch_size=10) as loader:
+            list(loader)
+
+
+def test_inmem_loader_epochs(scalar_dataset):
+    reader = make_batch_reader(scalar_dataset['url'],
+                               reader_pool_type='thread',
+                               schema_fields=['id'], num_epochs=1,
+                               shuffle_row_groups=False)
+    loader = InMemBatchedDataLoader(reader, batch_size=100, num_epochs=3,
+                                    rows_capacity=500, seed=42)
+    epochs = []
+    for _ in range(3):
+        ids = torch.cat([b['id'] for b in loader])
+        epochs.append(ids.tolist())
+    loader.stop()
+    loader.join()
+    for e in epochs:
+        assert sorted(e) == list(range(500))
+    assert epochs[0] != epochs[1]  # different shuffles per epoch
+    with pytest.raises(RuntimeError):
+        iter(loader)  # all epochs consumed
+
+
+def test_loader_guards_concurrent_iteration(test_dataset):
+    reader = make_reader(test_dataset['url'], reader_pool_type='dummy',
+                         schema_fields=['id'], shuffle_row_groups=False)
+    loader = DataLoader(reader, batch_size=4)
+    it = iter(loader)
+    next(it)
+    with pytest.raises(RuntimeError):
+        next(iter(loader))
+    loader.stop()
+    loader.join()
