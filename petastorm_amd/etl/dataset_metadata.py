@@ -111,13 +111,32 @@ class DatasetWriter(object):
             full = posixpath.join(self._path, fname)
             self._writer = pq.ParquetWriter(
                 self._fs.open(full, 'wb'), self._arrow_schema,
-                compression=self._compression,
+                compression=self._column_compression(),
                 # one table write == one row group:
                 use_dictionary=False, write_statistics=True,
                 data_page_size=1 << 20)
         self._writer.write_table(table)
         self._buffer = []
         self._buffer_bytes = 0
+
+    def _column_compression(self):
+        """Per-column codec choice: columns whose payloads are already
+        compressed (jpeg/png images, zlib ndarrays) are stored UNCOMPRESSED —
+        re-snappy-ing them wastes CPU at write time AND, for images, lets the
+        MI355X jpeg decoder parse headers straight from the pinned host
+        buffer without a decompress round-trip."""
+        from petastorm_amd.codecs import (CompressedImageCodec,
+                                          CompressedNdarrayCodec)
+        if self._compression == 'none':
+            return 'none'
+        spec = {}
+        for f in self._schema.fields.values():
+            if isinstance(f.codec, (CompressedImageCodec,
+                                    CompressedNdarrayCodec)):
+                spec[f.name] = 'none'
+            else:
+                spec[f.name] = self._compression
+        return spec
 
     def close(self):
         self._flush_row_group()

@@ -1,0 +1,475 @@
+"""GPU row-group decoder: raw Parquet column-chunk bytes -> device tensors.
+
+Replaces the Arrow C++ decode inside ``piece.read`` of the reference
+(petastorm/arrow_reader_worker.py:358) with HIP kernels:
+
+* page headers: native thrift-compact walk (ops csrc/thrift_pages.cpp, host)
+* snappy pages: ``snappy_decompress_batch`` (wave-per-page)
+* PLAIN fixed-width values: ``varlen_gather`` (funnel-shift copy) straight
+  into the column tensor
+* RLE/bit-packed definition levels & dictionary indices:
+  ``rle_hybrid_decode_batch``
+* PLAIN byte-array values: per-value (offset, length) tables; blob bytes are
+  decoded *in place* by the codec kernels (jpeg/npy) — never copied per value
+* NdarrayCodec: ``npy_payload_offsets`` + ``varlen_gather`` into a dense
+  [n, *shape] tensor (reference np.load, petastorm/codecs.py:155-157)
+* CompressedImageCodec(jpeg): restart-parallel decode (ops csrc/jpeg.hip)
+
+Columns whose physical encoding falls outside the GPU fast path (strings for
+Python consumption, png/zlib payloads until the inflate kernel lands,
+exotic encodings) are decoded with the CPU codec path and uploaded — the
+decoder reports which columns took the assist so benchmarks and tests can
+assert the hot path is native.
+"""
+
+import numpy as np
+import torch
+
+from petastorm_amd import codecs as _codecs
+from petastorm_amd import ops
+from petastorm_amd.codecs import (CompressedImageCodec, NdarrayCodec)
+
+# parquet enums
+_ENC_PLAIN = 0
+_ENC_PLAIN_DICT = 2
+_ENC_RLE = 3
+_ENC_RLE_DICT = 8
+_PAGE_DATA_V1 = 0
+_PAGE_DICT = 2
+_PAGE_DATA_V2 = 3
+
+_PHYS_TO_TORCH = {
+    'INT32': (torch.int32, 4),
+    'INT64': (torch.int64, 8),
+    'FLOAT': (torch.float32, 4),
+    'DOUBLE': (torch.float64, 8),
+}
+
+_SLACK = 16  # kernels may read a few bytes past the end of a stream
+
+
+class ByteArrayColumn(object):
+    """A decoded-to-offsets binary column: values live in ``device_buf``
+    (and, when the chunk was stored uncompressed, also in ``host_buf`` at
+    ``host_off`` — the jpeg host parser needs CPU visibility)."""
+
+    def __init__(self, device_buf, val_off, val_len, host_buf=None,
+                 host_val_off=None, n=0):
+        self.device_buf = device_buf
+        self.val_off = val_off          # int64 device tensor [n]
+        self.val_len = val_len          # int32 device tensor [n]
+        self.host_buf = host_buf        # uint8 cpu tensor or None
+        self.host_val_off = host_val_off  # int64 cpu tensor or None
+        self.n = n
+
+
+class GpuRowGroupDecoder(object):
+    def __init__(self, device='cuda'):
+        self.device = torch.device(device)
+        self._ext = ops.ext()
+        self.cpu_assist_columns = set()
+
+    # ------------------------------------------------------------------
+    def read_rowgroup_bytes(self, path, file_metadata, parquet_schema, rg,
+                            columns, pinned_pool=None):
+        """Read the raw byte span covering the requested column chunks into a
+        (pinned) host buffer.  Returns (host_buf, chunk_meta)."""
+        md = file_metadata.row_group(rg)
+        name_to_idx = {}
+        for ci in range(md.num_columns):
+            name_to_idx[md.column(ci).path_in_schema] = ci
+        chunks = []
+        lo, hi = None, 0
+        for name in columns:
+            ci = name_to_idx[name]
+            col = md.column(ci)
+            start = col.data_page_offset
+            if col.dictionary_page_offset is not None:
+                start = min(start, col.dictionary_page_offset)
+            end = start + col.total_compressed_size
+            lo = start if lo is None else min(lo, start)
+            hi = max(hi, end)
+            chunks.append((name, ci, start, col.total_compressed_size,
+                           col.physical_type, col.compression,
+                           col.num_values,
+                           parquet_schema.column(ci).max_definition_level))
+        nbytes = hi - lo
+        if pinned_pool is not None:
+            host = pinned_pool.get(nbytes + _SLACK)
+        else:
+            host = torch.empty(nbytes + _SLACK, dtype=torch.uint8,
+                               pin_memory=torch.cuda.is_available())
+        with open(path, 'rb') as f:
+            f.seek(lo)
+            mv = memoryview(host.numpy())[:nbytes]
+            f.readinto(mv)
+        chunk_meta = {
+            'base_offset': lo,
+            'num_rows': md.num_rows,
+            'chunks': [
+                dict(name=n, col_index=ci, offset=s - lo, length=ln,
+                     physical=pt, compression=comp, num_values=nv,
+                     max_def=mdl)
+                for (n, ci, s, ln, pt, comp, nv, mdl) in chunks],
+        }
+        return host, chunk_meta
+
+    # ------------------------------------------------------------------
+    def decode(self, host_buf, chunk_meta, schema, stream=None):
+        """Decode the requested columns.  Returns dict name ->
+        torch tensor (fixed columns) or ByteArrayColumn (binary)."""
+        ext = self._ext
+        dev = self.device
+        n_rows = chunk_meta['num_rows']
+        dbuf = host_buf.to(dev, non_blocking=True)
+
+        out = {}
+        for ch in chunk_meta['chunks']:
+            name = ch['name']
+            pages = ext.parquet_walk_pages(
+                host_buf, torch.tensor([ch['offset']], dtype=torch.int64),
+                torch.tensor([ch['length']], dtype=torch.int64))
+            comp = ch['compression']
+            if comp not in ('UNCOMPRESSED', 'SNAPPY'):
+                out[name] = self._cpu_assist_marker(name)
+                continue
+            col = self._decode_chunk(ext, dev, dbuf, host_buf, ch, pages,
+                                     n_rows, schema)
+            out[name] = col
+        return out, dbuf
+
+    def _cpu_assist_marker(self, name):
+        self.cpu_assist_columns.add(name)
+        return None
+
+    # ------------------------------------------------------------------
+    def _decode_chunk(self, ext, dev, dbuf, host_buf, ch, pages, n_rows,
+                      schema):
+        page_type = pages['page_type'].numpy()
+        data_off = pages['data_off'].numpy()       # relative to chunk walk
+        comp_size = pages['comp_size'].numpy()
+        uncomp_size = pages['uncomp_size'].numpy()
+        num_values = pages['num_values'].numpy()
+        encoding = pages['encoding'].numpy()
+        n_pages = len(page_type)
+        snappy = ch['compression'] == 'SNAPPY'
+
+        # 1) page payload location: either in dbuf directly, or in a
+        #    decompressed scratch buffer
+        if snappy:
+            total_un = int(uncomp_size.sum())
+            ubuf = torch.empty(total_un + _SLACK, dtype=torch.uint8,
+                               device=dev)
+            u_off = np.zeros(n_pages + 1, dtype=np.int64)
+            u_off[1:] = np.cumsum(uncomp_size)
+            c_off = np.zeros(n_pages + 1, dtype=np.int64)
+            c_off[:-1] = data_off
+            c_off[-1] = data_off[-1] + comp_size[-1]
+            status = torch.zeros(n_pages, dtype=torch.int32, device=dev)
+            ext.snappy_decompress_batch(
+                dbuf, torch.from_numpy(c_off).to(dev),
+                ubuf, torch.from_numpy(u_off).to(dev), status)
+            self._check(status, 'snappy:' + ch['name'])
+            page_buf = ubuf
+            page_start = u_off[:-1]
+            host_visible = False
+        else:
+            page_buf = dbuf
+            page_start = data_off.astype(np.int64)
+            host_visible = True
+
+        # 2) split dict page / data pages
+        dict_idx = [i for i in range(n_pages) if page_type[i] == _PAGE_DICT]
+        data_idx = [i for i in range(n_pages)
+                    if page_type[i] in (_PAGE_DATA_V1, _PAGE_DATA_V2)]
+        if not data_idx:
+            return torch.empty(0, device=dev)
+        data_enc = encoding[data_idx[0]]
+        for i in data_idx:
+            if page_type[i] == _PAGE_DATA_V2:
+                return self._cpu_assist_marker(ch['name'])
+
+        max_def = ch['max_def']
+        phys = ch['physical']
+
+        # 3) locate per-page def-level and value sections
+        val_start = np.empty(len(data_idx), dtype=np.int64)
+        val_end = np.empty(len(data_idx), dtype=np.int64)
+        def_start = np.empty(len(data_idx), dtype=np.int64)
+        def_end = np.empty(len(data_idx), dtype=np.int64)
+        page_nval = num_values[data_idx]
+        if max_def > 0:
+            if not host_visible:
+                # def-level section length prefix lives in the decompressed
+                # buffer; fetch the 4-byte prefixes (one small D2H)
+                prefs = torch.stack([
+                    page_buf[page_start[i]:page_start[i] + 4]
+                    for i in data_idx]).cpu().numpy()
+                dl_len = (prefs[:, 0].astype(np.int64)
+                          | (prefs[:, 1].astype(np.int64) << 8)
+                          | (prefs[:, 2].astype(np.int64) << 16)
+                          | (prefs[:, 3].astype(np.int64) << 24))
+            else:
+                hb = host_buf.numpy()
+                dl_len = np.empty(len(data_idx), dtype=np.int64)
+                for j, i in enumerate(data_idx):
+                    p = page_start[i]
+                    dl_len[j] = int.from_bytes(
+                        hb[p:p + 4].tobytes(), 'little')
+            for j, i in enumerate(data_idx):
+                def_start[j] = page_start[i] + 4
+                def_end[j] = def_start[j] + dl_len[j]
+                val_start[j] = def_end[j]
+                val_end[j] = page_start[i] + uncomp_size[i] if snappy \
+                    else page_start[i] + comp_size[i]
+        else:
+            for j, i in enumerate(data_idx):
+                val_start[j] = page_start[i]
+                val_end[j] = page_start[i] + (uncomp_size[i] if snappy
+                                              else comp_size[i])
+
+        # 4) definition levels -> validity
+        valid = None
+        nonnull_per_page = None
+        if max_def > 0:
+            lv_off = np.zeros(len(data_idx) + 1, dtype=np.int64)
+            lv_off[1:] = np.cumsum(page_nval)
+            levels = torch.empty(int(lv_off[-1]), dtype=torch.int32,
+                                 device=dev)
+            status = torch.zeros(len(data_idx), dtype=torch.int32,
+                                 device=dev)
+            ext.rle_hybrid_decode_batch(
+                page_buf, torch.from_numpy(def_start).to(dev),
+                torch.from_numpy(def_end).to(dev),
+                torch.ones(len(data_idx), dtype=torch.int32, device=dev),
+                torch.from_numpy(page_nval.astype(np.int32)).to(dev),
+                torch.from_numpy(lv_off[:-1]).to(dev), levels, status)
+            self._check(status, 'deflevels:' + ch['name'])
+            valid = levels.bool()
+            # per-page non-null counts (needed to place value sections)
+            vmat = valid.split([int(x) for x in page_nval])
+            nonnull_per_page = np.array([int(v.sum().item()) for v in vmat],
+                                        dtype=np.int64)
+
+        # 5) values by encoding
+        if data_enc == _ENC_PLAIN and phys in _PHYS_TO_TORCH:
+            return self._plain_fixed(ext, dev, page_buf, val_start, val_end,
+                                     page_nval, nonnull_per_page, valid,
+                                     n_rows, phys)
+        if data_enc == _ENC_PLAIN and phys == 'BYTE_ARRAY':
+            return self._plain_byte_array(
+                ext, dev, page_buf, host_buf, val_start, val_end, page_nval,
+                nonnull_per_page, valid, host_visible, ch)
+        if data_enc in (_ENC_PLAIN_DICT, _ENC_RLE_DICT) and \
+                phys in _PHYS_TO_TORCH and dict_idx:
+            return self._dict_fixed(ext, dev, page_buf, page_start, dict_idx,
+                                    num_values, uncomp_size if snappy
+                                    else comp_size, val_start, val_end,
+                                    page_nval, nonnull_per_page, valid,
+                                    n_rows, phys)
+        return self._cpu_assist_marker(ch['name'])
+
+    # ------------------------------------------------------------------
+    def _plain_fixed(self, ext, dev, page_buf, val_start, val_end, page_nval,
+                     nonnull_per_page, valid, n_rows, phys):
+        dtype, esize = _PHYS_TO_TORCH[phys]
+        counts = nonnull_per_page if nonnull_per_page is not None \
+            else page_nval
+        nbytes = counts * esize
+        dst_off = np.zeros(len(counts), dtype=np.int64)
+        dst_off[1:] = np.cumsum(nbytes)[:-1]
+        total_vals = int(counts.sum())
+        flat = torch.empty(total_vals * esize + _SLACK, dtype=torch.uint8,
+                           device=dev)
+        ext.varlen_gather(page_buf,
+                          torch.from_numpy(val_start).to(dev),
+                          torch.from_numpy(nbytes.astype(np.int64)).to(dev),
+                          flat, torch.from_numpy(dst_off).to(dev))
+        values = flat[:total_vals * esize].view(dtype)
+        if valid is None:
+            return values
+        out = torch.zeros(n_rows, dtype=dtype, device=dev)
+        if dtype.is_floating_point:
+            out.fill_(float('nan'))
+        out[valid] = values
+        return out
+
+    def _plain_byte_array(self, ext, dev, page_buf, host_buf, val_start,
+                          val_end, page_nval, nonnull_per_page, valid,
+                          host_visible, ch):
+        counts = nonnull_per_page if nonnull_per_page is not None \
+            else page_nval
+        total = int(counts.sum())
+        o_off = np.zeros(len(counts), dtype=np.int64)
+        o_off[1:] = np.cumsum(counts)[:-1]
+        val_off = torch.empty(total, dtype=torch.int64, device=dev)
+        val_len = torch.empty(total, dtype=torch.int32, device=dev)
+        status = torch.zeros(len(counts), dtype=torch.int32, device=dev)
+        ext.byte_array_offsets_batch(
+            page_buf, torch.from_numpy(val_start).to(dev),
+            torch.from_numpy(val_end).to(dev),
+            torch.from_numpy(counts.astype(np.int32)).to(dev),
+            torch.from_numpy(o_off).to(dev), val_off, val_len, status)
+        self._check(status, 'bytearray:' + ch['name'])
+        host_off = None
+        if host_visible:
+            # mirror the scan on host (cheap: one u32 read per value) so
+            # codecs needing header parsing (jpeg) can see the bytes
+            hb = host_buf.numpy()
+            host_off = np.empty(total, dtype=np.int64)
+            k = 0
+            for j in range(len(counts)):
+                pos = int(val_start[j])
+                for _ in range(int(counts[j])):
+                    ln = int.from_bytes(hb[pos:pos + 4].tobytes(), 'little')
+                    host_off[k] = pos + 4
+                    k += 1
+                    pos += 4 + ln
+        return ByteArrayColumn(page_buf, val_off, val_len,
+                               host_buf if host_visible else None,
+                               host_off, total)
+
+    def _dict_fixed(self, ext, dev, page_buf, page_start, dict_idx,
+                    num_values, size_arr, val_start, val_end, page_nval,
+                    nonnull_per_page, valid, n_rows, phys):
+        dtype, esize = _PHYS_TO_TORCH[phys]
+        di = dict_idx[0]
+        dict_n = int(num_values[di])
+        dstart = int(page_start[di])
+        flat = torch.empty(dict_n * esize + _SLACK, dtype=torch.uint8,
+                           device=dev)
+        ext.varlen_gather(page_buf,
+                          torch.tensor([dstart], dtype=torch.int64,
+                                       device=dev),
+                          torch.tensor([dict_n * esize], dtype=torch.int64,
+                                       device=dev),
+                          flat,
+                          torch.tensor([0], dtype=torch.int64, device=dev))
+        dict_vals = flat[:dict_n * esize].view(dtype)
+
+        counts = nonnull_per_page if nonnull_per_page is not None \
+            else page_nval
+        # data page: first byte = bit width, then hybrid runs
+        bw = np.empty(len(val_start), dtype=np.int32)
+        # read bit-width bytes (tiny D2H)
+        bw_t = torch.stack([page_buf[int(s)] for s in val_start]).cpu()
+        bw[:] = bw_t.numpy()
+        i_off = np.zeros(len(counts) + 1, dtype=np.int64)
+        i_off[1:] = np.cumsum(counts)
+        indices = torch.empty(int(i_off[-1]), dtype=torch.int32, device=dev)
+        status = torch.zeros(len(counts), dtype=torch.int32, device=dev)
+        ext.rle_hybrid_decode_batch(
+            page_buf, torch.from_numpy(val_start + 1).to(dev),
+            torch.from_numpy(val_end).to(dev),
+            torch.from_numpy(bw).to(dev),
+            torch.from_numpy(counts.astype(np.int32)).to(dev),
+            torch.from_numpy(i_off[:-1]).to(dev), indices, status)
+        self._check(status, 'dictidx:' + phys)
+        values = dict_vals[indices.long()]
+        if valid is None:
+            return values
+        out = torch.zeros(n_rows, dtype=dtype, device=dev)
+        if dtype.is_floating_point:
+            out.fill_(float('nan'))
+        out[valid] = values
+        return out
+
+    # ------------------------------------------------------------------
+    def _check(self, status, what):
+        s = int(status.abs().sum().item())
+        if s != 0:
+            raise RuntimeError('GPU decode error in {}: status={}'
+                               .format(what, status.cpu().tolist()))
+
+    # ------------------------------------------------------------------
+    # codec stages over ByteArrayColumn
+    # ------------------------------------------------------------------
+    def decode_ndarray_column(self, col, field):
+        """NdarrayCodec: npy containers -> dense [n, *shape] tensor."""
+        ext = self._ext
+        dev = self.device
+        np_dtype = np.dtype(field.numpy_dtype)
+        shape = tuple(field.shape)
+        if any(d is None for d in shape):
+            return None  # variable shape -> CPU assist
+        elem = int(np.prod(shape)) if shape else 1
+        row_bytes = elem * np_dtype.itemsize
+        pay_off = torch.empty(col.n, dtype=torch.int64, device=dev)
+        pay_len = torch.empty(col.n, dtype=torch.int64, device=dev)
+        status = torch.zeros(1, dtype=torch.int32, device=dev)
+        ext.npy_payload_offsets(col.device_buf, col.val_off, col.val_len,
+                                pay_off, pay_len, status)
+        self._check(status, 'npy:' + field.name)
+        out = torch.empty(col.n * row_bytes + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        dst_off = torch.arange(col.n, dtype=torch.int64, device=dev) \
+            * row_bytes
+        ext.varlen_gather(col.device_buf, pay_off, pay_len, out, dst_off)
+        # reinterpret raw bytes at the SAME width, then widen unsigned types
+        # torch can't represent (mirrors _sanitize_pytorch_types, reference
+        # pytorch.py:40-70)
+        view_dtype = _np_view_torch(np_dtype)
+        if view_dtype is None:
+            return None
+        t = out[:col.n * row_bytes].view(view_dtype).view((col.n,) + shape)
+        if np_dtype == np.dtype(np.uint16):
+            t = t.to(torch.int32) & 0xFFFF
+        elif np_dtype == np.dtype(np.uint32):
+            t = t.to(torch.int64) & 0xFFFFFFFF
+        return t
+
+    def decode_jpeg_column(self, col, field):
+        """CompressedImageCodec(jpeg): restart-parallel GPU decode."""
+        if col.host_buf is None:
+            return None  # compressed storage: host can't parse headers
+        ext = self._ext
+        dev = self.device
+        host_off = torch.from_numpy(col.host_val_off)
+        host_len = col.val_len.cpu()
+        meta = ext.jpeg_parse_batch(col.host_buf, host_off, host_len)
+        n = col.n
+        widths = meta['width'].numpy()
+        heights = meta['height'].numpy()
+        ncomp = meta['ncomp'].numpy()
+        # move every tensor to device
+        meta_dev = {}
+        for k, v in meta.items():
+            meta_dev[k] = v.to(dev) if isinstance(v, torch.Tensor) else v
+        block_total = int(meta['block_total'])
+        samp_total = int(meta['samp_total'])
+        coef = torch.zeros(block_total * 64, dtype=torch.float32, device=dev)
+        samples = torch.empty(samp_total, dtype=torch.uint8, device=dev)
+        out_bytes = (widths.astype(np.int64) * heights *
+                     np.where(ncomp == 3, 3, 1))
+        out_off = np.zeros(n, dtype=np.int64)
+        out_off[1:] = np.cumsum(out_bytes)[:-1]
+        out = torch.empty(int(out_bytes.sum()), dtype=torch.uint8,
+                          device=dev)
+        n_segs = int(meta['seg_img'].numel())
+        status = torch.zeros(max(n_segs, 1), dtype=torch.int32, device=dev)
+        ext.jpeg_decode_batch(col.device_buf, meta_dev, coef, samples, out,
+                              torch.from_numpy(out_off).to(dev), status)
+        self._check(status, 'jpeg:' + field.name)
+        # uniform-shape batch -> dense [n, H, W, C]
+        if len(set(widths.tolist())) == 1 and len(set(heights.tolist())) == 1 \
+                and len(set(ncomp.tolist())) == 1:
+            c = 3 if ncomp[0] == 3 else 1
+            t = out.view(n, int(heights[0]), int(widths[0]), c)
+            return t if c == 3 else t.squeeze(-1)
+        return None
+
+
+def _np_view_torch(np_dtype):
+    """torch dtype of the SAME byte width (for raw reinterpretation)."""
+    m = {np.dtype(np.float32): torch.float32,
+         np.dtype(np.float64): torch.float64,
+         np.dtype(np.int64): torch.int64,
+         np.dtype(np.int32): torch.int32,
+         np.dtype(np.int16): torch.int16,
+         np.dtype(np.int8): torch.int8,
+         np.dtype(np.uint8): torch.uint8,
+         np.dtype(np.uint16): torch.int16,
+         np.dtype(np.uint32): torch.int32,
+         np.dtype(np.uint64): torch.int64,
+         np.dtype(np.bool_): torch.bool}
+    return m.get(np_dtype)

@@ -1,0 +1,335 @@
+"""GpuBatchReader: the MI355X HIP pipeline behind
+``make_batch_reader(..., device='cuda')``.
+
+The reference's worker pools (thread/process + zmq) become:
+
+* an IO prefetch thread filling *pinned host buffers* with raw row-group
+  bytes (depth-2 queue = the ventilator's backpressure, reference
+  reader.py:45,489)
+* async H2D copies + HIP decode kernels on the current stream
+  (petastorm_amd.gpu.decoder) — the decode "workers" are stream slots on the
+  same device
+* batches are HBM-resident torch tensors; the shuffling queue and batching
+  happen on-device in BatchedDataLoader
+
+Sharding follows the reference rule ``index % shard_count == cur_shard``
+(reference reader.py:573-597), the per-epoch permutation is broadcast from
+rank 0 over RCCL when torch.distributed is initialized, and epoch
+boundaries all-gather per-rank row counts (petastorm_amd.parallel.epochs).
+"""
+
+import logging
+import queue
+import threading
+
+import numpy as np
+import torch
+
+from petastorm_amd.errors import NoDataAvailableError
+from petastorm_amd.etl import dataset_metadata as dsm
+from petastorm_amd.gpu.decoder import ByteArrayColumn, GpuRowGroupDecoder
+from petastorm_amd.gpu.hbm_cache import HbmCache
+from petastorm_amd.ngram import NGram
+from petastorm_amd.parallel import epochs as epoch_sync
+from petastorm_amd.transform import transform_schema
+from petastorm_amd.unischema import match_unischema_fields
+from petastorm_amd.codecs import (CompressedImageCodec, NdarrayCodec,
+                                  CompressedNdarrayCodec)
+
+logger = logging.getLogger(__name__)
+
+_PREFETCH_DEPTH = 2
+
+
+class _PinnedPool(object):
+    """Reusable pinned host buffers (rounded up to 1 MiB steps)."""
+
+    def __init__(self):
+        self._free = {}
+        self._lock = threading.Lock()
+
+    def get(self, nbytes):
+        size = ((nbytes + (1 << 20) - 1) >> 20) << 20
+        with self._lock:
+            lst = self._free.get(size)
+            if lst:
+                return lst.pop()[:nbytes] if False else lst.pop()
+        return torch.empty(size, dtype=torch.uint8,
+                           pin_memory=torch.cuda.is_available())
+
+    def put(self, buf):
+        with self._lock:
+            self._free.setdefault(buf.numel(), []).append(buf)
+
+
+class GpuBatchReader(object):
+    def __init__(self, fs, path_or_paths, schema_fields=None,
+                 shuffle_row_groups=True, shuffle_rows=False, predicate=None,
+                 num_epochs=1, cur_shard=None, shard_count=None, seed=None,
+                 transform_spec=None, device='cuda',
+                 cache_type=None, cache_size_limit=None):
+        if isinstance(schema_fields, NGram):
+            raise NotImplementedError('NGram is a make_reader feature; use '
+                                      'the sequence reader path')
+        self._fs = fs
+        self._paths = path_or_paths
+        self.device = torch.device(device)
+        self._decoder = GpuRowGroupDecoder(self.device)
+        self._pin_pool = _PinnedPool()
+
+        storage_schema, _ = dsm.infer_or_load_unischema(fs, path_or_paths)
+        if schema_fields is not None:
+            matched = match_unischema_fields(storage_schema, schema_fields)
+            if not matched:
+                raise ValueError('schema_fields matched nothing')
+            self._view_schema = storage_schema.create_schema_view(matched)
+        else:
+            self._view_schema = storage_schema
+        self._storage_schema = storage_schema
+        self.transform_spec = transform_spec
+        self.schema = transform_schema(self._view_schema, transform_spec) \
+            if transform_spec else self._view_schema
+        self.batched_output = True
+        self.ngram = None
+        self.last_row_consumed = False
+
+        self._pieces = dsm.load_row_groups(fs, path_or_paths)
+        if not self._pieces:
+            raise NoDataAvailableError('Dataset has no row groups')
+        cur_shard, shard_count = epoch_sync.shard_for_rank(cur_shard,
+                                                           shard_count)
+        self._cur_shard, self._shard_count = cur_shard, shard_count
+        if shard_count is not None and shard_count > len(self._pieces):
+            raise NoDataAvailableError(
+                'Number of row-groups ({}) < shard_count ({})'
+                .format(len(self._pieces), shard_count))
+        self._seed = seed
+        self._shuffle_row_groups = shuffle_row_groups
+        self._shuffle_rows = shuffle_rows
+        self._predicate = predicate
+        self._num_epochs = num_epochs
+
+        self._cache = HbmCache(cache_size_limit) \
+            if cache_type == 'hbm' and cache_size_limit else None
+
+        # per-file metadata handles (footer parse once per file)
+        self._file_md = {}
+        self._stopped = False
+        self._rows_epoch = 0
+        self._gen = self._generate()
+
+    # ------------------------------------------------------------------
+    def _metadata(self, path):
+        if path not in self._file_md:
+            import pyarrow.parquet as pq
+            pf = pq.ParquetFile(self._fs.open(path, 'rb'))
+            self._file_md[path] = (pf.metadata, pf.schema)
+        return self._file_md[path]
+
+    def _epoch_pieces(self, epoch):
+        n = len(self._pieces)
+        perm = epoch_sync.epoch_permutation(
+            n, epoch, self._seed, self._shuffle_row_groups)
+        if self._shard_count is not None:
+            perm = [p for pos, p in enumerate(perm)
+                    if pos % self._shard_count == self._cur_shard]
+        return [self._pieces[i] for i in perm]
+
+    # ------------------------------------------------------------------
+    def _io_worker(self, pieces, out_q):
+        columns = list(self._view_schema.fields.keys())
+        try:
+            for piece in pieces:
+                if self._stopped:
+                    break
+                if self._cache is not None and \
+                        self._cache_key(piece) in self._cache._store:
+                    out_q.put(('cached', piece, None, None))
+                    continue
+                md, pschema = self._metadata(piece.path)
+                host, meta = self._decoder.read_rowgroup_bytes(
+                    piece.path, md, pschema, piece.row_group, columns,
+                    self._pin_pool)
+                out_q.put(('data', piece, host, meta))
+            out_q.put(('end', None, None, None))
+        except Exception as e:  # noqa: BLE001 - forwarded to consumer
+            out_q.put(('error', e, None, None))
+
+    @staticmethod
+    def _cache_key(piece):
+        return '{}:{}'.format(piece.path, piece.row_group)
+
+    # ------------------------------------------------------------------
+    def _generate(self):
+        epoch = 0
+        while self._num_epochs is None or epoch < self._num_epochs:
+            pieces = self._epoch_pieces(epoch)
+            q = queue.Queue(maxsize=_PREFETCH_DEPTH)
+            t = threading.Thread(target=self._io_worker, args=(pieces, q),
+                                 daemon=True)
+            t.start()
+            self._rows_epoch = 0
+            while True:
+                kind, piece, host, meta = q.get()
+                if kind == 'end':
+                    break
+                if kind == 'error':
+                    raise piece
+                if kind == 'cached':
+                    columns = self._cache._store[self._cache_key(piece)]
+                    self._cache.get(self._cache_key(piece), lambda: columns)
+                else:
+                    if self._cache is not None:
+                        columns = self._cache.get(
+                            self._cache_key(piece),
+                            lambda: self._decode_piece(piece, host, meta))
+                    else:
+                        columns = self._decode_piece(piece, host, meta)
+                    self._pin_pool.put(host) if host is not None else None
+                batch = self._postprocess(piece, columns)
+                if batch is None:
+                    continue
+                self._rows_epoch += len(next(iter(batch.values())))
+                yield self.schema.make_namedtuple(**batch)
+            t.join()
+            epoch_sync.epoch_end_sync(self._rows_epoch)
+            epoch += 1
+
+    # ------------------------------------------------------------------
+    def _decode_piece(self, piece, host, meta):
+        raw, dbuf = self._decoder.decode(host, meta, self._storage_schema)
+        columns = {}
+        assist = []
+        for name, col in raw.items():
+            field = self._storage_schema.fields.get(name)
+            if col is None:
+                assist.append(name)
+                continue
+            if isinstance(col, ByteArrayColumn):
+                decoded = None
+                codec = field.codec if field is not None else None
+                if isinstance(codec, CompressedImageCodec) and \
+                        codec.image_codec == 'jpeg':
+                    decoded = self._decoder.decode_jpeg_column(col, field)
+                elif isinstance(codec, NdarrayCodec) or (
+                        codec is None and field is not None and
+                        field.shape not in ((), None)):
+                    decoded = self._decoder.decode_ndarray_column(col, field)
+                if decoded is None:
+                    assist.append(name)
+                else:
+                    columns[name] = decoded
+            else:
+                columns[name] = col
+        if assist:
+            columns.update(self._cpu_assist(piece, assist))
+        return columns
+
+    def _cpu_assist(self, piece, names):
+        """CPU decode + upload for columns outside the GPU fast path."""
+        from petastorm_amd.workers.batch_worker import \
+            arrow_table_to_numpy_dict
+        md, _ = self._metadata(piece.path)
+        import pyarrow.parquet as pq
+        pf = pq.ParquetFile(self._fs.open(piece.path, 'rb'))
+        table = pf.read_row_group(piece.row_group, columns=names)
+        np_dict = arrow_table_to_numpy_dict(table, self._storage_schema, True)
+        out = {}
+        for k, v in np_dict.items():
+            if isinstance(v, np.ndarray) and v.dtype.kind in 'iufb':
+                out[k] = torch.from_numpy(np.ascontiguousarray(v)).to(
+                    self.device, non_blocking=True)
+            else:
+                out[k] = v  # strings/objects stay host-side
+        return out
+
+    # ------------------------------------------------------------------
+    def _postprocess(self, piece, columns):
+        columns = dict(columns)
+        n = None
+        for v in columns.values():
+            n = len(v)
+            break
+        if not n:
+            return None
+        if self._predicate is not None:
+            mask = self._predicate_mask(columns)
+            if mask is not None:
+                idx = torch.nonzero(mask, as_tuple=False).squeeze(1)
+                if idx.numel() == 0:
+                    return None
+                columns = {k: (v.index_select(0, idx)
+                               if isinstance(v, torch.Tensor)
+                               else v[idx.cpu().numpy()])
+                           for k, v in columns.items()}
+        if self._shuffle_rows:
+            n2 = len(next(iter(columns.values())))
+            g = None
+            if self._seed is not None:
+                g = torch.Generator(device='cpu')
+                g.manual_seed((self._seed + piece.index) % (2 ** 31))
+            perm = torch.randperm(n2, generator=g).to(self.device)
+            columns = {k: (v.index_select(0, perm)
+                           if isinstance(v, torch.Tensor)
+                           else v[perm.cpu().numpy()])
+                       for k, v in columns.items()}
+        if self.transform_spec is not None:
+            if self.transform_spec.func:
+                columns = self.transform_spec.func(columns)
+            keep = set(self.schema.fields.keys())
+            columns = {k: v for k, v in columns.items() if k in keep}
+        return columns
+
+    def _predicate_mask(self, columns):
+        names = list(self._predicate.get_fields())
+        # predicate columns are typically small scalars: evaluate with the
+        # numpy vectorized path, apply the mask on-device
+        host_cols = {}
+        for f in names:
+            v = columns[f]
+            host_cols[f] = v.cpu().numpy() if isinstance(v, torch.Tensor) \
+                else np.asarray(v)
+        mask = self._predicate.do_include_vectorized(host_cols)
+        return torch.from_numpy(np.asarray(mask, dtype=bool)).to(self.device)
+
+    # ------------------------------------------------------------------
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        if self._stopped:
+            raise StopIteration
+        try:
+            return next(self._gen)
+        except StopIteration:
+            self.last_row_consumed = True
+            raise
+
+    next = __next__
+
+    def reset(self):
+        self._gen = self._generate()
+        self.last_row_consumed = False
+
+    def stop(self):
+        self._stopped = True
+
+    def join(self):
+        if self._cache is not None:
+            self._cache.cleanup()
+
+    @property
+    def diagnostics(self):
+        d = {'cpu_assist_columns': sorted(self._decoder.cpu_assist_columns)}
+        if self._cache is not None:
+            d.update(hbm_cache_hits=self._cache.hits,
+                     hbm_cache_misses=self._cache.misses,
+                     hbm_cache_bytes=self._cache.size_bytes)
+        return d
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, exc_type, exc_val, exc_tb):
+        self.stop()
+        self.join()

@@ -1,0 +1,68 @@
+// petastorm_amd HIP extension: python bindings.
+//
+// Every function here either launches gfx950 HIP kernels (snappy, parquet
+// page decode, jpeg, layout transforms) or performs the host half of a GPU
+// pipeline stage (jpeg header parse, parquet page-header walk).  There is no
+// CPU fallback in this module by design: if it imports, the GPU path is the
+// path that runs.
+#include <torch/extension.h>
+
+namespace psa {
+
+// snappy.hip
+void snappy_decompress_batch(torch::Tensor comp, torch::Tensor comp_offsets,
+                             torch::Tensor out, torch::Tensor out_offsets,
+                             torch::Tensor status);
+// parquet_decode.hip
+void rle_hybrid_decode_batch(torch::Tensor data, torch::Tensor start,
+                             torch::Tensor end, torch::Tensor bit_width,
+                             torch::Tensor n_values, torch::Tensor out_off,
+                             torch::Tensor out, torch::Tensor status);
+void byte_array_offsets_batch(torch::Tensor data, torch::Tensor start,
+                              torch::Tensor end, torch::Tensor n_values,
+                              torch::Tensor out_off, torch::Tensor val_off,
+                              torch::Tensor val_len, torch::Tensor status);
+void varlen_gather(torch::Tensor src, torch::Tensor src_off,
+                   torch::Tensor lengths, torch::Tensor dst,
+                   torch::Tensor dst_off);
+void npy_payload_offsets(torch::Tensor data, torch::Tensor val_off,
+                         torch::Tensor val_len, torch::Tensor pay_off,
+                         torch::Tensor pay_len, torch::Tensor status);
+// transforms.hip
+void nhwc_to_nchw_normalize(torch::Tensor in, torch::Tensor out,
+                            torch::Tensor mean, torch::Tensor inv_std,
+                            double scale);
+// jpeg_host.cpp
+py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
+                          torch::Tensor val_len);
+// thrift_pages.cpp
+py::dict parquet_walk_pages(torch::Tensor buf, torch::Tensor chunk_off,
+                            torch::Tensor chunk_len);
+// jpeg.hip
+void jpeg_decode_batch(torch::Tensor data, py::dict meta, torch::Tensor coef,
+                       torch::Tensor samples, torch::Tensor out,
+                       torch::Tensor out_off, torch::Tensor status);
+
+}  // namespace psa
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "petastorm_amd MI355X (gfx950) decode kernels";
+  m.def("snappy_decompress_batch", &psa::snappy_decompress_batch,
+        "Batched snappy page decompression (wave-per-page)");
+  m.def("rle_hybrid_decode_batch", &psa::rle_hybrid_decode_batch,
+        "Parquet RLE/bit-packed hybrid decode (levels & dict indices)");
+  m.def("byte_array_offsets_batch", &psa::byte_array_offsets_batch,
+        "PLAIN byte-array page -> per-value (offset, length)");
+  m.def("varlen_gather", &psa::varlen_gather,
+        "Unaligned variable-length byte gather (funnel-shift copy)");
+  m.def("npy_payload_offsets", &psa::npy_payload_offsets,
+        ".npy container -> payload (offset, length)");
+  m.def("nhwc_to_nchw_normalize", &psa::nhwc_to_nchw_normalize,
+        "Fused uint8 NHWC -> float NCHW normalize (LDS-tiled)");
+  m.def("parquet_walk_pages", &psa::parquet_walk_pages,
+        "Walk Parquet page headers (thrift compact) in a raw column chunk");
+  m.def("jpeg_parse_batch", &psa::jpeg_parse_batch,
+        "Host-side JPEG header/segment parse");
+  m.def("jpeg_decode_batch", &psa::jpeg_decode_batch,
+        "GPU baseline JPEG decode (huffman/idct/color kernels)");
+}

@@ -1,0 +1,307 @@
+// GPU Parquet page decode: RLE/bit-packed hybrid, PLAIN byte-array offset
+// extraction, unaligned variable-length gather, .npy payload location.
+//
+// Replaces the Arrow C++ Parquet value decoders the reference reaches through
+// piece.read() (reference petastorm/arrow_reader_worker.py:358,
+// py_dict_reader_worker.py:267) and the NdarrayCodec np.load
+// (reference petastorm/codecs.py:155-157).
+//
+// Parallel structure mirrors snappy.hip: bitstream run headers are parsed by
+// lane 0 of a wave and the 64 lanes expand values in parallel; page-level
+// parallelism fills the 256 CUs.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include "common.h"
+
+namespace psa {
+
+// ---------------------------------------------------------------------------
+// RLE / bit-packed hybrid (Parquet spec "RLE" encoding): definition levels
+// and dictionary indices.
+//
+//   run-header varint h: (h & 1) == 0 -> RLE run of (h >> 1) copies of a
+//   ceil(bw/8)-byte LE value; (h & 1) == 1 -> (h >> 1) groups of 8
+//   bit-packed values, LSB-first.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint32_t read_varint_u32(const uint8_t* p,
+                                                    int64_t& pos,
+                                                    int64_t end) {
+  uint32_t result = 0;
+  int shift = 0;
+  while (pos < end && shift < 35) {
+    uint8_t b = p[pos++];
+    result |= (uint32_t)(b & 0x7f) << shift;
+    if (!(b & 0x80)) break;
+    shift += 7;
+  }
+  return result;
+}
+
+__device__ __forceinline__ uint32_t extract_bits(const uint8_t* base,
+                                                 int64_t bit_off, int bw) {
+  // read up to 24+8 bits spanning <=5 bytes
+  int64_t byte = bit_off >> 3;
+  int shift = (int)(bit_off & 7);
+  uint64_t w = 0;
+  for (int i = 0; i < 5; ++i) w |= (uint64_t)base[byte + i] << (8 * i);
+  return (uint32_t)((w >> shift) & ((1u << bw) - 1u));
+}
+
+// One wave per stream. streams are (data, start, end, bit_width, n_values,
+// out_offset) tuples.
+__global__ void rle_hybrid_decode_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
+    const int64_t* __restrict__ end, const int32_t* __restrict__ bit_width,
+    const int32_t* __restrict__ n_values, const int64_t* __restrict__ out_off,
+    int32_t* __restrict__ out, int32_t* __restrict__ status, int n_streams) {
+  const int waves_per_block = blockDim.x / PSA_WAVE;
+  const int s = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  if (s >= n_streams) return;
+  const int lane = lane_id();
+
+  const int bw = bit_width[s];
+  const int64_t lo = start[s], hi = end[s];
+  const int32_t want = n_values[s];
+  int32_t* dst = out + out_off[s];
+
+  if (bw == 0) {
+    // all values are zero (e.g. required column def levels)
+    for (int32_t i = lane; i < want; i += PSA_WAVE) dst[i] = 0;
+    return;
+  }
+  if (bw > 24) {
+    if (lane == 0) status[s] = 4;  // unsupported width
+    return;
+  }
+
+  int64_t pos = lo;
+  int32_t produced = 0;
+  while (produced < want) {
+    uint32_t header = 0;
+    int64_t payload = 0;
+    int32_t run_len = 0, is_packed = 0;
+    uint32_t rle_value = 0;
+    int done = 0;
+    if (lane == 0) {
+      if (pos >= hi) {
+        done = 1;
+      } else {
+        header = read_varint_u32(data, pos, hi);
+        is_packed = header & 1;
+        if (is_packed) {
+          run_len = (int32_t)(header >> 1) * 8;
+          payload = pos;               // bit-packed payload starts here
+          pos += (int64_t)(header >> 1) * bw;
+        } else {
+          run_len = (int32_t)(header >> 1);
+          int nbytes = (bw + 7) / 8;
+          rle_value = 0;
+          for (int i = 0; i < nbytes; ++i)
+            rle_value |= (uint32_t)data[pos + i] << (8 * i);
+          pos += nbytes;
+        }
+      }
+    }
+    done = wave_bcast(done);
+    if (done) {
+      if (lane == 0 && produced < want) status[s] = 5;  // truncated
+      break;
+    }
+    is_packed = wave_bcast(is_packed);
+    run_len = wave_bcast(run_len);
+    if (is_packed) {
+      payload = wave_bcast(payload);
+      int32_t emit = min(run_len, want - produced);
+      for (int32_t i = lane; i < emit; i += PSA_WAVE)
+        dst[produced + i] =
+            (int32_t)extract_bits(data, payload * 8 + (int64_t)i * bw, bw);
+      produced += emit;
+    } else {
+      rle_value = wave_bcast(rle_value);
+      int32_t emit = min(run_len, want - produced);
+      for (int32_t i = lane; i < emit; i += PSA_WAVE)
+        dst[produced + i] = (int32_t)rle_value;
+      produced += emit;
+    }
+  }
+}
+
+void rle_hybrid_decode_batch(torch::Tensor data, torch::Tensor start,
+                             torch::Tensor end, torch::Tensor bit_width,
+                             torch::Tensor n_values, torch::Tensor out_off,
+                             torch::Tensor out, torch::Tensor status) {
+  int n = (int)start.numel();
+  if (!n) return;
+  const int WPB = 4;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(rle_hybrid_decode_kernel,
+                     dim3((n + WPB - 1) / WPB), dim3(WPB * PSA_WAVE), 0,
+                     stream, data.data_ptr<uint8_t>(),
+                     start.data_ptr<int64_t>(), end.data_ptr<int64_t>(),
+                     bit_width.data_ptr<int32_t>(),
+                     n_values.data_ptr<int32_t>(),
+                     out_off.data_ptr<int64_t>(), out.data_ptr<int32_t>(),
+                     status.data_ptr<int32_t>(), n);
+}
+
+// ---------------------------------------------------------------------------
+// PLAIN byte-array pages: [u32 len][bytes]... -> absolute (offset, length)
+// per value.  Downstream kernels (jpeg/npy/inflate) read values in place —
+// no copy of the blob bytes.
+// ---------------------------------------------------------------------------
+
+__global__ void byte_array_offsets_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
+    const int64_t* __restrict__ end, const int32_t* __restrict__ n_values,
+    const int64_t* __restrict__ out_off, int64_t* __restrict__ val_off,
+    int32_t* __restrict__ val_len, int32_t* __restrict__ status,
+    int n_pages) {
+  // lane 0 of a wave scans its page (the scan per value is one u32 read);
+  // other lanes idle — page parallelism dominates.
+  const int waves_per_block = blockDim.x / PSA_WAVE;
+  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  if (page >= n_pages || lane_id() != 0) return;
+  int64_t pos = start[page];
+  const int64_t hi = end[page];
+  const int32_t want = n_values[page];
+  int64_t* o = val_off + out_off[page];
+  int32_t* l = val_len + out_off[page];
+  for (int32_t i = 0; i < want; ++i) {
+    if (pos + 4 > hi) { status[page] = 6; return; }
+    uint32_t len = load_u32_unaligned(data + pos);
+    pos += 4;
+    if (pos + len > hi) { status[page] = 6; return; }
+    o[i] = pos;
+    l[i] = (int32_t)len;
+    pos += len;
+  }
+}
+
+void byte_array_offsets_batch(torch::Tensor data, torch::Tensor start,
+                              torch::Tensor end, torch::Tensor n_values,
+                              torch::Tensor out_off, torch::Tensor val_off,
+                              torch::Tensor val_len, torch::Tensor status) {
+  int n = (int)start.numel();
+  if (!n) return;
+  const int WPB = 4;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(byte_array_offsets_kernel,
+                     dim3((n + WPB - 1) / WPB), dim3(WPB * PSA_WAVE), 0,
+                     stream, data.data_ptr<uint8_t>(),
+                     start.data_ptr<int64_t>(), end.data_ptr<int64_t>(),
+                     n_values.data_ptr<int32_t>(),
+                     out_off.data_ptr<int64_t>(),
+                     val_off.data_ptr<int64_t>(),
+                     val_len.data_ptr<int32_t>(),
+                     status.data_ptr<int32_t>(), n);
+}
+
+// ---------------------------------------------------------------------------
+// Unaligned variable-length gather: copy item i's bytes
+// src[src_off[i] .. +len[i]) -> dst[dst_off[i] ..).
+//
+// The src side is arbitrarily aligned (values sit mid-page); the copy uses
+// aligned u32 loads + funnel shift so each lane still moves 4B per
+// instruction (guide §6 Guideline 13: never scalar-byte a memory-bound loop).
+// One block per item, threads stride the length.
+// ---------------------------------------------------------------------------
+
+__global__ void varlen_gather_kernel(
+    const uint8_t* __restrict__ src, const int64_t* __restrict__ src_off,
+    const int64_t* __restrict__ len64, uint8_t* __restrict__ dst,
+    const int64_t* __restrict__ dst_off, int n_items) {
+  for (int item = blockIdx.x; item < n_items; item += gridDim.x) {
+    const int64_t n = len64[item];
+    const uint8_t* s = src + src_off[item];
+    uint8_t* d = dst + dst_off[item];
+    // head: bytes until dst is 4-aligned
+    int64_t head = min(n, (int64_t)((4 - ((uintptr_t)d & 3)) & 3));
+    for (int64_t i = threadIdx.x; i < head; i += blockDim.x) d[i] = s[i];
+    s += head; d += head;
+    int64_t body = (n - head) & ~(int64_t)3;
+    // aligned-u32 + funnel shift over the body
+    const int shift = (int)((uintptr_t)s & 3) * 8;
+    const uint32_t* s4 = (const uint32_t*)((uintptr_t)s & ~(uintptr_t)3);
+    uint32_t* d4 = (uint32_t*)d;
+    const int64_t words = body >> 2;
+    if (shift == 0) {
+      for (int64_t w = threadIdx.x; w < words; w += blockDim.x)
+        d4[w] = s4[w];
+    } else {
+      for (int64_t w = threadIdx.x; w < words; w += blockDim.x) {
+        uint32_t lo = s4[w], hi = s4[w + 1];
+        d4[w] = (lo >> shift) | (hi << (32 - shift));
+      }
+    }
+    // tail
+    for (int64_t i = body + threadIdx.x; i < n - head; i += blockDim.x)
+      d[i] = s[i];
+  }
+}
+
+void varlen_gather(torch::Tensor src, torch::Tensor src_off,
+                   torch::Tensor lengths, torch::Tensor dst,
+                   torch::Tensor dst_off) {
+  int n = (int)src_off.numel();
+  if (!n) return;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  int blocks = n < 2048 ? n : 2048;
+  hipLaunchKernelGGL(varlen_gather_kernel, dim3(blocks), dim3(256), 0,
+                     stream, src.data_ptr<uint8_t>(),
+                     src_off.data_ptr<int64_t>(),
+                     lengths.data_ptr<int64_t>(), dst.data_ptr<uint8_t>(),
+                     dst_off.data_ptr<int64_t>(), n);
+}
+
+// ---------------------------------------------------------------------------
+// .npy payload location: value bytes are a full .npy container
+// (reference NdarrayCodec, petastorm/codecs.py:133-171).  Header:
+//   \x93NUMPY <ver_major> <ver_minor> <u16 hlen> <hlen dict bytes> payload
+// This kernel turns (value offset, value length) into (payload offset,
+// payload length) so varlen_gather can assemble the dense batch tensor.
+// ---------------------------------------------------------------------------
+
+__global__ void npy_payload_offsets_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ val_off,
+    const int32_t* __restrict__ val_len, int64_t* __restrict__ pay_off,
+    int64_t* __restrict__ pay_len, int32_t* __restrict__ status, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const uint8_t* p = data + val_off[i];
+  if (val_len[i] < 10 || p[0] != 0x93 || p[1] != 'N' || p[2] != 'U') {
+    status[0] = 7;  // not an npy container
+    pay_off[i] = val_off[i];
+    pay_len[i] = 0;
+    return;
+  }
+  int hlen;
+  int hdr;
+  if (p[6] == 1) {           // version 1.0: u16 header length
+    hlen = (int)p[8] | ((int)p[9] << 8);
+    hdr = 10 + hlen;
+  } else {                   // version 2.0+: u32 header length
+    hlen = (int)p[8] | ((int)p[9] << 8) | ((int)p[10] << 16) |
+           ((int)p[11] << 24);
+    hdr = 12 + hlen;
+  }
+  pay_off[i] = val_off[i] + hdr;
+  pay_len[i] = (int64_t)val_len[i] - hdr;
+}
+
+void npy_payload_offsets(torch::Tensor data, torch::Tensor val_off,
+                         torch::Tensor val_len, torch::Tensor pay_off,
+                         torch::Tensor pay_len, torch::Tensor status) {
+  int n = (int)val_off.numel();
+  if (!n) return;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(npy_payload_offsets_kernel,
+                     dim3((n + 255) / 256), dim3(256), 0, stream,
+                     data.data_ptr<uint8_t>(), val_off.data_ptr<int64_t>(),
+                     val_len.data_ptr<int32_t>(), pay_off.data_ptr<int64_t>(),
+                     pay_len.data_ptr<int64_t>(),
+                     status.data_ptr<int32_t>(), n);
+}
+
+}  // namespace psa

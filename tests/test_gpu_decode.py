@@ -1,0 +1,528 @@
+"""GPU kernel numerics tests: every HIP kernel vs a plain CPU reference.
+
+All tests in this file require an MI355X (marked gpu).
+"""
+import io
+
+import numpy as np
+import pytest
+import torch
+
+from petastorm_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope='module')
+def ext():
+    assert torch.cuda.is_available(), 'gpu tests need a GPU'
+    return ops.ext()
+
+
+# ---------------------------------------------------------------------------
+# snappy
+# ---------------------------------------------------------------------------
+
+def _snappy_compress(data):
+    import pyarrow as pa
+    return pa.compress(data, codec='snappy', asbytes=True)
+
+
+def test_snappy_roundtrip_random_and_repetitive(ext):
+    rng = np.random.RandomState(0)
+    payloads = [
+        rng.randint(0, 255, 100000).astype(np.uint8).tobytes(),   # literals
+        (b'abcdefgh' * 20000),                                    # long copies
+        (b'\x00' * 65536),                                        # RLE-ish
+        rng.randint(0, 4, 50000).astype(np.uint8).tobytes(),      # mixed
+        b'x',                                                     # tiny
+    ]
+    comp = [_snappy_compress(p) for p in payloads]
+    comp_cat = b''.join(comp)
+    c_off = np.zeros(len(comp) + 1, dtype=np.int64)
+    c_off[1:] = np.cumsum([len(c) for c in comp])
+    u_off = np.zeros(len(payloads) + 1, dtype=np.int64)
+    u_off[1:] = np.cumsum([len(p) for p in payloads])
+
+    dev = 'cuda'
+    comp_t = torch.frombuffer(bytearray(comp_cat + b'\0' * 16),
+                              dtype=torch.uint8).to(dev)
+    out = torch.zeros(int(u_off[-1]) + 16, dtype=torch.uint8, device=dev)
+    status = torch.zeros(len(payloads), dtype=torch.int32, device=dev)
+    ext.snappy_decompress_batch(comp_t, torch.from_numpy(c_off).to(dev),
+                                out, torch.from_numpy(u_off).to(dev), status)
+    torch.cuda.synchronize()
+    assert status.cpu().tolist() == [0] * len(payloads)
+    got = out[:int(u_off[-1])].cpu().numpy().tobytes()
+    assert got == b''.join(payloads)
+
+
+# ---------------------------------------------------------------------------
+# RLE/bit-packed hybrid
+# ---------------------------------------------------------------------------
+
+def _write_uvarint(v):
+    out = bytearray()
+    while True:
+        b = v & 0x7f
+        v >>= 7
+        if v:
+            out.append(b | 0x80)
+        else:
+            out.append(b)
+            return bytes(out)
+
+
+def _encode_hybrid(values, bit_width):
+    """Simple encoder: alternating RLE and bit-packed runs."""
+    out = bytearray()
+    i = 0
+    n = len(values)
+    while i < n:
+        # run-length detect
+        j = i
+        while j < n and values[j] == values[i]:
+            j += 1
+        if j - i >= 8:
+            out += _write_uvarint((j - i) << 1)
+            v = int(values[i])
+            for k in range((bit_width + 7) // 8):
+                out.append((v >> (8 * k)) & 0xFF)
+            i = j
+        else:
+            # bit-pack the next up-to-504 values in groups of 8
+            take = min(n - i, 504)
+            groups = (take + 7) // 8
+            out += _write_uvarint((groups << 1) | 1)
+            bits = 0
+            acc = 0
+            cnt = 0
+            for k in range(groups * 8):
+                v = int(values[i + k]) if i + k < n else 0
+                acc |= v << bits
+                bits += bit_width
+                while bits >= 8:
+                    out.append(acc & 0xFF)
+                    acc >>= 8
+                    bits -= 8
+                cnt += 1
+            if bits:
+                out.append(acc & 0xFF)
+            i += take
+    return bytes(out)
+
+
+@pytest.mark.parametrize('bit_width', [1, 2, 5, 8, 12, 20])
+def test_rle_hybrid_decode(ext, bit_width):
+    rng = np.random.RandomState(bit_width)
+    maxv = (1 << bit_width) - 1
+    streams, expected = [], []
+    for s in range(6):
+        n = rng.randint(1, 3000)
+        vals = rng.randint(0, maxv + 1, n)
+        if s % 2 == 0:
+            vals[: n // 2] = vals[0]  # force RLE run
+        streams.append(_encode_hybrid(vals, bit_width))
+        expected.append(vals)
+    cat = b''.join(streams)
+    starts = np.zeros(len(streams), dtype=np.int64)
+    ends = np.zeros(len(streams), dtype=np.int64)
+    pos = 0
+    for i, s in enumerate(streams):
+        starts[i] = pos
+        pos += len(s)
+        ends[i] = pos
+    nvals = np.array([len(e) for e in expected], dtype=np.int32)
+    out_off = np.zeros(len(streams), dtype=np.int64)
+    out_off[1:] = np.cumsum(nvals)[:-1]
+
+    dev = 'cuda'
+    data = torch.frombuffer(bytearray(cat + b'\0' * 16),
+                            dtype=torch.uint8).to(dev)
+    out = torch.zeros(int(nvals.sum()), dtype=torch.int32, device=dev)
+    status = torch.zeros(len(streams), dtype=torch.int32, device=dev)
+    ext.rle_hybrid_decode_batch(
+        data, torch.from_numpy(starts).to(dev), torch.from_numpy(ends).to(dev),
+        torch.full((len(streams),), bit_width, dtype=torch.int32, device=dev),
+        torch.from_numpy(nvals).to(dev), torch.from_numpy(out_off).to(dev),
+        out, status)
+    torch.cuda.synchronize()
+    assert status.cpu().tolist() == [0] * len(streams)
+    got = out.cpu().numpy()
+    exp = np.concatenate(expected)
+    np.testing.assert_array_equal(got, exp)
+
+
+# ---------------------------------------------------------------------------
+# varlen gather
+# ---------------------------------------------------------------------------
+
+def test_varlen_gather_misaligned(ext):
+    rng = np.random.RandomState(1)
+    src = rng.randint(0, 255, 300000).astype(np.uint8)
+    items = []
+    pos = 1  # deliberately odd start
+    while pos + 2000 < len(src):
+        ln = int(rng.randint(1, 1999))
+        items.append((pos, ln))
+        pos += ln + int(rng.randint(0, 3))
+    dst_off = np.zeros(len(items), dtype=np.int64)
+    lens = np.array([ln for _, ln in items], dtype=np.int64)
+    dst_off[1:] = np.cumsum(lens)[:-1]
+    dev = 'cuda'
+    src_t = torch.from_numpy(src).to(dev)
+    dst = torch.zeros(int(lens.sum()) + 16, dtype=torch.uint8, device=dev)
+    ext.varlen_gather(src_t,
+                      torch.tensor([p for p, _ in items],
+                                   dtype=torch.int64, device=dev),
+                      torch.from_numpy(lens).to(dev), dst,
+                      torch.from_numpy(dst_off).to(dev))
+    torch.cuda.synchronize()
+    expected = np.concatenate([src[p:p + ln] for p, ln in items])
+    np.testing.assert_array_equal(dst[:len(expected)].cpu().numpy(), expected)
+
+
+# ---------------------------------------------------------------------------
+# full rowgroup decode vs pyarrow (the CPU oracle)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize('compression', ['snappy', 'none'])
+def test_scalar_rowgroup_decode_vs_pyarrow(ext, tmp_path, compression):
+    import pyarrow.parquet as pq
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    from petastorm_amd.gpu.decoder import GpuRowGroupDecoder
+    from petastorm_amd.etl import dataset_metadata as dsm
+    from petastorm_amd.fs_utils import get_filesystem_and_path_or_paths
+
+    url = 'file://' + str(tmp_path / ('ds_' + compression))
+    create_scalar_dataset(url, num_rows=5000, rowgroup_size=1024,
+                          compression=compression)
+    fs, path = get_filesystem_and_path_or_paths(url)
+    pieces = dsm.load_row_groups(fs, path)
+    schema, _ = dsm.infer_or_load_unischema(fs, path)
+    dec = GpuRowGroupDecoder('cuda')
+    cols = ['id', 'f0', 'f3', 'i2']
+    for piece in pieces[:3]:
+        pf = pq.ParquetFile(piece.path)
+        host, meta = dec.read_rowgroup_bytes(piece.path, pf.metadata,
+                                             pf.schema, piece.row_group, cols)
+        out, _ = dec.decode(host, meta, schema)
+        oracle = pf.read_row_group(piece.row_group, columns=cols)
+        for c in cols:
+            got = out[c].cpu().numpy()
+            exp = oracle.column(c).to_numpy()
+            np.testing.assert_array_equal(got, exp, err_msg=c)
+
+
+def test_nullable_rowgroup_decode(ext, tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd.gpu.decoder import GpuRowGroupDecoder
+    from petastorm_amd.unischema import Unischema
+
+    rng = np.random.RandomState(0)
+    vals = rng.rand(4000)
+    mask = rng.rand(4000) < 0.3
+    col = pa.array([None if m else float(v) for m, v in zip(mask, vals)],
+                   type=pa.float64())
+    ids = pa.array(np.arange(4000, dtype=np.int64))
+    path = str(tmp_path / 'nullable.parquet')
+    pq.write_table(pa.table({'id': ids, 'x': col}), path,
+                   compression='snappy', use_dictionary=False,
+                   row_group_size=1500)
+    pf = pq.ParquetFile(path)
+    schema = Unischema.from_arrow_schema(pf.schema_arrow)
+    dec = GpuRowGroupDecoder('cuda')
+    for rg in range(pf.metadata.num_row_groups):
+        host, meta = dec.read_rowgroup_bytes(path, pf.metadata, pf.schema,
+                                             rg, ['id', 'x'])
+        out, _ = dec.decode(host, meta, schema)
+        oracle = pf.read_row_group(rg, columns=['id', 'x'])
+        exp = oracle.column('x').to_numpy(zero_copy_only=False)
+        got = out['x'].cpu().numpy()
+        np.testing.assert_array_equal(np.isnan(got), np.isnan(exp))
+        np.testing.assert_allclose(got[~np.isnan(got)], exp[~np.isnan(exp)])
+
+
+def test_dictionary_encoded_decode(ext, tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd.gpu.decoder import GpuRowGroupDecoder
+    from petastorm_amd.unischema import Unischema
+
+    rng = np.random.RandomState(0)
+    vals = rng.randint(0, 50, 10000).astype(np.int64)  # few distinct values
+    path = str(tmp_path / 'dict.parquet')
+    pq.write_table(pa.table({'v': vals}), path, compression='snappy',
+                   use_dictionary=True, row_group_size=4000)
+    pf = pq.ParquetFile(path)
+    schema = Unischema.from_arrow_schema(pf.schema_arrow)
+    dec = GpuRowGroupDecoder('cuda')
+    for rg in range(pf.metadata.num_row_groups):
+        host, meta = dec.read_rowgroup_bytes(path, pf.metadata, pf.schema,
+                                             rg, ['v'])
+        out, _ = dec.decode(host, meta, schema)
+        exp = pf.read_row_group(rg, columns=['v']).column('v').to_numpy()
+        np.testing.assert_array_equal(out['v'].cpu().numpy(), exp)
+
+
+# ---------------------------------------------------------------------------
+# ndarray (npy) column
+# ---------------------------------------------------------------------------
+
+def test_ndarray_column_decode(ext, tmp_path):
+    from petastorm_amd.test_util.dataset_gen import (SequenceSchema,
+                                                     create_sequence_dataset)
+    from petastorm_amd.gpu.decoder import GpuRowGroupDecoder, ByteArrayColumn
+    from petastorm_amd.etl import dataset_metadata as dsm
+    from petastorm_amd.fs_utils import get_filesystem_and_path_or_paths
+    import pyarrow.parquet as pq
+
+    url = 'file://' + str(tmp_path / 'seq')
+    create_sequence_dataset(url, num_rows=100, rowgroup_size_mb=0.5)
+    fs, path = get_filesystem_and_path_or_paths(url)
+    pieces = dsm.load_row_groups(fs, path)
+    schema = dsm.get_schema(fs, path)
+    dec = GpuRowGroupDecoder('cuda')
+    piece = pieces[0]
+    pf = pq.ParquetFile(piece.path)
+    host, meta = dec.read_rowgroup_bytes(piece.path, pf.metadata, pf.schema,
+                                         piece.row_group,
+                                         ['timestamp', 'tokens'])
+    out, _ = dec.decode(host, meta, schema)
+    col = out['tokens']
+    assert isinstance(col, ByteArrayColumn)
+    decoded = dec.decode_ndarray_column(col, schema.fields['tokens'])
+    assert decoded.shape[1:] == (1024,)
+    # oracle: CPU codec decode
+    oracle = pf.read_row_group(piece.row_group, columns=['tokens'])
+    import io as _io
+    exp = np.stack([np.load(_io.BytesIO(v.as_py()))
+                    for v in oracle.column('tokens')])
+    np.testing.assert_array_equal(decoded.cpu().numpy(), exp)
+
+
+# ---------------------------------------------------------------------------
+# jpeg decode vs PIL
+# ---------------------------------------------------------------------------
+
+def _jpeg_batch_to_gpu(ext, blobs):
+    dev = 'cuda'
+    buf = b''.join(blobs)
+    off, lens, pos = [], [], 0
+    for d in blobs:
+        off.append(pos)
+        lens.append(len(d))
+        pos += len(d)
+    host = torch.frombuffer(bytearray(buf + b'\0' * 16), dtype=torch.uint8)
+    meta = ext.jpeg_parse_batch(host, torch.tensor(off, dtype=torch.int64),
+                                torch.tensor(lens, dtype=torch.int64))
+    dbuf = host.to(dev)
+    meta_dev = {k: (v.to(dev) if isinstance(v, torch.Tensor) else v)
+                for k, v in meta.items()}
+    n = len(blobs)
+    widths = meta['width'].numpy()
+    heights = meta['height'].numpy()
+    ncomp = meta['ncomp'].numpy()
+    coef = torch.zeros(int(meta['block_total']) * 64, dtype=torch.float32,
+                       device=dev)
+    samples = torch.empty(int(meta['samp_total']), dtype=torch.uint8,
+                          device=dev)
+    out_bytes = widths.astype(np.int64) * heights * np.where(ncomp == 3, 3, 1)
+    out_off = np.zeros(n, dtype=np.int64)
+    out_off[1:] = np.cumsum(out_bytes)[:-1]
+    out = torch.empty(int(out_bytes.sum()), dtype=torch.uint8, device=dev)
+    status = torch.zeros(max(1, int(meta['seg_img'].numel())),
+                         dtype=torch.int32, device=dev)
+    ext.jpeg_decode_batch(dbuf, meta_dev, coef, samples, out,
+                          torch.from_numpy(out_off).to(dev), status)
+    torch.cuda.synchronize()
+    assert int(status.abs().sum()) == 0, status.cpu()
+    outs = []
+    for i in range(n):
+        c = 3 if ncomp[i] == 3 else 1
+        img = out[out_off[i]:out_off[i] + out_bytes[i]].cpu().numpy()
+        img = img.reshape(heights[i], widths[i], c)
+        outs.append(img.squeeze(-1) if c == 1 else img)
+    return outs
+
+
+def _pil_decode(blobs):
+    from PIL import Image
+    return [np.asarray(Image.open(io.BytesIO(b))) for b in blobs]
+
+
+def _make_jpegs(n, size=(48, 64), quality=90, gray=False, subsampling=None,
+                smooth=False, seed=0):
+    from PIL import Image
+    rng = np.random.RandomState(seed)
+    blobs = []
+    h, w = size[1], size[0]
+    for i in range(n):
+        if smooth:
+            yy, xx = np.mgrid[0:h, 0:w].astype(np.float32)
+            base = np.sin(xx / 7 + i) * 60 + np.cos(yy / 9) * 50 + 128
+            if gray:
+                arr = np.clip(base, 0, 255).astype(np.uint8)
+            else:
+                arr = np.clip(np.stack([base, base * 0.8, 255 - base],
+                                       axis=-1), 0, 255).astype(np.uint8)
+        elif gray:
+            arr = rng.randint(0, 255, (h, w)).astype(np.uint8)
+        else:
+            arr = rng.randint(0, 255, (h, w, 3)).astype(np.uint8)
+        img = Image.fromarray(arr)
+        b = io.BytesIO()
+        kw = dict(format='JPEG', quality=quality, restart_marker_rows=1)
+        if subsampling is not None:
+            kw['subsampling'] = subsampling
+        img.save(b, **kw)
+        blobs.append(b.getvalue())
+    return blobs
+
+
+@pytest.mark.parametrize('subsampling,gray', [
+    (0, False),   # 4:4:4
+    (2, False),   # 4:2:0
+    (1, False),   # 4:2:2
+    (None, True),  # grayscale
+])
+def test_jpeg_decode_matches_pil(ext, subsampling, gray):
+    blobs = _make_jpegs(6, size=(48, 64), gray=gray, subsampling=subsampling,
+                        smooth=True)
+    got = _jpeg_batch_to_gpu(ext, blobs)
+    exp = _pil_decode(blobs)
+    for g, e in zip(got, exp):
+        assert g.shape == e.shape
+        diff = np.abs(g.astype(int) - e.astype(int))
+        assert diff.mean() < 1.5, diff.mean()
+        assert diff.max() <= 8, diff.max()
+
+
+def test_jpeg_decode_random_noise_image(ext):
+    # noise stresses the Huffman decoder (max bitstream entropy)
+    blobs = _make_jpegs(4, size=(96, 80), quality=95, subsampling=0)
+    got = _jpeg_batch_to_gpu(ext, blobs)
+    exp = _pil_decode(blobs)
+    for g, e in zip(got, exp):
+        diff = np.abs(g.astype(int) - e.astype(int))
+        assert diff.mean() < 2.0
+        assert diff.max() <= 12
+
+
+def test_jpeg_decode_odd_sizes(ext):
+    # sizes not multiples of the MCU: 50x35 (4:2:0 -> partial MCUs)
+    blobs = _make_jpegs(3, size=(35, 50), subsampling=2, smooth=True)
+    got = _jpeg_batch_to_gpu(ext, blobs)
+    exp = _pil_decode(blobs)
+    for g, e in zip(got, exp):
+        assert g.shape == e.shape
+        assert np.abs(g.astype(int) - e.astype(int)).mean() < 2.0
+
+
+# ---------------------------------------------------------------------------
+# nhwc -> nchw normalize
+# ---------------------------------------------------------------------------
+
+def test_nhwc_to_nchw_normalize(ext):
+    rng = np.random.RandomState(0)
+    x = torch.from_numpy(
+        rng.randint(0, 255, (4, 37, 53, 3)).astype(np.uint8)).cuda()
+    mean = torch.tensor([0.485, 0.456, 0.406], device='cuda')
+    std = torch.tensor([0.229, 0.224, 0.225], device='cuda')
+    out = torch.empty(4, 3, 37, 53, dtype=torch.float32, device='cuda')
+    ext.nhwc_to_nchw_normalize(x, out, mean, 1.0 / std, 1.0 / 255.0)
+    torch.cuda.synchronize()
+    ref = (x.permute(0, 3, 1, 2).float() / 255.0 -
+           mean.view(1, 3, 1, 1)) / std.view(1, 3, 1, 1)
+    torch.testing.assert_close(out, ref, rtol=1e-5, atol=1e-5)
+
+
+def test_nhwc_to_nchw_normalize_bf16(ext):
+    rng = np.random.RandomState(0)
+    x = torch.from_numpy(
+        rng.randint(0, 255, (2, 64, 64, 3)).astype(np.uint8)).cuda()
+    mean = torch.zeros(3, device='cuda')
+    inv_std = torch.ones(3, device='cuda')
+    out = torch.empty(2, 3, 64, 64, dtype=torch.bfloat16, device='cuda')
+    ext.nhwc_to_nchw_normalize(x, out, mean, inv_std, 1.0 / 255.0)
+    torch.cuda.synchronize()
+    ref = (x.permute(0, 3, 1, 2).float() / 255.0).bfloat16()
+    torch.testing.assert_close(out.float(), ref.float(), rtol=0.02,
+                               atol=0.01)
+
+
+# ---------------------------------------------------------------------------
+# GpuBatchReader end-to-end
+# ---------------------------------------------------------------------------
+
+def test_gpu_batch_reader_imagenet(ext, tmp_path):
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_imagenet_dataset
+    url = 'file://' + str(tmp_path / 'imnet')
+    create_imagenet_dataset(url, num_rows=64, rowgroup_size_mb=8)
+    with make_batch_reader(url, device='cuda',
+                           shuffle_row_groups=False) as r:
+        batches = list(r)
+    total = sum(b.image.shape[0] for b in batches)
+    assert total == 64
+    b0 = batches[0]
+    assert b0.image.is_cuda and b0.image.dtype == torch.uint8
+    assert b0.image.shape[1:] == (224, 224, 3)
+    assert b0.label.is_cuda
+    # decode correctness vs CPU reader
+    from petastorm_amd import make_reader
+    with make_reader(url, reader_pool_type='dummy',
+                     shuffle_row_groups=False) as cr:
+        cpu_rows = {int(row.label): row.image for row in cr}
+    labels = torch.cat([b.label for b in batches]).cpu().numpy()
+    images = torch.cat([b.image for b in batches]).cpu().numpy()
+    for i in range(0, 64, 16):
+        diff = np.abs(images[i].astype(int) -
+                      cpu_rows[int(labels[i])].astype(int))
+        assert diff.mean() < 2.0
+
+
+def test_gpu_batch_reader_scalar_sharded(ext, tmp_path):
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'scal')
+    create_scalar_dataset(url, num_rows=2000, rowgroup_size=250)
+    all_ids = []
+    for shard in range(2):
+        with make_batch_reader(url, device='cuda', shuffle_row_groups=False,
+                               cur_shard=shard, shard_count=2,
+                               schema_fields=['id', 'f0']) as r:
+            for b in r:
+                all_ids.extend(b.id.cpu().tolist())
+    assert sorted(all_ids) == list(range(2000))
+
+
+def test_gpu_batch_reader_hbm_cache_and_epochs(ext, tmp_path):
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_sequence_dataset
+    url = 'file://' + str(tmp_path / 'seqc')
+    create_sequence_dataset(url, num_rows=64, rowgroup_size_mb=0.5)
+    with make_batch_reader(url, device='cuda', shuffle_row_groups=False,
+                           num_epochs=3,
+                           gpu_options=dict(cache_type='hbm',
+                                            cache_size_limit=1 << 30)) as r:
+        batches = list(r)
+        diag = r.diagnostics
+    total = sum(b.tokens.shape[0] for b in batches)
+    assert total == 3 * 64
+    assert diag['hbm_cache_hits'] > 0
+
+
+def test_gpu_predicate(ext, tmp_path):
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.predicates import in_lambda
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'pred')
+    create_scalar_dataset(url, num_rows=1000, rowgroup_size=200)
+    pred = in_lambda(['id'], lambda v: v['id'] % 4 == 0)
+    with make_batch_reader(url, device='cuda', shuffle_row_groups=False,
+                           predicate=pred,
+                           schema_fields=['id', 'i1']) as r:
+        ids = torch.cat([b.id for b in r]).cpu().numpy()
+    assert (ids % 4 == 0).all() and len(ids) == 250

@@ -1,0 +1,96 @@
+"""Host-side pieces of the HIP extension (run on CPU-only machines):
+thrift page walker and the JPEG header parser."""
+import io
+
+import numpy as np
+import pytest
+import torch
+
+from petastorm_amd import ops
+
+pytestmark = pytest.mark.skipif(not ops.available(),
+                                reason='HIP extension not built')
+
+
+def _ext():
+    return ops.ext()
+
+
+def test_parquet_walk_pages_matches_pyarrow(scalar_dataset):
+    import pyarrow.parquet as pq
+    e = _ext()
+    import glob
+    f = sorted(glob.glob(scalar_dataset['path'] + '/*.parquet'))[0]
+    raw = open(f, 'rb').read()
+    host = torch.frombuffer(bytearray(raw), dtype=torch.uint8)
+    md = pq.ParquetFile(f).metadata
+    for rg in range(md.num_row_groups):
+        offs, lens, nvals = [], [], []
+        for ci in range(md.num_columns):
+            col = md.row_group(rg).column(ci)
+            start = col.data_page_offset
+            if col.dictionary_page_offset is not None:
+                start = min(start, col.dictionary_page_offset)
+            offs.append(start)
+            lens.append(col.total_compressed_size)
+            nvals.append(col.num_values)
+        pages = e.parquet_walk_pages(host, torch.tensor(offs),
+                                     torch.tensor(lens))
+        for ci in range(md.num_columns):
+            got = int(pages['num_values'][
+                (pages['page_chunk'] == ci) & (pages['page_type'] == 0)].sum())
+            assert got == nvals[ci]
+
+
+def _make_jpegs(n, size=(48, 64), quality=90, gray=False, subsampling=None):
+    from PIL import Image
+    rng = np.random.RandomState(0)
+    blobs, arrays = [], []
+    for i in range(n):
+        if gray:
+            arr = rng.randint(0, 255, size[::-1]).astype(np.uint8)
+        else:
+            arr = rng.randint(0, 255, size[::-1] + (3,)).astype(np.uint8)
+        img = Image.fromarray(arr)
+        b = io.BytesIO()
+        kw = dict(format='JPEG', quality=quality, restart_marker_rows=1)
+        if subsampling is not None:
+            kw['subsampling'] = subsampling
+        img.save(b, **kw)
+        blobs.append(b.getvalue())
+        arrays.append(arr)
+    return blobs, arrays
+
+
+def test_jpeg_parse_batch_geometry():
+    e = _ext()
+    blobs, _ = _make_jpegs(4)
+    buf = b''.join(blobs)
+    off, lens, pos = [], [], 0
+    for d in blobs:
+        off.append(pos)
+        lens.append(len(d))
+        pos += len(d)
+    t = torch.frombuffer(bytearray(buf), dtype=torch.uint8)
+    meta = e.jpeg_parse_batch(t, torch.tensor(off), torch.tensor(lens))
+    assert meta['width'].tolist() == [48] * 4
+    assert meta['height'].tolist() == [64] * 4
+    assert meta['ncomp'].tolist() == [3] * 4
+    # 4:2:0 -> mcu 16x16 -> 3x4 grid; one restart segment per MCU row
+    assert meta['mcus_x'].tolist() == [3] * 4
+    assert meta['mcus_y'].tolist() == [4] * 4
+    assert meta['seg_img'].numel() == 16
+    assert int(meta['seg_nmcu'].sum()) == 4 * 12
+
+
+def test_jpeg_parse_rejects_progressive():
+    from PIL import Image
+    e = _ext()
+    rng = np.random.RandomState(0)
+    img = Image.fromarray(rng.randint(0, 255, (32, 32, 3)).astype(np.uint8))
+    b = io.BytesIO()
+    img.save(b, format='JPEG', progressive=True)
+    d = b.getvalue()
+    t = torch.frombuffer(bytearray(d), dtype=torch.uint8)
+    with pytest.raises(RuntimeError, match='non-baseline'):
+        e.jpeg_parse_batch(t, torch.tensor([0]), torch.tensor([len(d)]))
