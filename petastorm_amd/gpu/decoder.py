@@ -61,6 +61,8 @@ class ByteArrayColumn(object):
         self.host_buf = host_buf        # uint8 cpu tensor or None
         self.host_val_off = host_val_off  # int64 cpu tensor or None
         self.n = n
+        self.jpeg_meta = None           # precomputed by prepare_host
+        self.png_meta = None
 
 
 class GpuRowGroupDecoder(object):
@@ -68,6 +70,10 @@ class GpuRowGroupDecoder(object):
         self.device = torch.device(device)
         self._ext = ops.ext()
         self.cpu_assist_columns = set()
+        # deferred status checks: each decode stage appends its status
+        # tensor; flush_status() does ONE host sync per row-group instead of
+        # one per kernel
+        self._pending_status = []
 
     # ------------------------------------------------------------------
     def read_rowgroup_bytes(self, path, file_metadata, parquet_schema, rg,
@@ -115,26 +121,70 @@ class GpuRowGroupDecoder(object):
         return host, chunk_meta
 
     # ------------------------------------------------------------------
-    def decode(self, host_buf, chunk_meta, schema, stream=None):
+    def prepare_host(self, host_buf, chunk_meta, schema):
+        """All host-only parse work for one row-group: page walk, byte-array
+        offset scans and jpeg/png header parsing.  Pure CPU — the IO
+        prefetch thread runs this so it overlaps GPU decode of the previous
+        row-group."""
+        ext = self._ext
+        plan = {}
+        for ch in chunk_meta['chunks']:
+            name = ch['name']
+            entry = {}
+            entry['pages'] = ext.parquet_walk_pages(
+                host_buf, torch.tensor([ch['offset']], dtype=torch.int64),
+                torch.tensor([ch['length']], dtype=torch.int64))
+            pages = entry['pages']
+            # host-visible PLAIN byte-array of a REQUIRED column: offsets and
+            # image headers can be parsed before any GPU work
+            if ch['compression'] == 'UNCOMPRESSED' and \
+                    ch['physical'] == 'BYTE_ARRAY' and ch['max_def'] == 0:
+                ptype = pages['page_type'].numpy()
+                enc = pages['encoding'].numpy()
+                didx = [i for i in range(len(ptype)) if ptype[i] == 0]
+                if didx and all(enc[i] == _ENC_PLAIN for i in didx):
+                    starts = pages['data_off'].numpy()[didx].astype(np.int64)
+                    counts = pages['num_values'].numpy()[didx] \
+                        .astype(np.int64)
+                    ho = ext.byte_array_host_offsets(
+                        host_buf, torch.from_numpy(starts),
+                        torch.from_numpy(counts))
+                    entry['host_off'] = ho['off'].numpy()
+                    entry['host_len'] = ho['len'].numpy()
+                    field = schema.fields.get(name)
+                    codec = field.codec if field is not None else None
+                    try:
+                        if isinstance(codec, CompressedImageCodec):
+                            if codec.image_codec == 'jpeg':
+                                entry['jpeg_meta'] = ext.jpeg_parse_batch(
+                                    host_buf, ho['off'], ho['len'])
+                            else:
+                                entry['png_meta'] = ext.png_parse_batch(
+                                    host_buf, ho['off'], ho['len'])
+                    except RuntimeError:
+                        pass  # unsupported flavor -> device/CPU path decides
+            plan[name] = entry
+        return plan
+
+    def decode(self, host_buf, chunk_meta, schema, host_plan=None):
         """Decode the requested columns.  Returns dict name ->
         torch tensor (fixed columns) or ByteArrayColumn (binary)."""
         ext = self._ext
         dev = self.device
         n_rows = chunk_meta['num_rows']
+        if host_plan is None:
+            host_plan = self.prepare_host(host_buf, chunk_meta, schema)
         dbuf = host_buf.to(dev, non_blocking=True)
 
         out = {}
         for ch in chunk_meta['chunks']:
             name = ch['name']
-            pages = ext.parquet_walk_pages(
-                host_buf, torch.tensor([ch['offset']], dtype=torch.int64),
-                torch.tensor([ch['length']], dtype=torch.int64))
             comp = ch['compression']
             if comp not in ('UNCOMPRESSED', 'SNAPPY'):
                 out[name] = self._cpu_assist_marker(name)
                 continue
-            col = self._decode_chunk(ext, dev, dbuf, host_buf, ch, pages,
-                                     n_rows, schema)
+            col = self._decode_chunk(ext, dev, dbuf, host_buf, ch,
+                                     host_plan[name], n_rows, schema)
             out[name] = col
         return out, dbuf
 
@@ -143,8 +193,10 @@ class GpuRowGroupDecoder(object):
         return None
 
     # ------------------------------------------------------------------
-    def _decode_chunk(self, ext, dev, dbuf, host_buf, ch, pages, n_rows,
+    def _decode_chunk(self, ext, dev, dbuf, host_buf, ch, plan_entry, n_rows,
                       schema):
+        pages = plan_entry['pages']
+        self._plan_entry = plan_entry
         page_type = pages['page_type'].numpy()
         data_off = pages['data_off'].numpy()       # relative to chunk walk
         comp_size = pages['comp_size'].numpy()
@@ -312,22 +364,23 @@ class GpuRowGroupDecoder(object):
             torch.from_numpy(o_off).to(dev), val_off, val_len, status)
         self._check(status, 'bytearray:' + ch['name'])
         host_off = None
+        plan = getattr(self, '_plan_entry', {}) or {}
         if host_visible:
-            # mirror the scan on host (cheap: one u32 read per value) so
-            # codecs needing header parsing (jpeg) can see the bytes
-            hb = host_buf.numpy()
-            host_off = np.empty(total, dtype=np.int64)
-            k = 0
-            for j in range(len(counts)):
-                pos = int(val_start[j])
-                for _ in range(int(counts[j])):
-                    ln = int.from_bytes(hb[pos:pos + 4].tobytes(), 'little')
-                    host_off[k] = pos + 4
-                    k += 1
-                    pos += 4 + ln
-        return ByteArrayColumn(page_buf, val_off, val_len,
-                               host_buf if host_visible else None,
-                               host_off, total)
+            if 'host_off' in plan:
+                host_off = plan['host_off']
+            else:
+                # mirror the scan on host (native C++) so codecs needing
+                # header parsing (jpeg/png) can see the bytes
+                ho = ext.byte_array_host_offsets(
+                    host_buf, torch.from_numpy(val_start),
+                    torch.from_numpy(counts.astype(np.int64)))
+                host_off = ho['off'].numpy()
+        col = ByteArrayColumn(page_buf, val_off, val_len,
+                              host_buf if host_visible else None,
+                              host_off, total)
+        col.jpeg_meta = plan.get('jpeg_meta')
+        col.png_meta = plan.get('png_meta')
+        return col
 
     def _dict_fixed(self, ext, dev, page_buf, page_start, dict_idx,
                     num_values, size_arr, val_start, val_end, page_nval,
@@ -376,10 +429,22 @@ class GpuRowGroupDecoder(object):
 
     # ------------------------------------------------------------------
     def _check(self, status, what):
-        s = int(status.abs().sum().item())
-        if s != 0:
-            raise RuntimeError('GPU decode error in {}: status={}'
-                               .format(what, status.cpu().tolist()))
+        """Queue a status tensor for the end-of-row-group flush."""
+        self._pending_status.append((what, status))
+
+    def flush_status(self):
+        """One sync: verify every queued kernel status is clean."""
+        if not self._pending_status:
+            return
+        pending, self._pending_status = self._pending_status, []
+        total = torch.stack([s.abs().sum() for _, s in pending]).sum()
+        if int(total.item()) != 0:
+            for what, s in pending:
+                vals = s.cpu()
+                if int(vals.abs().sum()) != 0:
+                    raise RuntimeError(
+                        'GPU decode error in {}: status={}'
+                        .format(what, vals.tolist()))
 
     # ------------------------------------------------------------------
     # codec stages over ByteArrayColumn
@@ -418,15 +483,118 @@ class GpuRowGroupDecoder(object):
             t = t.to(torch.int64) & 0xFFFFFFFF
         return t
 
+    def decode_compressed_ndarray_column(self, col, field):
+        """CompressedNdarrayCodec: zlib(npy) -> inflate kernel -> dense
+        [n, *shape] tensor (reference petastorm/codecs.py:174-212)."""
+        ext = self._ext
+        dev = self.device
+        np_dtype = np.dtype(field.numpy_dtype)
+        shape = tuple(field.shape)
+        if any(d is None for d in shape):
+            return None
+        elem = int(np.prod(shape)) if shape else 1
+        row_bytes = elem * np_dtype.itemsize
+        cap = row_bytes + 256  # npy header upper bound
+        n = col.n
+        # each value is a single zlib stream: segment table == value table
+        seg_first = torch.arange(n, dtype=torch.int32, device=dev)
+        seg_count = torch.ones(n, dtype=torch.int32, device=dev)
+        raw = torch.empty(n * cap + _SLACK, dtype=torch.uint8, device=dev)
+        raw_off = torch.arange(n, dtype=torch.int64, device=dev) * cap
+        raw_cap = torch.full((n,), cap, dtype=torch.int64, device=dev)
+        produced = torch.zeros(n, dtype=torch.int64, device=dev)
+        status = torch.zeros(n, dtype=torch.int32, device=dev)
+        ext.inflate_batch(col.device_buf, col.val_off,
+                          col.val_len.to(torch.int64), seg_first, seg_count,
+                          raw, raw_off, raw_cap, produced, 0, status)
+        self._check(status, 'inflate:' + field.name)
+        pay_off = torch.empty(n, dtype=torch.int64, device=dev)
+        pay_len = torch.empty(n, dtype=torch.int64, device=dev)
+        st2 = torch.zeros(1, dtype=torch.int32, device=dev)
+        ext.npy_payload_offsets(raw, raw_off, produced.to(torch.int32),
+                                pay_off, pay_len, st2)
+        self._check(st2, 'npy:' + field.name)
+        out = torch.empty(n * row_bytes + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        dst_off = torch.arange(n, dtype=torch.int64, device=dev) * row_bytes
+        ext.varlen_gather(raw, pay_off, pay_len, out, dst_off)
+        view_dtype = _np_view_torch(np_dtype)
+        if view_dtype is None:
+            return None
+        t = out[:n * row_bytes].view(view_dtype).view((n,) + shape)
+        return _widen_unsigned(t, np_dtype)
+
+    def decode_png_column(self, col, field):
+        """CompressedImageCodec(png): inflate + unfilter kernels
+        (reference cv2.imdecode, petastorm/codecs.py:106)."""
+        if col.host_buf is None:
+            return None
+        ext = self._ext
+        dev = self.device
+        meta = col.png_meta
+        if meta is None:
+            meta = ext.png_parse_batch(col.host_buf,
+                                       torch.from_numpy(col.host_val_off),
+                                       col.val_len.cpu())
+        n = col.n
+        heights = meta['height'].numpy()
+        widths = meta['width'].numpy()
+        channels = meta['channels'].numpy()
+        depth = meta['bit_depth'].numpy()
+        row_bytes = meta['row_bytes'].numpy().astype(np.int64)
+        raw_size = meta['raw_size'].numpy()
+        raw_off = np.zeros(n, dtype=np.int64)
+        raw_off[1:] = np.cumsum(raw_size)[:-1]
+        raw = torch.empty(int(raw_size.sum()) + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        produced = torch.zeros(n, dtype=torch.int64, device=dev)
+        status = torch.zeros(n, dtype=torch.int32, device=dev)
+        ext.inflate_batch(col.device_buf, meta['seg_off'].to(dev),
+                          meta['seg_len'].to(dev),
+                          meta['seg_first'].to(dev),
+                          meta['seg_count'].to(dev), raw,
+                          torch.from_numpy(raw_off).to(dev),
+                          torch.from_numpy(raw_size).to(dev), produced, 0,
+                          status)
+        self._check(status, 'png-inflate:' + field.name)
+        out_bytes = row_bytes * heights
+        out_off = np.zeros(n, dtype=np.int64)
+        out_off[1:] = np.cumsum(out_bytes)[:-1]
+        out = torch.empty(int(out_bytes.sum()) + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        st2 = torch.zeros(n, dtype=torch.int32, device=dev)
+        ext.png_unfilter_batch(raw, torch.from_numpy(raw_off).to(dev), out,
+                               torch.from_numpy(out_off).to(dev),
+                               meta['height'].to(dev),
+                               meta['row_bytes'].to(dev),
+                               meta['bpp'].to(dev), st2)
+        self._check(st2, 'png-unfilter:' + field.name)
+        if len(set(widths.tolist())) != 1 or len(set(heights.tolist())) != 1 \
+                or len(set(channels.tolist())) != 1 \
+                or len(set(depth.tolist())) != 1:
+            return None  # ragged batch -> CPU assist
+        h, w, c, d = int(heights[0]), int(widths[0]), int(channels[0]), \
+            int(depth[0])
+        out = out[:int(out_bytes.sum())]
+        if d == 16:
+            ext.bswap16(out)
+            t = out.view(torch.int16).view(n, h, w, c)
+            t = t.to(torch.int32) & 0xFFFF  # uint16 semantics
+        else:
+            t = out.view(n, h, w, c)
+        return t.squeeze(-1) if c == 1 else t
+
     def decode_jpeg_column(self, col, field):
         """CompressedImageCodec(jpeg): restart-parallel GPU decode."""
         if col.host_buf is None:
             return None  # compressed storage: host can't parse headers
         ext = self._ext
         dev = self.device
-        host_off = torch.from_numpy(col.host_val_off)
-        host_len = col.val_len.cpu()
-        meta = ext.jpeg_parse_batch(col.host_buf, host_off, host_len)
+        meta = col.jpeg_meta
+        if meta is None:
+            host_off = torch.from_numpy(col.host_val_off)
+            host_len = col.val_len.cpu()
+            meta = ext.jpeg_parse_batch(col.host_buf, host_off, host_len)
         n = col.n
         widths = meta['width'].numpy()
         heights = meta['height'].numpy()
@@ -457,6 +625,14 @@ class GpuRowGroupDecoder(object):
             t = out.view(n, int(heights[0]), int(widths[0]), c)
             return t if c == 3 else t.squeeze(-1)
         return None
+
+
+def _widen_unsigned(t, np_dtype):
+    if np_dtype == np.dtype(np.uint16):
+        return t.to(torch.int32) & 0xFFFF
+    if np_dtype == np.dtype(np.uint32):
+        return t.to(torch.int64) & 0xFFFFFFFF
+    return t
 
 
 def _np_view_torch(np_dtype):

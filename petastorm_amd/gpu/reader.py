@@ -150,7 +150,12 @@ class GpuBatchReader(object):
                 host, meta = self._decoder.read_rowgroup_bytes(
                     piece.path, md, pschema, piece.row_group, columns,
                     self._pin_pool)
-                out_q.put(('data', piece, host, meta))
+                # host-only parse work (page walk, offset scans, image
+                # headers) runs HERE so it overlaps GPU decode of the
+                # previous row-group
+                plan = self._decoder.prepare_host(host, meta,
+                                                  self._storage_schema)
+                out_q.put(('data', piece, host, (meta, plan)))
             out_q.put(('end', None, None, None))
         except Exception as e:  # noqa: BLE001 - forwarded to consumer
             out_q.put(('error', e, None, None))
@@ -179,12 +184,14 @@ class GpuBatchReader(object):
                     columns = self._cache._store[self._cache_key(piece)]
                     self._cache.get(self._cache_key(piece), lambda: columns)
                 else:
+                    meta, plan = meta
                     if self._cache is not None:
                         columns = self._cache.get(
                             self._cache_key(piece),
-                            lambda: self._decode_piece(piece, host, meta))
+                            lambda: self._decode_piece(piece, host, meta,
+                                                       plan))
                     else:
-                        columns = self._decode_piece(piece, host, meta)
+                        columns = self._decode_piece(piece, host, meta, plan)
                     self._pin_pool.put(host) if host is not None else None
                 batch = self._postprocess(piece, columns)
                 if batch is None:
@@ -196,8 +203,9 @@ class GpuBatchReader(object):
             epoch += 1
 
     # ------------------------------------------------------------------
-    def _decode_piece(self, piece, host, meta):
-        raw, dbuf = self._decoder.decode(host, meta, self._storage_schema)
+    def _decode_piece(self, piece, host, meta, plan=None):
+        raw, dbuf = self._decoder.decode(host, meta, self._storage_schema,
+                                         plan)
         columns = {}
         assist = []
         for name, col in raw.items():
@@ -211,6 +219,11 @@ class GpuBatchReader(object):
                 if isinstance(codec, CompressedImageCodec) and \
                         codec.image_codec == 'jpeg':
                     decoded = self._decoder.decode_jpeg_column(col, field)
+                elif isinstance(codec, CompressedImageCodec):
+                    decoded = self._decoder.decode_png_column(col, field)
+                elif isinstance(codec, CompressedNdarrayCodec):
+                    decoded = self._decoder.decode_compressed_ndarray_column(
+                        col, field)
                 elif isinstance(codec, NdarrayCodec) or (
                         codec is None and field is not None and
                         field.shape not in ((), None)):
@@ -223,6 +236,8 @@ class GpuBatchReader(object):
                 columns[name] = col
         if assist:
             columns.update(self._cpu_assist(piece, assist))
+        # one host sync verifying every decode kernel's status
+        self._decoder.flush_status()
         return columns
 
     def _cpu_assist(self, piece, names):

@@ -38,6 +38,23 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
 // thrift_pages.cpp
 py::dict parquet_walk_pages(torch::Tensor buf, torch::Tensor chunk_off,
                             torch::Tensor chunk_len);
+// inflate.hip
+void inflate_batch(torch::Tensor src, torch::Tensor seg_off,
+                   torch::Tensor seg_len, torch::Tensor seg_first,
+                   torch::Tensor seg_count, torch::Tensor dst,
+                   torch::Tensor dst_off, torch::Tensor dst_cap,
+                   torch::Tensor produced, int64_t mode,
+                   torch::Tensor status);
+void png_unfilter_batch(torch::Tensor raw, torch::Tensor raw_off,
+                        torch::Tensor out, torch::Tensor out_off,
+                        torch::Tensor height, torch::Tensor row_bytes,
+                        torch::Tensor bpp, torch::Tensor status);
+void bswap16(torch::Tensor data);
+// png_host.cpp
+py::dict png_parse_batch(torch::Tensor buf, torch::Tensor val_off,
+                         torch::Tensor val_len);
+py::dict byte_array_host_offsets(torch::Tensor buf, torch::Tensor val_start,
+                                 torch::Tensor counts);
 // jpeg.hip
 void jpeg_decode_batch(torch::Tensor data, py::dict meta, torch::Tensor coef,
                        torch::Tensor samples, torch::Tensor out,
@@ -61,6 +78,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused uint8 NHWC -> float NCHW normalize (LDS-tiled)");
   m.def("parquet_walk_pages", &psa::parquet_walk_pages,
         "Walk Parquet page headers (thrift compact) in a raw column chunk");
+  m.def("inflate_batch", &psa::inflate_batch,
+        "Batched DEFLATE inflate (zlib/raw), thread-per-stream");
+  m.def("png_unfilter_batch", &psa::png_unfilter_batch,
+        "PNG scanline unfilter (wave-per-image, bpp-lane chains)");
+  m.def("bswap16", &psa::bswap16, "byte-swap 16-bit samples in place");
+  m.def("png_parse_batch", &psa::png_parse_batch,
+        "Host-side PNG container parse");
+  m.def("byte_array_host_offsets", &psa::byte_array_host_offsets,
+        "Host-side PLAIN byte-array offset scan");
   m.def("jpeg_parse_batch", &psa::jpeg_parse_batch,
         "Host-side JPEG header/segment parse");
   m.def("jpeg_decode_batch", &psa::jpeg_decode_batch,

@@ -17,6 +17,7 @@ SRC = [
     'csrc/ext.cpp',
     'csrc/jpeg_host.cpp',
     'csrc/thrift_pages.cpp',
+    'csrc/png_host.cpp',
     'csrc/snappy.hip',
     'csrc/parquet_decode.hip',
     'csrc/jpeg.hip',

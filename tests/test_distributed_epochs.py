@@ -1,0 +1,104 @@
+"""Multi-process (gloo, world_size=2) coverage of the RCCL epoch-sync path.
+
+The driver runs these on CPU; on the GPU node the same code paths run over
+RCCL (backend 'nccl' on ROCm).
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _worker_perm(rank, world, file_store, out_q):
+    import torch.distributed as dist
+    dist.init_process_group('gloo', init_method='file://' + file_store,
+                            rank=rank, world_size=world)
+    from petastorm_amd.parallel.epochs import (epoch_end_sync,
+                                               epoch_permutation,
+                                               shard_for_rank)
+    perms = [epoch_permutation(20, e, seed=None).tolist() for e in range(3)]
+    counts = epoch_end_sync(100 + rank)
+    shard = shard_for_rank()
+    out_q.put((rank, perms, counts, shard))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_epoch_permutation_broadcast_consistency(tmp_path):
+    """Unseeded permutations must still agree across ranks (rank-0
+    broadcast replaces the reference's same-seed convention)."""
+    ctx = mp.get_context('spawn')
+    q = ctx.Queue()
+    store = str(tmp_path / 'store')
+    procs = [ctx.Process(target=_worker_perm, args=(r, 2, store, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, perms, counts, shard = q.get(timeout=90)
+        results[rank] = (perms, counts, shard)
+    for p in procs:
+        p.join(timeout=30)
+    assert results[0][0] == results[1][0], 'permutations diverged'
+    # epoch-end all-gather saw both ranks' counts
+    assert sorted(results[0][1]) == [100, 101]
+    assert sorted(results[1][1]) == [100, 101]
+    # shard defaults to (rank, world)
+    assert results[0][2] == (0, 2) and results[1][2] == (1, 2)
+
+
+def _worker_sharded_read(rank, world, file_store, url, out_q):
+    import torch.distributed as dist
+    dist.init_process_group('gloo', init_method='file://' + file_store,
+                            rank=rank, world_size=world)
+    from petastorm_amd import make_batch_reader
+    ids = []
+    with make_batch_reader(url, reader_pool_type='dummy',
+                           shuffle_row_groups=True, seed=17, num_epochs=2,
+                           cur_shard=rank, shard_count=world) as r:
+        for b in r:
+            ids.extend(int(x) for x in b.id)
+    out_q.put((rank, ids))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_sharded_readers_cover_dataset_across_processes(tmp_path,
+                                                        scalar_dataset):
+    ctx = mp.get_context('spawn')
+    q = ctx.Queue()
+    store = str(tmp_path / 'store2')
+    procs = [ctx.Process(target=_worker_sharded_read,
+                         args=(r, 2, store, scalar_dataset['url'], q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, ids = q.get(timeout=150)
+        results[rank] = ids
+    for p in procs:
+        p.join(timeout=30)
+    union = sorted(results[0] + results[1])
+    # 2 epochs x 500 rows, disjoint shards per epoch
+    assert union == sorted(list(range(500)) * 2)
+    assert not (set(results[0]) & set(results[1]))
+
+
+def test_epoch_permutation_single_process():
+    from petastorm_amd.parallel.epochs import epoch_permutation
+    a = epoch_permutation(50, 0, seed=3)
+    b = epoch_permutation(50, 0, seed=3)
+    c = epoch_permutation(50, 1, seed=3)
+    np.testing.assert_array_equal(a, b)
+    assert a.tolist() != c.tolist()
+    noshuffle = epoch_permutation(10, 0, seed=3, shuffle=False)
+    np.testing.assert_array_equal(noshuffle, np.arange(10))
+
+
+def test_epoch_end_sync_single_process():
+    from petastorm_amd.parallel.epochs import epoch_end_sync
+    assert epoch_end_sync(42) == [42]
