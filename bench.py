@@ -119,6 +119,9 @@ def bench_imagenet(args, rank, world, device, dist):
     if rank == 0 and reader.diagnostics.get('cpu_assist_columns'):
         print('WARNING: cpu-assist columns: {}'.format(
             reader.diagnostics['cpu_assist_columns']), file=sys.stderr)
+    if os.environ.get('PSA_TIMING') == '1' and rank == 0:
+        print('stage_times:', reader.diagnostics.get('stage_times'),
+              file=sys.stderr)
     return result, {
         'model': 'ImageNetSchema(224x224x3 jpeg CompressedImageCodec + '
                  'int32 label)',

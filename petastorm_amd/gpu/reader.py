@@ -19,8 +19,10 @@ boundaries all-gather per-rank row counts (petastorm_amd.parallel.epochs).
 """
 
 import logging
+import os
 import queue
 import threading
+import time
 
 import numpy as np
 import torch
@@ -116,6 +118,11 @@ class GpuBatchReader(object):
         self._file_md = {}
         self._stopped = False
         self._rows_epoch = 0
+        # stage timing (enabled with PSA_TIMING=1; reported in diagnostics)
+        self._timing_enabled = os.environ.get('PSA_TIMING') == '1'
+        self.stage_times = {'io_wait': 0.0, 'decode': 0.0, 'codec': 0.0,
+                            'postprocess': 0.0, 'io_read': 0.0,
+                            'io_parse': 0.0}
         self._gen = self._generate()
 
     # ------------------------------------------------------------------
@@ -147,14 +154,18 @@ class GpuBatchReader(object):
                     out_q.put(('cached', piece, None, None))
                     continue
                 md, pschema = self._metadata(piece.path)
+                t0 = time.perf_counter()
                 host, meta = self._decoder.read_rowgroup_bytes(
                     piece.path, md, pschema, piece.row_group, columns,
                     self._pin_pool)
+                t1 = time.perf_counter()
                 # host-only parse work (page walk, offset scans, image
                 # headers) runs HERE so it overlaps GPU decode of the
                 # previous row-group
                 plan = self._decoder.prepare_host(host, meta,
                                                   self._storage_schema)
+                self.stage_times['io_read'] += t1 - t0
+                self.stage_times['io_parse'] += time.perf_counter() - t1
                 out_q.put(('data', piece, host, (meta, plan)))
             out_q.put(('end', None, None, None))
         except Exception as e:  # noqa: BLE001 - forwarded to consumer
@@ -175,7 +186,9 @@ class GpuBatchReader(object):
             t.start()
             self._rows_epoch = 0
             while True:
+                t0 = time.perf_counter()
                 kind, piece, host, meta = q.get()
+                self.stage_times['io_wait'] += time.perf_counter() - t0
                 if kind == 'end':
                     break
                 if kind == 'error':
@@ -193,7 +206,9 @@ class GpuBatchReader(object):
                     else:
                         columns = self._decode_piece(piece, host, meta, plan)
                     self._pin_pool.put(host) if host is not None else None
+                t2 = time.perf_counter()
                 batch = self._postprocess(piece, columns)
+                self.stage_times['postprocess'] += time.perf_counter() - t2
                 if batch is None:
                     continue
                 self._rows_epoch += len(next(iter(batch.values())))
@@ -204,8 +219,11 @@ class GpuBatchReader(object):
 
     # ------------------------------------------------------------------
     def _decode_piece(self, piece, host, meta, plan=None):
+        t0 = time.perf_counter()
         raw, dbuf = self._decoder.decode(host, meta, self._storage_schema,
                                          plan)
+        self.stage_times['decode'] += time.perf_counter() - t0
+        t0 = time.perf_counter()
         columns = {}
         assist = []
         for name, col in raw.items():
@@ -238,6 +256,7 @@ class GpuBatchReader(object):
             columns.update(self._cpu_assist(piece, assist))
         # one host sync verifying every decode kernel's status
         self._decoder.flush_status()
+        self.stage_times['codec'] += time.perf_counter() - t0
         return columns
 
     def _cpu_assist(self, piece, names):
@@ -335,7 +354,8 @@ class GpuBatchReader(object):
 
     @property
     def diagnostics(self):
-        d = {'cpu_assist_columns': sorted(self._decoder.cpu_assist_columns)}
+        d = {'cpu_assist_columns': sorted(self._decoder.cpu_assist_columns),
+             'stage_times': dict(self.stage_times)}
         if self._cache is not None:
             d.update(hbm_cache_hits=self._cache.hits,
                      hbm_cache_misses=self._cache.misses,
