@@ -113,6 +113,7 @@ class GpuBatchReader(object):
 
         self._cache = HbmCache(cache_size_limit) \
             if cache_type == 'hbm' and cache_size_limit else None
+        self._inflight_hosts = []
 
         # per-file metadata handles (footer parse once per file)
         self._file_md = {}
@@ -175,6 +176,16 @@ class GpuBatchReader(object):
     def _cache_key(piece):
         return '{}:{}'.format(piece.path, piece.row_group)
 
+    def _reclaim_hosts(self):
+        """Return pinned buffers whose H2D copies have completed."""
+        still = []
+        for ev, buf in self._inflight_hosts:
+            if ev.query():
+                self._pin_pool.put(buf)
+            else:
+                still.append((ev, buf))
+        self._inflight_hosts = still
+
     # ------------------------------------------------------------------
     def _generate(self):
         epoch = 0
@@ -185,6 +196,11 @@ class GpuBatchReader(object):
                                  daemon=True)
             t.start()
             self._rows_epoch = 0
+            # one-deep software pipeline: decode of row-group N+1 is LAUNCHED
+            # before row-group N's batch is yielded, so N+1's H2D + kernels
+            # overlap the consumer's work on N; each row-group's status sync
+            # happens just before ITS batch is yielded
+            pending = None
             while True:
                 t0 = time.perf_counter()
                 kind, piece, host, meta = q.get()
@@ -205,14 +221,27 @@ class GpuBatchReader(object):
                                                        plan))
                     else:
                         columns = self._decode_piece(piece, host, meta, plan)
-                    self._pin_pool.put(host) if host is not None else None
+                    if host is not None:
+                        # the async H2D of `host` may still be in flight;
+                        # reclaim the pinned buffer only after an event
+                        # recorded behind it completes
+                        ev = torch.cuda.Event()
+                        ev.record()
+                        self._inflight_hosts.append((ev, host))
+                self._reclaim_hosts()
                 t2 = time.perf_counter()
                 batch = self._postprocess(piece, columns)
                 self.stage_times['postprocess'] += time.perf_counter() - t2
+                if pending is not None:
+                    yield pending
+                    pending = None
                 if batch is None:
                     continue
+                self._decoder.flush_status()
                 self._rows_epoch += len(next(iter(batch.values())))
-                yield self.schema.make_namedtuple(**batch)
+                pending = self.schema.make_namedtuple(**batch)
+            if pending is not None:
+                yield pending
             t.join()
             epoch_sync.epoch_end_sync(self._rows_epoch)
             epoch += 1
@@ -254,8 +283,9 @@ class GpuBatchReader(object):
                 columns[name] = col
         if assist:
             columns.update(self._cpu_assist(piece, assist))
-        # one host sync verifying every decode kernel's status
-        self._decoder.flush_status()
+        # NB: flush_status is called by the pipeline loop just before this
+        # row-group's batch is yielded, so the sync overlaps decode of the
+        # next row-group
         self.stage_times['codec'] += time.perf_counter() - t0
         return columns
 

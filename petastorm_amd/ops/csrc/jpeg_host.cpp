@@ -17,6 +17,7 @@
 #include <torch/extension.h>
 #include <cstdint>
 #include <cstring>
+#include <map>
 #include <vector>
 
 namespace psa {
@@ -106,9 +107,14 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
 
   std::vector<ImgInfo> imgs(n);
   // batch-level table pools; per image we remember which pool slot each of
-  // its table ids (0..3) resolved to at SOS time
+  // its table ids (0..3) resolved to at SOS time.  Tables are de-duplicated
+  // by content: a batch of images from one encoder collapses to a handful
+  // of pool entries instead of thousands (this parse runs per row-group on
+  // the hot path).
   std::vector<std::array<uint16_t, 64>> qpool;
   std::vector<HuffTable> hpool;
+  std::map<std::array<uint16_t, 64>, int> qdedup;
+  std::map<std::vector<uint8_t>, int> hdedup;
 
   std::vector<int32_t> seg_img;
   std::vector<int64_t> seg_pos, seg_end;
@@ -144,8 +150,12 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
                         "unsupported");
             std::array<uint16_t, 64> tab;
             for (int k2 = 0; k2 < 64; ++k2) tab[k2] = s[q + 1 + k2];
-            qpool.push_back(tab);
-            cur_q[id] = (int)qpool.size() - 1;
+            auto it = qdedup.find(tab);
+            if (it == qdedup.end()) {
+              qpool.push_back(tab);
+              it = qdedup.emplace(tab, (int)qpool.size() - 1).first;
+            }
+            cur_q[id] = it->second;
             q += 65;
           }
           break;
@@ -158,13 +168,19 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
             int nvals = 0;
             for (int l = 0; l < 16; ++l) nvals += bits[l];
             TORCH_CHECK(nvals <= 256, "image ", i, ": bad DHT");
-            HuffTable t{};
-            build_huff(bits, s + q + 17, nvals, t);
-            hpool.push_back(t);
+            std::vector<uint8_t> key(s + q + 1, s + q + 17 + nvals);
+            auto it = hdedup.find(key);
+            if (it == hdedup.end()) {
+              HuffTable t{};
+              build_huff(bits, s + q + 17, nvals, t);
+              hpool.push_back(t);
+              it = hdedup.emplace(std::move(key),
+                                  (int)hpool.size() - 1).first;
+            }
             if (cls == 0)
-              cur_dc[id] = (int)hpool.size() - 1;
+              cur_dc[id] = it->second;
             else
-              cur_ac[id] = (int)hpool.size() - 1;
+              cur_ac[id] = it->second;
             q += 17 + nvals;
           }
           break;
@@ -293,19 +309,34 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
   torch::Tensor ncomp = torch::empty({nimg}, i32);
   torch::Tensor mcus_x_t = torch::empty({nimg}, i32);
   torch::Tensor mcus_y_t = torch::empty({nimg}, i32);
-  torch::Tensor comp_h = torch::empty({nimg, 3}, i32);
-  torch::Tensor comp_v = torch::empty({nimg, 3}, i32);
-  torch::Tensor comp_q = torch::empty({nimg, 3}, i32);
-  torch::Tensor comp_dc = torch::empty({nimg, 3}, i32);
-  torch::Tensor comp_ac = torch::empty({nimg, 3}, i32);
-  torch::Tensor samp_off = torch::empty({nimg, 3}, i64);
-  torch::Tensor samp_stride = torch::empty({nimg, 3}, i32);
+  torch::Tensor comp_h = torch::zeros({nimg, 3}, i32);
+  torch::Tensor comp_v = torch::zeros({nimg, 3}, i32);
+  torch::Tensor comp_q = torch::zeros({nimg, 3}, i32);
+  torch::Tensor comp_dc = torch::zeros({nimg, 3}, i32);
+  torch::Tensor comp_ac = torch::zeros({nimg, 3}, i32);
+  torch::Tensor samp_off = torch::zeros({nimg, 3}, i64);
+  torch::Tensor samp_stride = torch::zeros({nimg, 3}, i32);
   torch::Tensor img_block0 = torch::empty({nimg + 1}, i64);
   torch::Tensor bpm_t = torch::empty({nimg}, i32);
   torch::Tensor kmap = torch::zeros({nimg, 8}, i32);
 
   int64_t samp_total = 0;
   int64_t block_total = 0;
+  int32_t* W = width.data_ptr<int32_t>();
+  int32_t* H = height.data_ptr<int32_t>();
+  int32_t* NC = ncomp.data_ptr<int32_t>();
+  int32_t* MX = mcus_x_t.data_ptr<int32_t>();
+  int32_t* MY = mcus_y_t.data_ptr<int32_t>();
+  int32_t* CH = comp_h.data_ptr<int32_t>();
+  int32_t* CV = comp_v.data_ptr<int32_t>();
+  int32_t* CQ = comp_q.data_ptr<int32_t>();
+  int32_t* CDC = comp_dc.data_ptr<int32_t>();
+  int32_t* CAC = comp_ac.data_ptr<int32_t>();
+  int64_t* SOFF = samp_off.data_ptr<int64_t>();
+  int32_t* SSTR = samp_stride.data_ptr<int32_t>();
+  int64_t* IB0 = img_block0.data_ptr<int64_t>();
+  int32_t* BPM = bpm_t.data_ptr<int32_t>();
+  int32_t* KMAP = kmap.data_ptr<int32_t>();
   for (int64_t i = 0; i < n; ++i) {
     ImgInfo& im = imgs[i];
     int hmax = 1, vmax = 1;
@@ -315,38 +346,39 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
     }
     int mx = (im.w + 8 * hmax - 1) / (8 * hmax);
     int my = (im.h + 8 * vmax - 1) / (8 * vmax);
-    width[i] = im.w;
-    height[i] = im.h;
-    ncomp[i] = im.ncomp;
-    mcus_x_t[i] = mx;
-    mcus_y_t[i] = my;
-    img_block0[i] = block_total;
+    W[i] = im.w;
+    H[i] = im.h;
+    NC[i] = im.ncomp;
+    MX[i] = mx;
+    MY[i] = my;
+    IB0[i] = block_total;
     int bpm = 0;
     for (int c = 0; c < im.ncomp; ++c) {
-      comp_h[i][c] = im.comp_h[c];
-      comp_v[i][c] = im.comp_v[c];
-      comp_q[i][c] = im.comp_q[c];
-      comp_dc[i][c] = im.comp_dc[c];
-      comp_ac[i][c] = im.comp_ac[c];
+      CH[i * 3 + c] = im.comp_h[c];
+      CV[i * 3 + c] = im.comp_v[c];
+      CQ[i * 3 + c] = im.comp_q[c];
+      CDC[i * 3 + c] = im.comp_dc[c];
+      CAC[i * 3 + c] = im.comp_ac[c];
       int pw = mx * im.comp_h[c] * 8;   // padded plane width in samples
       int ph = my * im.comp_v[c] * 8;
-      samp_off[i][c] = samp_total;
-      samp_stride[i][c] = pw;
+      SOFF[i * 3 + c] = samp_total;
+      SSTR[i * 3 + c] = pw;
       samp_total += (int64_t)pw * ph;
       for (int v = 0; v < im.comp_v[c]; ++v)
         for (int hh = 0; hh < im.comp_h[c]; ++hh)
-          kmap[i][bpm++] = (c << 8) | (v << 4) | hh;
+          KMAP[i * 8 + bpm++] = (c << 8) | (v << 4) | hh;
     }
-    bpm_t[i] = bpm;
+    BPM[i] = bpm;
     block_total += (int64_t)mx * my * bpm;
   }
-  img_block0[nimg] = block_total;
+  IB0[nimg] = block_total;
 
   int64_t nq = (int64_t)qpool.size();
   torch::Tensor qtabs = torch::empty({std::max<int64_t>(nq, 1), 64}, f32);
+  float* QT = qtabs.data_ptr<float>();
   for (int64_t t = 0; t < nq; ++t)
     for (int k2 = 0; k2 < 64; ++k2)
-      qtabs[t][k2] = (float)qpool[t][k2];
+      QT[t * 64 + k2] = (float)qpool[t][k2];
 
   int64_t nh = (int64_t)hpool.size();
   torch::Tensor lut = torch::empty({std::max<int64_t>(nh, 1), 256}, i32);
