@@ -80,10 +80,12 @@ def bench_imagenet(args, rank, world, device, dist):
     from petastorm_amd.test_util.dataset_gen import create_imagenet_dataset
     from petastorm_amd.unischema import UnischemaField
 
-    n_rows = args.rows or 2048
+    # >= 8 row-groups are required so every rank of an 8-GPU run gets data
+    # (sharding is per row-group, reference reader.py:573-597)
+    n_rows = args.rows or 6144
     url = _dataset_dir('imagenet_{}'.format(n_rows), rank, dist,
                        lambda u: create_imagenet_dataset(
-                           u, num_rows=n_rows, rowgroup_size_mb=32))
+                           u, num_rows=n_rows, rowgroup_size_mb=16))
 
     ext = ops.ext()
     mean = torch.tensor([0.485, 0.456, 0.406], device=device)
@@ -174,10 +176,10 @@ def bench_helloworld_cpu(args, rank, world, device, dist):
     from petastorm_amd import make_reader
     from petastorm_amd.test_util.dataset_gen import create_hello_world_dataset
 
-    n_rows = args.rows or 200
+    n_rows = args.rows or 400
     url = _dataset_dir('hello_{}'.format(n_rows), rank, dist,
                        lambda u: create_hello_world_dataset(
-                           u, num_rows=n_rows, rowgroup_size_mb=16))
+                           u, num_rows=n_rows, rowgroup_size_mb=4))
     reader = make_reader(url, reader_pool_type='thread', workers_count=3,
                          num_epochs=None, shuffle_row_groups=True)
     it = iter(reader)

@@ -352,19 +352,28 @@ class GpuRowGroupDecoder(object):
         counts = nonnull_per_page if nonnull_per_page is not None \
             else page_nval
         total = int(counts.sum())
-        o_off = np.zeros(len(counts), dtype=np.int64)
-        o_off[1:] = np.cumsum(counts)[:-1]
-        val_off = torch.empty(total, dtype=torch.int64, device=dev)
-        val_len = torch.empty(total, dtype=torch.int32, device=dev)
-        status = torch.zeros(len(counts), dtype=torch.int32, device=dev)
-        ext.byte_array_offsets_batch(
-            page_buf, torch.from_numpy(val_start).to(dev),
-            torch.from_numpy(val_end).to(dev),
-            torch.from_numpy(counts.astype(np.int32)).to(dev),
-            torch.from_numpy(o_off).to(dev), val_off, val_len, status)
-        self._check(status, 'bytearray:' + ch['name'])
-        host_off = None
         plan = getattr(self, '_plan_entry', {}) or {}
+        if host_visible and 'host_off' in plan:
+            # offsets were already scanned on host by prepare_host (in the
+            # IO thread) — upload them instead of launching the serial scan
+            # kernel (which is latency-bound at a handful of pages)
+            val_off = torch.from_numpy(plan['host_off']).to(
+                dev, non_blocking=True)
+            val_len = torch.from_numpy(
+                plan['host_len'].astype(np.int32)).to(dev, non_blocking=True)
+        else:
+            o_off = np.zeros(len(counts), dtype=np.int64)
+            o_off[1:] = np.cumsum(counts)[:-1]
+            val_off = torch.empty(total, dtype=torch.int64, device=dev)
+            val_len = torch.empty(total, dtype=torch.int32, device=dev)
+            status = torch.zeros(len(counts), dtype=torch.int32, device=dev)
+            ext.byte_array_offsets_batch(
+                page_buf, torch.from_numpy(val_start).to(dev),
+                torch.from_numpy(val_end).to(dev),
+                torch.from_numpy(counts.astype(np.int32)).to(dev),
+                torch.from_numpy(o_off).to(dev), val_off, val_len, status)
+            self._check(status, 'bytearray:' + ch['name'])
+        host_off = None
         if host_visible:
             if 'host_off' in plan:
                 host_off = plan['host_off']
