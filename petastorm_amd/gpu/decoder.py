@@ -244,6 +244,21 @@ class GpuRowGroupDecoder(object):
         max_def = ch['max_def']
         phys = ch['physical']
 
+        # fast path: PLAIN fixed-width columns decode def levels + values in
+        # ONE fused kernel with zero host syncs (common case: pyarrow marks
+        # every column OPTIONAL, so max_def==1 even for null-free data)
+        if data_enc == _ENC_PLAIN and phys in _PHYS_TO_TORCH:
+            sizes = uncomp_size if snappy else comp_size
+            p_start = np.array([page_start[i] for i in data_idx],
+                               dtype=np.int64)
+            p_end = np.array([page_start[i] + sizes[i] for i in data_idx],
+                             dtype=np.int64)
+            row0 = np.zeros(len(data_idx), dtype=np.int64)
+            row0[1:] = np.cumsum(page_nval)[:-1]
+            return self._plain_fixed_fused(ext, dev, page_buf, p_start,
+                                           p_end, page_nval, row0, max_def,
+                                           n_rows, phys, ch['name'])
+
         # 3) locate per-page def-level and value sections
         val_start = np.empty(len(data_idx), dtype=np.int64)
         val_end = np.empty(len(data_idx), dtype=np.int64)
@@ -304,10 +319,6 @@ class GpuRowGroupDecoder(object):
                                         dtype=np.int64)
 
         # 5) values by encoding
-        if data_enc == _ENC_PLAIN and phys in _PHYS_TO_TORCH:
-            return self._plain_fixed(ext, dev, page_buf, val_start, val_end,
-                                     page_nval, nonnull_per_page, valid,
-                                     n_rows, phys)
         if data_enc == _ENC_PLAIN and phys == 'BYTE_ARRAY':
             return self._plain_byte_array(
                 ext, dev, page_buf, host_buf, val_start, val_end, page_nval,
@@ -322,29 +333,29 @@ class GpuRowGroupDecoder(object):
         return self._cpu_assist_marker(ch['name'])
 
     # ------------------------------------------------------------------
-    def _plain_fixed(self, ext, dev, page_buf, val_start, val_end, page_nval,
-                     nonnull_per_page, valid, n_rows, phys):
+    _FILL_PATTERNS = {
+        'FLOAT': 0x7FC00000,            # fp32 quiet NaN
+        'DOUBLE': 0x7FF8000000000000,   # fp64 quiet NaN
+        'INT32': 0, 'INT64': 0,
+    }
+
+    def _plain_fixed_fused(self, ext, dev, page_buf, p_start, p_end,
+                           page_nval, row0, max_def, n_rows, phys, name):
         dtype, esize = _PHYS_TO_TORCH[phys]
-        counts = nonnull_per_page if nonnull_per_page is not None \
-            else page_nval
-        nbytes = counts * esize
-        dst_off = np.zeros(len(counts), dtype=np.int64)
-        dst_off[1:] = np.cumsum(nbytes)[:-1]
-        total_vals = int(counts.sum())
-        flat = torch.empty(total_vals * esize + _SLACK, dtype=torch.uint8,
-                           device=dev)
-        ext.varlen_gather(page_buf,
-                          torch.from_numpy(val_start).to(dev),
-                          torch.from_numpy(nbytes.astype(np.int64)).to(dev),
-                          flat, torch.from_numpy(dst_off).to(dev))
-        values = flat[:total_vals * esize].view(dtype)
-        if valid is None:
-            return values
-        out = torch.zeros(n_rows, dtype=dtype, device=dev)
-        if dtype.is_floating_point:
-            out.fill_(float('nan'))
-        out[valid] = values
-        return out
+        total = int(page_nval.sum())
+        out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        status = torch.zeros(len(p_start), dtype=torch.int32, device=dev)
+        ext.plain_fixed_decode_batch(
+            page_buf, torch.from_numpy(p_start).to(dev, non_blocking=True),
+            torch.from_numpy(p_end).to(dev, non_blocking=True),
+            torch.from_numpy(page_nval.astype(np.int32)).to(
+                dev, non_blocking=True),
+            torch.from_numpy(row0).to(dev, non_blocking=True),
+            1 if max_def > 0 else 0, esize, self._FILL_PATTERNS[phys],
+            out, torch.empty(0, dtype=torch.uint8, device=dev), status)
+        self._check(status, 'plainfixed:' + name)
+        return out[:total * esize].view(dtype)
 
     def _plain_byte_array(self, ext, dev, page_buf, host_buf, val_start,
                           val_end, page_nval, nonnull_per_page, valid,

@@ -25,6 +25,13 @@ void byte_array_offsets_batch(torch::Tensor data, torch::Tensor start,
 void varlen_gather(torch::Tensor src, torch::Tensor src_off,
                    torch::Tensor lengths, torch::Tensor dst,
                    torch::Tensor dst_off);
+void plain_fixed_decode_batch(torch::Tensor page_buf,
+                              torch::Tensor payload_start,
+                              torch::Tensor payload_end,
+                              torch::Tensor n_values, torch::Tensor row0,
+                              int64_t has_def, int64_t esize,
+                              int64_t fill_pattern, torch::Tensor out,
+                              torch::Tensor valid_out, torch::Tensor status);
 void npy_payload_offsets(torch::Tensor data, torch::Tensor val_off,
                          torch::Tensor val_len, torch::Tensor pay_off,
                          torch::Tensor pay_len, torch::Tensor status);
@@ -72,6 +79,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "PLAIN byte-array page -> per-value (offset, length)");
   m.def("varlen_gather", &psa::varlen_gather,
         "Unaligned variable-length byte gather (funnel-shift copy)");
+  m.def("plain_fixed_decode_batch", &psa::plain_fixed_decode_batch,
+        "Fused PLAIN data-page decode: def levels + prefix scan + scatter");
   m.def("npy_payload_offsets", &psa::npy_payload_offsets,
         ".npy container -> payload (offset, length)");
   m.def("nhwc_to_nchw_normalize", &psa::nhwc_to_nchw_normalize,

@@ -305,3 +305,193 @@ void npy_payload_offsets(torch::Tensor data, torch::Tensor val_off,
 }
 
 }  // namespace psa
+
+namespace psa {
+
+// ---------------------------------------------------------------------------
+// Fused PLAIN fixed-width data-page decode: definition levels + non-null
+// prefix scan + value scatter in ONE kernel, one wave per page.
+//
+// Replaces the multi-launch path (rle levels -> cumsum -> masked scatter ->
+// varlen gather) which cost one host sync per page on nullable columns
+// (pyarrow writes every top-level column as OPTIONAL, so this is the common
+// case even for null-free data).
+//
+// Page payload layout (v1 data page, PLAIN values):
+//   [u32 dl_len][def-level RLE hybrid, bw=1][values...]   when max_def == 1
+//   [values...]                                            when max_def == 0
+//
+// Wave algorithm for bit-packed runs: each chunk of 64 levels ballots its
+// valid bits; lane's value index = running val_cursor + popc(mask & lanes
+// below me).  RLE runs copy (valid) or fill (null) in parallel.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ void copy_elem(uint8_t* dst, const uint8_t* src,
+                                          int esize) {
+  if (esize == 8) {
+    uint32_t lo = load_u32_unaligned(src);
+    uint32_t hi = load_u32_unaligned(src + 4);
+    // dst is naturally aligned (column base + row*esize)
+    *(uint32_t*)dst = lo;
+    *(uint32_t*)(dst + 4) = hi;
+  } else if (esize == 4) {
+    *(uint32_t*)dst = load_u32_unaligned(src);
+  } else if (esize == 2) {
+    dst[0] = src[0];
+    dst[1] = src[1];
+  } else {
+    dst[0] = src[0];
+  }
+}
+
+__device__ __forceinline__ void fill_elem(uint8_t* dst, uint64_t pattern,
+                                          int esize) {
+  for (int b = 0; b < esize; ++b) dst[b] = (uint8_t)(pattern >> (8 * (b & 7)));
+}
+
+__global__ void plain_fixed_decode_kernel(
+    const uint8_t* __restrict__ page_buf,
+    const int64_t* __restrict__ payload_start,  // per page
+    const int64_t* __restrict__ payload_end,
+    const int32_t* __restrict__ n_values,       // rows in page (incl nulls)
+    const int64_t* __restrict__ row0,           // first output row of page
+    int32_t has_def, int32_t esize, uint64_t fill_pattern,
+    uint8_t* __restrict__ out,                  // column base (row-major)
+    uint8_t* __restrict__ valid_out,            // [total rows] or nullptr
+    int32_t* __restrict__ status, int n_pages) {
+  const int waves_per_block = blockDim.x / PSA_WAVE;
+  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  if (page >= n_pages) return;
+  const int lane = lane_id();
+  const int32_t want = n_values[page];
+  uint8_t* out_base = out + row0[page] * esize;
+  uint8_t* vbase = valid_out ? valid_out + row0[page] : nullptr;
+
+  int64_t pos = payload_start[page];
+  const int64_t hi = payload_end[page];
+
+  if (!has_def) {
+    // required column: straight parallel copy
+    const uint8_t* src = page_buf + pos;
+    for (int32_t i = lane; i < want; i += PSA_WAVE)
+      copy_elem(out_base + (int64_t)i * esize, src + (int64_t)i * esize,
+                esize);
+    if (vbase)
+      for (int32_t i = lane; i < want; i += PSA_WAVE) vbase[i] = 1;
+    return;
+  }
+
+  // definition levels: u32 length prefix then RLE hybrid at bit width 1
+  uint32_t dl_len = 0;
+  if (lane == 0) dl_len = load_u32_unaligned(page_buf + pos);
+  dl_len = wave_bcast(dl_len);
+  int64_t def_pos = pos + 4;
+  const int64_t def_end = def_pos + dl_len;
+  const uint8_t* values = page_buf + def_end;
+
+  int32_t row_cursor = 0;   // rows emitted
+  int32_t val_cursor = 0;   // non-null values consumed
+  while (row_cursor < want) {
+    uint32_t header = 0;
+    int is_packed = 0, run_len = 0;
+    uint32_t rle_value = 0;
+    int64_t payload = 0;
+    int done = 0;
+    if (lane == 0) {
+      if (def_pos >= def_end) {
+        done = 1;
+      } else {
+        header = read_varint_u32(page_buf, def_pos, def_end);
+        is_packed = header & 1;
+        if (is_packed) {
+          run_len = (int32_t)(header >> 1) * 8;
+          payload = def_pos;
+          def_pos += (int64_t)(header >> 1);  // bw=1: one byte per group
+        } else {
+          run_len = (int32_t)(header >> 1);
+          rle_value = page_buf[def_pos];
+          def_pos += 1;
+        }
+      }
+    }
+    done = wave_bcast(done);
+    if (done) { if (lane == 0) status[page] = 8; return; }
+    is_packed = wave_bcast(is_packed);
+    run_len = wave_bcast(run_len);
+    int32_t emit = min(run_len, want - row_cursor);
+    if (!is_packed) {
+      rle_value = wave_bcast(rle_value);
+      if (rle_value) {
+        const uint8_t* src = values + (int64_t)val_cursor * esize;
+        for (int32_t i = lane; i < emit; i += PSA_WAVE)
+          copy_elem(out_base + (int64_t)(row_cursor + i) * esize,
+                    src + (int64_t)i * esize, esize);
+        val_cursor += emit;
+      } else {
+        for (int32_t i = lane; i < emit; i += PSA_WAVE)
+          fill_elem(out_base + (int64_t)(row_cursor + i) * esize,
+                    fill_pattern, esize);
+      }
+      if (vbase)
+        for (int32_t i = lane; i < emit; i += PSA_WAVE)
+          vbase[row_cursor + i] = (uint8_t)(rle_value ? 1 : 0);
+      row_cursor += emit;
+    } else {
+      payload = wave_bcast(payload);
+      // process 64 levels per iteration: ballot + prefix popcount
+      for (int32_t base = 0; base < emit; base += PSA_WAVE) {
+        int32_t i = base + lane;
+        int my_bit = 0;
+        if (i < emit) {
+          int64_t bit = (int64_t)i;  // bit index within this run
+          my_bit = (page_buf[payload + (bit >> 3)] >> (bit & 7)) & 1;
+        }
+        unsigned long long mask = __ballot(my_bit != 0);
+        // lanes-below-me mask; lane 63 special-cased ((1ull<<64) is UB)
+        unsigned long long below = mask & ((lane < 63)
+                                   ? ((1ull << lane) - 1ull)
+                                   : 0x7FFFFFFFFFFFFFFFull);
+        int my_val_idx = val_cursor + (int)__popcll(below);
+        if (i < emit) {
+          uint8_t* dst = out_base + (int64_t)(row_cursor + i) * esize;
+          if (my_bit) {
+            copy_elem(dst, values + (int64_t)my_val_idx * esize, esize);
+          } else {
+            fill_elem(dst, fill_pattern, esize);
+          }
+          if (vbase) vbase[row_cursor + i] = (uint8_t)my_bit;
+        }
+        // lanes past `emit` contributed my_bit=0, so mask is already clean
+        val_cursor += (int)__popcll(mask);
+      }
+      row_cursor += emit;
+    }
+    (void)hi;
+  }
+}
+
+void plain_fixed_decode_batch(torch::Tensor page_buf,
+                              torch::Tensor payload_start,
+                              torch::Tensor payload_end,
+                              torch::Tensor n_values, torch::Tensor row0,
+                              int64_t has_def, int64_t esize,
+                              int64_t fill_pattern, torch::Tensor out,
+                              torch::Tensor valid_out, torch::Tensor status) {
+  int n = (int)payload_start.numel();
+  if (!n) return;
+  const int WPB = 4;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  uint8_t* vptr = valid_out.numel() ? valid_out.data_ptr<uint8_t>() : nullptr;
+  hipLaunchKernelGGL(plain_fixed_decode_kernel, dim3((n + WPB - 1) / WPB),
+                     dim3(WPB * PSA_WAVE), 0, stream,
+                     page_buf.data_ptr<uint8_t>(),
+                     payload_start.data_ptr<int64_t>(),
+                     payload_end.data_ptr<int64_t>(),
+                     n_values.data_ptr<int32_t>(),
+                     row0.data_ptr<int64_t>(), (int32_t)has_def,
+                     (int32_t)esize, (uint64_t)fill_pattern,
+                     out.data_ptr<uint8_t>(), vptr,
+                     status.data_ptr<int32_t>(), n);
+}
+
+}  // namespace psa
