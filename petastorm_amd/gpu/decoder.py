@@ -249,6 +249,7 @@ class GpuRowGroupDecoder(object):
         # every column OPTIONAL, so max_def==1 even for null-free data)
         if data_enc == _ENC_PLAIN and phys in _PHYS_TO_TORCH:
             sizes = uncomp_size if snappy else comp_size
+            page_nval = num_values[data_idx]
             p_start = np.array([page_start[i] for i in data_idx],
                                dtype=np.int64)
             p_end = np.array([page_start[i] + sizes[i] for i in data_idx],

@@ -119,6 +119,9 @@ class GpuBatchReader(object):
         self._file_md = {}
         self._stopped = False
         self._rows_epoch = 0
+        self._epoch = 0
+        self._piece_pos = 0
+        self._resume = None
         # stage timing (enabled with PSA_TIMING=1; reported in diagnostics)
         self._timing_enabled = os.environ.get('PSA_TIMING') == '1'
         self.stage_times = {'io_wait': 0.0, 'decode': 0.0, 'codec': 0.0,
@@ -189,8 +192,20 @@ class GpuBatchReader(object):
     # ------------------------------------------------------------------
     def _generate(self):
         epoch = 0
+        skip = 0
+        if self._resume is not None:
+            epoch = self._resume.get('epoch', 0)
+            skip = self._resume.get('piece_pos', 0)
+            self._resume = None
         while self._num_epochs is None or epoch < self._num_epochs:
+            self._epoch = epoch
             pieces = self._epoch_pieces(epoch)
+            if skip:
+                pieces = pieces[skip:]
+                self._piece_pos = skip
+                skip = 0
+            else:
+                self._piece_pos = 0
             q = queue.Queue(maxsize=_PREFETCH_DEPTH)
             t = threading.Thread(target=self._io_worker, args=(pieces, q),
                                  daemon=True)
@@ -201,6 +216,7 @@ class GpuBatchReader(object):
             # overlap the consumer's work on N; each row-group's status sync
             # happens just before ITS batch is yielded
             pending = None
+            dispatched = self._piece_pos  # pieces fully processed so far
             while True:
                 t0 = time.perf_counter()
                 kind, piece, host, meta = q.get()
@@ -232,16 +248,23 @@ class GpuBatchReader(object):
                 t2 = time.perf_counter()
                 batch = self._postprocess(piece, columns)
                 self.stage_times['postprocess'] += time.perf_counter() - t2
+                dispatched += 1
                 if pending is not None:
-                    yield pending
+                    nt, pos = pending
                     pending = None
+                    self._piece_pos = pos
+                    yield nt
                 if batch is None:
                     continue
                 self._decoder.flush_status()
                 self._rows_epoch += len(next(iter(batch.values())))
-                pending = self.schema.make_namedtuple(**batch)
+                # cursor value once THIS batch is consumed = pieces
+                # dispatched up to and including it
+                pending = (self.schema.make_namedtuple(**batch), dispatched)
             if pending is not None:
-                yield pending
+                nt, pos = pending
+                self._piece_pos = pos
+                yield nt
             t.join()
             epoch_sync.epoch_end_sync(self._rows_epoch)
             epoch += 1
@@ -372,6 +395,27 @@ class GpuBatchReader(object):
     next = __next__
 
     def reset(self):
+        self._gen = self._generate()
+        self.last_row_consumed = False
+
+    # ------------------------------------------------------------------
+    # iterator-state checkpointing (the reference offers
+    # only determinism-by-seed as a resume substitute; this framework saves
+    # the cursor explicitly).  Requires a seeded reader (or an initialized
+    # process group, where rank 0's broadcast makes epochs reproducible
+    # within one job but NOT across restarts — so persist with seed set).
+    # ------------------------------------------------------------------
+    def state_dict(self):
+        return {'epoch': self._epoch, 'piece_pos': self._piece_pos,
+                'seed': self._seed}
+
+    def load_state_dict(self, state):
+        if state.get('seed') != self._seed:
+            raise ValueError('state was saved with seed={!r}; reader has '
+                             'seed={!r} — epoch permutations would diverge'
+                             .format(state.get('seed'), self._seed))
+        self._resume = {'epoch': state['epoch'],
+                        'piece_pos': state['piece_pos']}
         self._gen = self._generate()
         self.last_row_consumed = False
 

@@ -671,3 +671,24 @@ def test_gpu_batch_reader_helloworld_png(ext, tmp_path):
         imgs = b.image1.cpu().numpy()
         for i in range(0, len(ids), 8):
             np.testing.assert_array_equal(imgs[i], cpu[int(ids[i])])
+
+
+def test_gpu_reader_state_dict_resume(ext, tmp_path):
+    """Checkpoint/resume: a reloaded reader continues at the saved cursor."""
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'ckpt')
+    create_scalar_dataset(url, num_rows=1000, rowgroup_size=100)
+    kwargs = dict(device='cuda', shuffle_row_groups=True, seed=99,
+                  num_epochs=2, schema_fields=['id'])
+    with make_batch_reader(url, **kwargs) as r1:
+        it = iter(r1)
+        seen = [next(it) for _ in range(4)]
+        state = r1.state_dict()
+        rest_a = [b.id.cpu() for b in it]
+    with make_batch_reader(url, **kwargs) as r2:
+        r2.load_state_dict(state)
+        rest_b = [b.id.cpu() for b in r2]
+    assert len(rest_a) == len(rest_b)
+    for a, b in zip(rest_a, rest_b):
+        assert torch.equal(a, b)
