@@ -160,6 +160,9 @@ def bench_scalar(args, rank, world, device, dist):
     result = _run_timed(args, step, device, dist, world)
     reader.stop()
     reader.join()
+    if os.environ.get('PSA_TIMING') == '1' and rank == 0:
+        print('stage_times:', reader.diagnostics.get('stage_times'),
+              file=sys.stderr)
     return result, {
         'model': 'scalar-parquet (8 float64 + 8 int64 cols, snappy)',
         'global_batch': args.batch_size * 64 * world,
@@ -237,6 +240,9 @@ def bench_ngram(args, rank, world, device, dist):
     result = _run_timed(args, step, device, dist, world)
     reader.stop()
     reader.join()
+    if os.environ.get('PSA_TIMING') == '1' and rank == 0:
+        print('stage_times:', reader.diagnostics.get('stage_times'),
+              file=sys.stderr)
     return result, {
         'model': 'SequenceSchema (1024-token int32 NdarrayCodec + predicate '
                  '+ shuffling queue, HBM cache)',

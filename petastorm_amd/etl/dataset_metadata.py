@@ -33,9 +33,27 @@ DEFAULT_ROWGROUP_SIZE_MB = 32
 
 #: One Parquet row group: the framework's unit of IO, sharding and shuffling.
 #: ``index`` is the global ordinal across the sorted file list (the value
-#: sharding is computed over, reference reader.py:573-597).
+#: sharding is computed over, reference reader.py:573-597).  ``partitions``
+#: holds hive-style ``key=value`` directory values for partitioned stores
+#: (reference supports these through pq.ParquetDataset partitions,
+#: arrow_reader_worker.py:358).
 RowGroupPiece = namedtuple('RowGroupPiece',
-                           ['index', 'path', 'row_group', 'num_rows'])
+                           ['index', 'path', 'row_group', 'num_rows',
+                            'partitions'])
+RowGroupPiece.__new__.__defaults__ = ({},)
+
+
+def parse_partition_values(root, file_path):
+    """Extract hive-style key=value path segments between ``root`` and the
+    file: .../root/color=red/size=2/part.parquet -> {'color': 'red',
+    'size': '2'}."""
+    rel = file_path[len(root):].lstrip('/')
+    out = {}
+    for seg in rel.split('/')[:-1]:
+        if '=' in seg:
+            k, v = seg.split('=', 1)
+            out[k] = v
+    return out
 
 
 class DatasetWriter(object):
@@ -195,15 +213,25 @@ def list_parquet_files(fs, path_or_paths):
 
 def load_row_groups(fs, path_or_paths):
     """Enumerate every row group of the dataset as RowGroupPieces
-    (reference load_row_groups, :244-290)."""
+    (reference load_row_groups, :244-290).  Hive-partition directory values
+    are attached to each piece."""
     import pyarrow.parquet as pq
+    roots = path_or_paths if isinstance(path_or_paths, list) \
+        else [path_or_paths]
+    dir_roots = [r for r in roots if fs.isdir(r)]
     pieces = []
     index = 0
     for fpath in list_parquet_files(fs, path_or_paths):
+        partitions = {}
+        for r in dir_roots:
+            if fpath.startswith(r.rstrip('/') + '/'):
+                partitions = parse_partition_values(r.rstrip('/'), fpath)
+                break
         md = pq.ParquetFile(fs.open(fpath, 'rb')).metadata
         for rg in range(md.num_row_groups):
             pieces.append(RowGroupPiece(index, fpath, rg,
-                                        md.row_group(rg).num_rows))
+                                        md.row_group(rg).num_rows,
+                                        partitions))
             index += 1
     return pieces
 
@@ -241,12 +269,36 @@ def get_schema_from_dataset_url(dataset_url_or_urls, storage_options=None):
 def infer_or_load_unischema(fs, path_or_paths):
     """Stored schema when present, else inference from the Arrow schema
     (reference infer_or_load_unischema, :410-418)."""
+    import numpy as _np
+
+    from petastorm_amd.unischema import UnischemaField
+
+    def _with_partition_fields(schema):
+        pieces = load_row_groups(fs, path_or_paths)
+        part_keys = sorted({k for p in pieces for k in p.partitions})
+        extra = []
+        for k in part_keys:
+            if k in schema.fields:
+                continue
+            values = {p.partitions.get(k) for p in pieces}
+            try:
+                all(int(v) for v in values if v is not None)
+                dtype = _np.int64
+            except (TypeError, ValueError):
+                dtype = _np.str_
+            extra.append(UnischemaField(k, dtype, (), None, False))
+        if not extra:
+            return schema
+        return Unischema(schema._name,
+                         list(schema.fields.values()) + extra)
+
     try:
-        return get_schema(fs, path_or_paths), True
+        return _with_partition_fields(get_schema(fs, path_or_paths)), True
     except ValueError:
         import pyarrow.parquet as pq
         files = list_parquet_files(fs, path_or_paths)
         if not files:
             raise ValueError('No parquet files found at {}'.format(path_or_paths))
         arrow_schema = pq.ParquetFile(fs.open(files[0], 'rb')).schema_arrow
-        return Unischema.from_arrow_schema(arrow_schema), False
+        return _with_partition_fields(
+            Unischema.from_arrow_schema(arrow_schema)), False

@@ -247,6 +247,9 @@ class Reader(object):
         if not self._pieces:
             raise NoDataAvailableError('Dataset has no row groups')
         selected = self._apply_row_group_selector(rowgroup_selector)
+        predicate = self._push_down_partition_predicate(predicate, selected)
+        if isinstance(predicate, tuple):  # (pushed_down_selection, None)
+            selected, predicate = predicate
         selected = self._apply_shard(selected, cur_shard, shard_count, seed,
                                      shuffle_row_groups)
 
@@ -290,6 +293,35 @@ class Reader(object):
         self._row_buffer = []
 
     # ------------------------------------------------------------------
+    def _push_down_partition_predicate(self, predicate, selected):
+        """When every predicate field is a hive-partition key, filter row
+        groups by path values and drop the worker-side predicate entirely
+        (reference _apply_predicate_to_row_groups, reader.py:620-652)."""
+        if predicate is None:
+            return None
+        try:
+            pred_fields = set(predicate.get_fields())
+        except Exception:  # noqa: BLE001 - user predicate may defer fields
+            return predicate
+        if not pred_fields or not all(
+                pred_fields <= set(self._pieces[i].partitions)
+                for i in selected):
+            return predicate
+        from petastorm_amd.workers.row_worker import _partition_value
+
+        def keep(i):
+            p = self._pieces[i]
+            vals = {f: _partition_value(self._storage_schema.fields.get(f),
+                                        p.partitions[f])
+                    for f in pred_fields}
+            return predicate.do_include(vals)
+
+        kept = [i for i in selected if keep(i)]
+        if not kept:
+            logger.warning('Partition predicate filtered out every row '
+                           'group (reference reader.py:567-569)')
+        return (kept, None)
+
     def _apply_row_group_selector(self, selector):
         """reference :599-618"""
         indexes = list(range(len(self._pieces)))

@@ -158,8 +158,18 @@ class BatchReaderWorker(WorkerBase):
         available = set(pf.schema_arrow.names)
         cols = [c for c in column_names if c in available]
         table = pf.read_row_group(piece.row_group, columns=cols)
-        return arrow_table_to_numpy_dict(table, self._a.schema,
-                                         self._a.decode_codecs)
+        out = arrow_table_to_numpy_dict(table, self._a.schema,
+                                        self._a.decode_codecs)
+        for c in column_names:
+            if c not in available and c in piece.partitions:
+                field = self._a.schema.fields.get(c)
+                if field is not None and field.numpy_dtype not in (np.str_,):
+                    v = np.dtype(field.numpy_dtype).type(piece.partitions[c])
+                    out[c] = np.full(table.num_rows, v)
+                else:
+                    out[c] = np.full(table.num_rows, str(piece.partitions[c]),
+                                     dtype=object)
+        return out
 
     def _load_with_predicate(self, piece, predicate, needed):
         """Vectorized predicate: mask on predicate columns first, early exit,

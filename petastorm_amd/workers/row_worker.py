@@ -43,6 +43,13 @@ class RowWorkerArgs(object):
         self.seed = seed
 
 
+def _partition_value(field, raw):
+    """Cast a hive path value to the field's dtype."""
+    if field is None or field.numpy_dtype in (np.str_,):
+        return str(raw)
+    return np.dtype(field.numpy_dtype).type(raw)
+
+
 def _cache_key(path, row_group, column_names):
     h = hashlib.md5('{}:{}:{}'.format(path, row_group,
                                       ','.join(sorted(column_names)))
@@ -89,15 +96,26 @@ class RowReaderWorker(WorkerBase):
         return self._parquet_files[path]
 
     def _load_rows(self, piece, column_names):
-        """Read one row group as a list of raw (encoded) row dicts."""
+        """Read one row group as a list of raw (encoded) row dicts.
+        Hive-partition key columns are materialized from the piece's path
+        values (reference reads them through pq.ParquetDataset partitions,
+        py_dict_reader_worker.py:267)."""
         pf = self._parquet_file(piece.path)
         available = set(pf.schema_arrow.names)
         cols = [c for c in column_names if c in available]
         table = pf.read_row_group(piece.row_group, columns=cols)
         pydict = table.to_pydict()
         names = list(pydict.keys())
-        return [dict(zip(names, vals)) for vals in zip(*pydict.values())] \
-            if names else []
+        rows = [dict(zip(names, vals)) for vals in zip(*pydict.values())] \
+            if names else [{} for _ in range(table.num_rows)]
+        part_cols = [c for c in column_names
+                     if c not in available and c in piece.partitions]
+        for c in part_cols:
+            v = _partition_value(self._a.schema.fields.get(c),
+                                 piece.partitions[c])
+            for r in rows:
+                r[c] = v
+        return rows
 
     def _load_rows_with_predicate(self, piece, predicate):
         """Two-phase load: predicate columns first, then the rest only for
@@ -107,6 +125,18 @@ class RowReaderWorker(WorkerBase):
         other_fields = [f for f in all_fields if f not in predicate_fields]
 
         pf = self._parquet_file(piece.path)
+        available = set(pf.schema_arrow.names)
+        part_pred = [f for f in predicate_fields
+                     if f not in available and f in piece.partitions]
+        predicate_fields = [f for f in predicate_fields if f in available]
+        if part_pred and not predicate_fields:
+            # all predicate fields are partition keys: evaluate once
+            vals = {f: _partition_value(self._a.schema.fields.get(f),
+                                        piece.partitions[f])
+                    for f in part_pred}
+            if not predicate.do_include(vals):
+                return []
+            return self._load_rows(piece, all_fields)
         pred_table = pf.read_row_group(piece.row_group,
                                        columns=predicate_fields)
         pred_cols = {name: pred_table.column(name).to_pylist()
