@@ -20,6 +20,8 @@ class DummyPool(object):
         self._items = deque()
         self._results = deque()
         self._stopped = False
+        self._ventilated = 0
+        self._processed = 0
 
     def start(self, worker_class, worker_args=None, ventilator=None):
         self._worker = worker_class(0, self._results.append, worker_args)
@@ -29,6 +31,7 @@ class DummyPool(object):
 
     def ventilate(self, *args, **kwargs):
         item = kwargs if kwargs else (args if len(args) != 1 else args[0])
+        self._ventilated += 1
         self._items.append(item)
 
     def get_results(self):
@@ -48,6 +51,7 @@ class DummyPool(object):
                     else:
                         self._worker.process(item)
                 finally:
+                    self._processed += 1
                     if self._ventilator is not None:
                         self._ventilator.processed_item()
                 continue
@@ -69,4 +73,6 @@ class DummyPool(object):
 
     @property
     def diagnostics(self):
-        return {'output_queue_size': len(self._results)}
+        return {'output_queue_size': len(self._results),
+                'items_ventilated': self._ventilated,
+                'items_processed': self._processed}
