@@ -23,3 +23,16 @@ def decode_row(row, schema):
             continue
         decoded[name] = _codecs.decode_value(field, value)
     return decoded
+
+
+def run_in_subprocess(func, *args, **kwargs):
+    """Run ``func(*args, **kwargs)`` in a spawned subprocess and return its
+    result (reference petastorm/utils.py:28-45).
+
+    Used to isolate work that must not pollute the parent (e.g. library
+    global state).  The callable and its arguments must be picklable.
+    """
+    import multiprocessing as mp
+    ctx = mp.get_context('spawn')
+    with ctx.Pool(1) as pool:
+        return pool.apply(func, args, kwargs)
