@@ -74,6 +74,22 @@ class GpuRowGroupDecoder(object):
         # tensor; flush_status() does ONE host sync per row-group instead of
         # one per kernel
         self._pending_status = []
+        self._pin_memory = torch.cuda.is_available()
+
+    def _up(self, arr):
+        """Async host->device upload of a small numpy array / cpu tensor.
+
+        A plain ``torch.from_numpy(x).to(dev)`` on unpinned memory is a
+        BLOCKING copy that synchronizes the stream — profiled at >1ms per
+        call and the dominant cost of the scalar config.  Staging through
+        torch's cached pinned allocator keeps every upload asynchronous.
+        """
+        t = arr if isinstance(arr, torch.Tensor) else torch.from_numpy(arr)
+        if self._pin_memory and not t.is_pinned():
+            pinned = torch.empty_like(t, pin_memory=True)
+            pinned.copy_(t)
+            t = pinned
+        return t.to(self.device, non_blocking=True)
 
     # ------------------------------------------------------------------
     def read_rowgroup_bytes(self, path, file_metadata, parquet_schema, rg,
@@ -219,8 +235,8 @@ class GpuRowGroupDecoder(object):
             c_off[-1] = data_off[-1] + comp_size[-1]
             status = torch.zeros(n_pages, dtype=torch.int32, device=dev)
             ext.snappy_decompress_batch(
-                dbuf, torch.from_numpy(c_off).to(dev),
-                ubuf, torch.from_numpy(u_off).to(dev), status)
+                dbuf, self._up(c_off),
+                ubuf, self._up(u_off), status)
             self._check(status, 'snappy:' + ch['name'])
             page_buf = ubuf
             page_start = u_off[:-1]
@@ -307,11 +323,10 @@ class GpuRowGroupDecoder(object):
             status = torch.zeros(len(data_idx), dtype=torch.int32,
                                  device=dev)
             ext.rle_hybrid_decode_batch(
-                page_buf, torch.from_numpy(def_start).to(dev),
-                torch.from_numpy(def_end).to(dev),
+                page_buf, self._up(def_start), self._up(def_end),
                 torch.ones(len(data_idx), dtype=torch.int32, device=dev),
-                torch.from_numpy(page_nval.astype(np.int32)).to(dev),
-                torch.from_numpy(lv_off[:-1]).to(dev), levels, status)
+                self._up(page_nval.astype(np.int32)),
+                self._up(lv_off[:-1]), levels, status)
             self._check(status, 'deflevels:' + ch['name'])
             valid = levels.bool()
             # per-page non-null counts (needed to place value sections)
@@ -348,11 +363,8 @@ class GpuRowGroupDecoder(object):
                           device=dev)
         status = torch.zeros(len(p_start), dtype=torch.int32, device=dev)
         ext.plain_fixed_decode_batch(
-            page_buf, torch.from_numpy(p_start).to(dev, non_blocking=True),
-            torch.from_numpy(p_end).to(dev, non_blocking=True),
-            torch.from_numpy(page_nval.astype(np.int32)).to(
-                dev, non_blocking=True),
-            torch.from_numpy(row0).to(dev, non_blocking=True),
+            page_buf, self._up(p_start), self._up(p_end),
+            self._up(page_nval.astype(np.int32)), self._up(row0),
             1 if max_def > 0 else 0, esize, self._FILL_PATTERNS[phys],
             out, torch.empty(0, dtype=torch.uint8, device=dev), status)
         self._check(status, 'plainfixed:' + name)
@@ -369,10 +381,8 @@ class GpuRowGroupDecoder(object):
             # offsets were already scanned on host by prepare_host (in the
             # IO thread) — upload them instead of launching the serial scan
             # kernel (which is latency-bound at a handful of pages)
-            val_off = torch.from_numpy(plan['host_off']).to(
-                dev, non_blocking=True)
-            val_len = torch.from_numpy(
-                plan['host_len'].astype(np.int32)).to(dev, non_blocking=True)
+            val_off = self._up(plan['host_off'])
+            val_len = self._up(plan['host_len'].astype(np.int32))
         else:
             o_off = np.zeros(len(counts), dtype=np.int64)
             o_off[1:] = np.cumsum(counts)[:-1]
@@ -380,10 +390,9 @@ class GpuRowGroupDecoder(object):
             val_len = torch.empty(total, dtype=torch.int32, device=dev)
             status = torch.zeros(len(counts), dtype=torch.int32, device=dev)
             ext.byte_array_offsets_batch(
-                page_buf, torch.from_numpy(val_start).to(dev),
-                torch.from_numpy(val_end).to(dev),
-                torch.from_numpy(counts.astype(np.int32)).to(dev),
-                torch.from_numpy(o_off).to(dev), val_off, val_len, status)
+                page_buf, self._up(val_start), self._up(val_end),
+                self._up(counts.astype(np.int32)),
+                self._up(o_off), val_off, val_len, status)
             self._check(status, 'bytearray:' + ch['name'])
         host_off = None
         if host_visible:
@@ -433,11 +442,9 @@ class GpuRowGroupDecoder(object):
         indices = torch.empty(int(i_off[-1]), dtype=torch.int32, device=dev)
         status = torch.zeros(len(counts), dtype=torch.int32, device=dev)
         ext.rle_hybrid_decode_batch(
-            page_buf, torch.from_numpy(val_start + 1).to(dev),
-            torch.from_numpy(val_end).to(dev),
-            torch.from_numpy(bw).to(dev),
-            torch.from_numpy(counts.astype(np.int32)).to(dev),
-            torch.from_numpy(i_off[:-1]).to(dev), indices, status)
+            page_buf, self._up(val_start + 1), self._up(val_end),
+            self._up(bw), self._up(counts.astype(np.int32)),
+            self._up(i_off[:-1]), indices, status)
         self._check(status, 'dictidx:' + phys)
         values = dict_vals[indices.long()]
         if valid is None:
@@ -570,12 +577,12 @@ class GpuRowGroupDecoder(object):
                           device=dev)
         produced = torch.zeros(n, dtype=torch.int64, device=dev)
         status = torch.zeros(n, dtype=torch.int32, device=dev)
-        ext.inflate_batch(col.device_buf, meta['seg_off'].to(dev),
-                          meta['seg_len'].to(dev),
-                          meta['seg_first'].to(dev),
-                          meta['seg_count'].to(dev), raw,
-                          torch.from_numpy(raw_off).to(dev),
-                          torch.from_numpy(raw_size).to(dev), produced, 0,
+        ext.inflate_batch(col.device_buf, self._up(meta['seg_off']),
+                          self._up(meta['seg_len']),
+                          self._up(meta['seg_first']),
+                          self._up(meta['seg_count']), raw,
+                          self._up(raw_off),
+                          self._up(raw_size), produced, 0,
                           status)
         self._check(status, 'png-inflate:' + field.name)
         out_bytes = row_bytes * heights
@@ -584,11 +591,11 @@ class GpuRowGroupDecoder(object):
         out = torch.empty(int(out_bytes.sum()) + _SLACK, dtype=torch.uint8,
                           device=dev)
         st2 = torch.zeros(n, dtype=torch.int32, device=dev)
-        ext.png_unfilter_batch(raw, torch.from_numpy(raw_off).to(dev), out,
-                               torch.from_numpy(out_off).to(dev),
-                               meta['height'].to(dev),
-                               meta['row_bytes'].to(dev),
-                               meta['bpp'].to(dev), st2)
+        ext.png_unfilter_batch(raw, self._up(raw_off), out,
+                               self._up(out_off),
+                               self._up(meta['height']),
+                               self._up(meta['row_bytes']),
+                               self._up(meta['bpp']), st2)
         self._check(st2, 'png-unfilter:' + field.name)
         if len(set(widths.tolist())) != 1 or len(set(heights.tolist())) != 1 \
                 or len(set(channels.tolist())) != 1 \
@@ -623,7 +630,7 @@ class GpuRowGroupDecoder(object):
         # move every tensor to device
         meta_dev = {}
         for k, v in meta.items():
-            meta_dev[k] = v.to(dev) if isinstance(v, torch.Tensor) else v
+            meta_dev[k] = self._up(v) if isinstance(v, torch.Tensor) else v
         block_total = int(meta['block_total'])
         samp_total = int(meta['samp_total'])
         coef = torch.zeros(block_total * 64, dtype=torch.float32, device=dev)
@@ -637,7 +644,7 @@ class GpuRowGroupDecoder(object):
         n_segs = int(meta['seg_img'].numel())
         status = torch.zeros(max(n_segs, 1), dtype=torch.int32, device=dev)
         ext.jpeg_decode_batch(col.device_buf, meta_dev, coef, samples, out,
-                              torch.from_numpy(out_off).to(dev), status)
+                              self._up(out_off), status)
         self._check(status, 'jpeg:' + field.name)
         # uniform-shape batch -> dense [n, H, W, C]
         if len(set(widths.tolist())) == 1 and len(set(heights.tolist())) == 1 \

@@ -324,8 +324,7 @@ class GpuBatchReader(object):
         out = {}
         for k, v in np_dict.items():
             if isinstance(v, np.ndarray) and v.dtype.kind in 'iufb':
-                out[k] = torch.from_numpy(np.ascontiguousarray(v)).to(
-                    self.device, non_blocking=True)
+                out[k] = self._decoder._up(np.ascontiguousarray(v))
             else:
                 out[k] = v  # strings/objects stay host-side
         return out
@@ -355,7 +354,7 @@ class GpuBatchReader(object):
             if self._seed is not None:
                 g = torch.Generator(device='cpu')
                 g.manual_seed((self._seed + piece.index) % (2 ** 31))
-            perm = torch.randperm(n2, generator=g).to(self.device)
+            perm = self._decoder._up(torch.randperm(n2, generator=g))
             columns = {k: (v.index_select(0, perm)
                            if isinstance(v, torch.Tensor)
                            else v[perm.cpu().numpy()])
@@ -377,7 +376,7 @@ class GpuBatchReader(object):
             host_cols[f] = v.cpu().numpy() if isinstance(v, torch.Tensor) \
                 else np.asarray(v)
         mask = self._predicate.do_include_vectorized(host_cols)
-        return torch.from_numpy(np.asarray(mask, dtype=bool)).to(self.device)
+        return self._decoder._up(np.asarray(mask, dtype=bool))
 
     # ------------------------------------------------------------------
     def __iter__(self):
