@@ -115,21 +115,42 @@ class GpuRowGroupDecoder(object):
                            col.physical_type, col.compression,
                            col.num_values,
                            parquet_schema.column(ci).max_definition_level))
-        nbytes = hi - lo
+        # merge the requested chunk byte ranges into extents (gaps below 256
+        # KiB are read through rather than seeking) so unrequested columns
+        # between them are not read or uploaded
+        ranges = sorted((s, s + ln) for (_, _, s, ln, *_rest) in chunks)
+        extents = []
+        for s, e in ranges:
+            if extents and s - extents[-1][1] <= (256 << 10):
+                extents[-1][1] = max(extents[-1][1], e)
+            else:
+                extents.append([s, e])
+        nbytes = sum(e - s for s, e in extents)
         if pinned_pool is not None:
             host = pinned_pool.get(nbytes + _SLACK)
         else:
             host = torch.empty(nbytes + _SLACK, dtype=torch.uint8,
                                pin_memory=torch.cuda.is_available())
+        mv = memoryview(host.numpy())
+        dest = 0
+        extent_map = []  # (file_start, file_end, buffer_offset)
         with open(path, 'rb') as f:
-            f.seek(lo)
-            mv = memoryview(host.numpy())[:nbytes]
-            f.readinto(mv)
+            for s, e in extents:
+                f.seek(s)
+                f.readinto(mv[dest:dest + (e - s)])
+                extent_map.append((s, e, dest))
+                dest += e - s
+
+        def _rebase(file_off):
+            for s, e, d in extent_map:
+                if s <= file_off < e:
+                    return d + (file_off - s)
+            raise ValueError('offset outside read extents')
+
         chunk_meta = {
-            'base_offset': lo,
             'num_rows': md.num_rows,
             'chunks': [
-                dict(name=n, col_index=ci, offset=s - lo, length=ln,
+                dict(name=n, col_index=ci, offset=_rebase(s), length=ln,
                      physical=pt, compression=comp, num_values=nv,
                      max_def=mdl)
                 for (n, ci, s, ln, pt, comp, nv, mdl) in chunks],
