@@ -146,6 +146,40 @@ __device__ __forceinline__ int receive_extend(BitReader& br, int s) {
   return v;
 }
 
+// Decode one 8x8 block's Huffman-coded coefficients.  `pred` is a named
+// register passed by reference (a pred[c] array would be runtime-indexed and
+// spill to scratch — guide rule: dynamic-indexed locals live in local
+// memory).  Returns 0 on success, an error code otherwise.
+__device__ __forceinline__ int decode_block(BitReader& br,
+                                            const JpegTables& tabs,
+                                            const float* __restrict__ q,
+                                            int dc_t, int ac_t,
+                                            float* __restrict__ out,
+                                            int& pred) {
+  int t = huff_decode(br, tabs, dc_t);
+  if (t < 0 || t > 15) return 10;
+  int diff = t ? receive_extend(br, t) : 0;
+  pred += diff;
+  out[0] = (float)pred * q[0];
+  int kk = 1;
+  while (kk < 64) {
+    int rs = huff_decode(br, tabs, ac_t);
+    if (rs < 0) return 11;
+    int r = rs >> 4, sz = rs & 15;
+    if (sz == 0) {
+      if (r != 15) break;  // EOB
+      kk += 16;
+    } else {
+      kk += r;
+      if (kk > 63) return 12;
+      int v = receive_extend(br, sz);
+      out[ZIGZAG_NAT[kk]] = (float)v * q[kk];
+      ++kk;
+    }
+  }
+  return 0;
+}
+
 __global__ void jpeg_huffman_kernel(
     const uint8_t* __restrict__ data, JpegTables tabs, JpegGeom g,
     const int32_t* __restrict__ seg_img, const int64_t* __restrict__ seg_pos,
@@ -156,49 +190,44 @@ __global__ void jpeg_huffman_kernel(
   if (s >= n_segs) return;
   const int img = seg_img[s];
   const int bpm = g.bpm[img];
+  const int ncomp = g.ncomp[img];
 
   BitReader br;
   br.init(data, seg_pos[s], seg_end[s]);
 
-  int pred[3] = {0, 0, 0};
+  // hoist per-component tables into named registers (per-image constants)
+  const float* q0 = tabs.qtabs + (int64_t)g.comp_q[img * 3 + 0] * 64;
+  const int dc0 = g.comp_dc[img * 3 + 0], ac0 = g.comp_ac[img * 3 + 0];
+  const int rep0 = g.comp_h[img * 3 + 0] * g.comp_v[img * 3 + 0];
+  const float* q1 = q0;
+  const float* q2 = q0;
+  int dc1 = 0, ac1 = 0, dc2 = 0, ac2 = 0;
+  if (ncomp == 3) {
+    q1 = tabs.qtabs + (int64_t)g.comp_q[img * 3 + 1] * 64;
+    dc1 = g.comp_dc[img * 3 + 1];
+    ac1 = g.comp_ac[img * 3 + 1];
+    q2 = tabs.qtabs + (int64_t)g.comp_q[img * 3 + 2] * 64;
+    dc2 = g.comp_dc[img * 3 + 2];
+    ac2 = g.comp_ac[img * 3 + 2];
+  }
+
+  int pred0 = 0, pred1 = 0, pred2 = 0;
   const int mcu0 = seg_mcu0[s];
   const int nmcu = seg_nmcu[s];
   const int64_t blk0 = g.img_block0[img];
 
   for (int m = 0; m < nmcu; ++m) {
-    const int64_t mcu_blk = blk0 + (int64_t)(mcu0 + m) * bpm;
-    for (int k = 0; k < bpm; ++k) {
-      const int km = g.kmap[img * 8 + k];
-      const int c = km >> 8;
-      const float* q = tabs.qtabs + (int64_t)g.comp_q[img * 3 + c] * 64;
-      const int dc_t = g.comp_dc[img * 3 + c];
-      const int ac_t = g.comp_ac[img * 3 + c];
-      float* out = coef + (mcu_blk + k) * 64;
-
-      int t = huff_decode(br, tabs, dc_t);
-      if (t < 0 || t > 15) { status[s] = 10; return; }
-      int diff = t ? receive_extend(br, t) : 0;
-      pred[c] += diff;
-      out[0] = (float)pred[c] * q[0];
-
-      int kk = 1;
-      while (kk < 64) {
-        int rs = huff_decode(br, tabs, ac_t);
-        if (rs < 0) { status[s] = 11; return; }
-        int r = rs >> 4, sz = rs & 15;
-        if (sz == 0) {
-          if (r != 15) break;  // EOB
-          kk += 16;
-        } else {
-          kk += r;
-          if (kk > 63) { status[s] = 12; return; }
-          int v = receive_extend(br, sz);
-          int nat = ZIGZAG_NAT[kk];
-          out[nat] = (float)v * q[kk];
-          ++kk;
-        }
-      }
+    float* mcu_out = coef + (blk0 + (int64_t)(mcu0 + m) * bpm) * 64;
+    int rc = 0;
+    for (int r = 0; r < rep0 && !rc; ++r, mcu_out += 64)
+      rc = decode_block(br, tabs, q0, dc0, ac0, mcu_out, pred0);
+    if (!rc && ncomp == 3) {
+      rc = decode_block(br, tabs, q1, dc1, ac1, mcu_out, pred1);
+      mcu_out += 64;
+      if (!rc)
+        rc = decode_block(br, tabs, q2, dc2, ac2, mcu_out, pred2);
     }
+    if (rc) { status[s] = rc; return; }
   }
 }
 
