@@ -217,7 +217,7 @@ class GpuRowGroupDecoder(object):
         for ch in chunk_meta['chunks']:
             name = ch['name']
             comp = ch['compression']
-            if comp not in ('UNCOMPRESSED', 'SNAPPY'):
+            if comp not in ('UNCOMPRESSED', 'SNAPPY', 'GZIP'):
                 out[name] = self._cpu_assist_marker(name)
                 continue
             col = self._decode_chunk(ext, dev, dbuf, host_buf, ch,
@@ -241,7 +241,7 @@ class GpuRowGroupDecoder(object):
         num_values = pages['num_values'].numpy()
         encoding = pages['encoding'].numpy()
         n_pages = len(page_type)
-        snappy = ch['compression'] == 'SNAPPY'
+        snappy = ch['compression'] in ('SNAPPY', 'GZIP')
 
         # 1) page payload location: either in dbuf directly, or in a
         #    decompressed scratch buffer
@@ -255,10 +255,23 @@ class GpuRowGroupDecoder(object):
             c_off[:-1] = data_off
             c_off[-1] = data_off[-1] + comp_size[-1]
             status = torch.zeros(n_pages, dtype=torch.int32, device=dev)
-            ext.snappy_decompress_batch(
-                dbuf, self._up(c_off),
-                ubuf, self._up(u_off), status)
-            self._check(status, 'snappy:' + ch['name'])
+            if ch['compression'] == 'SNAPPY':
+                ext.snappy_decompress_batch(
+                    dbuf, self._up(c_off),
+                    ubuf, self._up(u_off), status)
+                self._check(status, 'snappy:' + ch['name'])
+            else:  # GZIP: each page is one gzip member -> inflate kernel
+                produced = torch.zeros(n_pages, dtype=torch.int64,
+                                       device=dev)
+                ext.inflate_batch(
+                    dbuf, self._up(data_off.astype(np.int64)),
+                    self._up(comp_size.astype(np.int64)),
+                    torch.arange(n_pages, dtype=torch.int32, device=dev),
+                    torch.ones(n_pages, dtype=torch.int32, device=dev),
+                    ubuf, self._up(u_off[:-1]),
+                    self._up(uncomp_size.astype(np.int64)), produced, 2,
+                    status)
+                self._check(status, 'gzip:' + ch['name'])
             page_buf = ubuf
             page_start = u_off[:-1]
             host_visible = False

@@ -133,6 +133,23 @@ __global__ void inflate_kernel(const uint8_t* __restrict__ src,
     br.next_byte();
     int flg = br.next_byte();
     if (flg & 0x20) { status[s] = 20; return; }  // FDICT unsupported
+  } else if (mode == 2) {  // gzip member header (RFC 1952)
+    int id1 = br.next_byte(), id2 = br.next_byte();
+    if (id1 != 0x1f || id2 != 0x8b) { status[s] = 26; return; }
+    int cm = br.next_byte();
+    if (cm != 8) { status[s] = 26; return; }
+    int flg = br.next_byte();
+    for (int i = 0; i < 6; ++i) br.next_byte();  // mtime, xfl, os
+    if (flg & 0x04) {  // FEXTRA
+      int xl = br.next_byte();
+      xl |= br.next_byte() << 8;
+      for (int i = 0; i < xl; ++i) br.next_byte();
+    }
+    if (flg & 0x08)  // FNAME
+      while (br.next_byte() > 0) {}
+    if (flg & 0x10)  // FCOMMENT
+      while (br.next_byte() > 0) {}
+    if (flg & 0x02) { br.next_byte(); br.next_byte(); }  // FHCRC
   }
   uint8_t* out = dst + dst_off[s];
   const int64_t cap = dst_cap[s];

@@ -692,3 +692,16 @@ def test_gpu_reader_state_dict_resume(ext, tmp_path):
     assert len(rest_a) == len(rest_b)
     for a, b in zip(rest_a, rest_b):
         assert torch.equal(a, b)
+
+
+def test_gzip_rowgroup_decode(ext, tmp_path):
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    from petastorm_amd import make_batch_reader
+    url = 'file://' + str(tmp_path / 'gz')
+    create_scalar_dataset(url, num_rows=3000, rowgroup_size=1000,
+                          compression='gzip')
+    with make_batch_reader(url, device='cuda', shuffle_row_groups=False,
+                           schema_fields=['id', 'f0']) as r:
+        ids = torch.cat([b.id for b in r]).cpu().numpy()
+        assert not r.diagnostics['cpu_assist_columns']
+    np.testing.assert_array_equal(np.sort(ids), np.arange(3000))
