@@ -5,7 +5,12 @@ Parity role: /root/reference/examples/hello_world/petastorm_dataset/
 
 Run:  python examples/hello_world/main.py [output_url]
 """
+import os
 import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                '..', '..'))
+
 import tempfile
 
 import numpy as np

@@ -7,7 +7,12 @@ loop receives ready NCHW fp32 CUDA tensors.
 
 Run on a GPU box:  python examples/imagenet_gpu/main.py
 """
+import os
 import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                '..', '..'))
+
 import tempfile
 
 import numpy as np

@@ -7,6 +7,12 @@ Runs on CPU or GPU; random data stands in for MNIST (no network access).
 
 Run:  python examples/mnist/main.py
 """
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                '..', '..'))
+
 import tempfile
 
 import numpy as np
