@@ -251,14 +251,13 @@ class GpuRowGroupDecoder(object):
                                device=dev)
             u_off = np.zeros(n_pages + 1, dtype=np.int64)
             u_off[1:] = np.cumsum(uncomp_size)
-            c_off = np.zeros(n_pages + 1, dtype=np.int64)
-            c_off[:-1] = data_off
-            c_off[-1] = data_off[-1] + comp_size[-1]
             status = torch.zeros(n_pages, dtype=torch.int32, device=dev)
             if ch['compression'] == 'SNAPPY':
                 ext.snappy_decompress_batch(
-                    dbuf, self._up(c_off),
-                    ubuf, self._up(u_off), status)
+                    dbuf, self._up(data_off.astype(np.int64)),
+                    self._up((data_off + comp_size).astype(np.int64)),
+                    ubuf, self._up(u_off[:-1]),
+                    self._up(uncomp_size.astype(np.int64)), status)
                 self._check(status, 'snappy:' + ch['name'])
             else:  # GZIP: each page is one gzip member -> inflate kernel
                 produced = torch.zeros(n_pages, dtype=torch.int64,
@@ -287,9 +286,9 @@ class GpuRowGroupDecoder(object):
         if not data_idx:
             return torch.empty(0, device=dev)
         data_enc = encoding[data_idx[0]]
-        for i in data_idx:
-            if page_type[i] == _PAGE_DATA_V2:
-                return self._cpu_assist_marker(ch['name'])
+        if any(page_type[i] == _PAGE_DATA_V2 for i in data_idx):
+            return self._decode_v2_chunk(ext, dev, dbuf, ch, pages, data_idx,
+                                         n_rows)
 
         max_def = ch['max_def']
         phys = ch['physical']
@@ -396,12 +395,87 @@ class GpuRowGroupDecoder(object):
         out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
                           device=dev)
         status = torch.zeros(len(p_start), dtype=torch.int32, device=dev)
+        empty8 = torch.empty(0, dtype=torch.uint8, device=dev)
+        empty64 = torch.empty(0, dtype=torch.int64, device=dev)
         ext.plain_fixed_decode_batch(
             page_buf, self._up(p_start), self._up(p_end),
             self._up(page_nval.astype(np.int32)), self._up(row0),
             1 if max_def > 0 else 0, esize, self._FILL_PATTERNS[phys],
-            out, torch.empty(0, dtype=torch.uint8, device=dev), status)
+            empty8, empty64, empty64,
+            out, empty8, status)
         self._check(status, 'plainfixed:' + name)
+        return out[:total * esize].view(dtype)
+
+    def _decode_v2_chunk(self, ext, dev, dbuf, ch, pages, data_idx, n_rows):
+        """DataPageV2: levels are stored uncompressed with explicit byte
+        lengths; only the values section is compressed (Parquet format
+        spec).  Supported for PLAIN fixed-width columns; everything else
+        takes the CPU assist."""
+        phys = ch['physical']
+        enc = pages['encoding'].numpy()
+        if phys not in _PHYS_TO_TORCH or \
+                any(enc[i] != _ENC_PLAIN for i in data_idx):
+            return self._cpu_assist_marker(ch['name'])
+        data_off = pages['data_off'].numpy()
+        comp_size = pages['comp_size'].numpy()
+        uncomp_size = pages['uncomp_size'].numpy()
+        num_values = pages['num_values'].numpy()
+        dl = pages['dl_bytes'].numpy()
+        rl = pages['rl_bytes'].numpy()
+        if rl[data_idx].any():
+            return self._cpu_assist_marker(ch['name'])  # nested types
+        idx = np.asarray(data_idx)
+        page_nval = num_values[idx].astype(np.int64)
+        lev_start = (data_off[idx] + rl[idx]).astype(np.int64)
+        lev_len = dl[idx].astype(np.int64)
+        val_comp_start = (data_off[idx] + rl[idx] + dl[idx]).astype(np.int64)
+        compressed = ch['compression'] in ('SNAPPY', 'GZIP')
+        n = len(idx)
+        if compressed:
+            v_un = (uncomp_size[idx] - dl[idx] - rl[idx]).astype(np.int64)
+            v_comp_end = (data_off[idx] + comp_size[idx]).astype(np.int64)
+            u_off = np.zeros(n, dtype=np.int64)
+            u_off[1:] = np.cumsum(v_un)[:-1]
+            vbuf = torch.empty(int(v_un.sum()) + _SLACK, dtype=torch.uint8,
+                               device=dev)
+            status = torch.zeros(n, dtype=torch.int32, device=dev)
+            if ch['compression'] == 'SNAPPY':
+                ext.snappy_decompress_batch(
+                    dbuf, self._up(val_comp_start), self._up(v_comp_end),
+                    vbuf, self._up(u_off), self._up(v_un), status)
+            else:
+                produced = torch.zeros(n, dtype=torch.int64, device=dev)
+                ext.inflate_batch(
+                    dbuf, self._up(val_comp_start),
+                    self._up((v_comp_end - val_comp_start)),
+                    torch.arange(n, dtype=torch.int32, device=dev),
+                    torch.ones(n, dtype=torch.int32, device=dev),
+                    vbuf, self._up(u_off), self._up(v_un), produced, 2,
+                    status)
+            self._check(status, 'v2-decompress:' + ch['name'])
+            val_buf = vbuf
+            val_start = u_off
+            val_end = u_off + v_un
+        else:
+            val_buf = dbuf
+            val_start = val_comp_start
+            val_end = (data_off[idx] + comp_size[idx]).astype(np.int64)
+        row0 = np.zeros(n, dtype=np.int64)
+        row0[1:] = np.cumsum(page_nval)[:-1]
+        dtype, esize = _PHYS_TO_TORCH[phys]
+        total = int(page_nval.sum())
+        out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        status2 = torch.zeros(n, dtype=torch.int32, device=dev)
+        has_def = 2 if ch['max_def'] > 0 else 0
+        empty8 = torch.empty(0, dtype=torch.uint8, device=dev)
+        ext.plain_fixed_decode_batch(
+            val_buf, self._up(val_start), self._up(val_end),
+            self._up(page_nval.astype(np.int32)), self._up(row0),
+            has_def, esize, self._FILL_PATTERNS[phys],
+            dbuf, self._up(lev_start), self._up(lev_len),
+            out, empty8, status2)
+        self._check(status2, 'v2-plainfixed:' + ch['name'])
         return out[:total * esize].view(dtype)
 
     def _plain_byte_array(self, ext, dev, page_buf, host_buf, val_start,

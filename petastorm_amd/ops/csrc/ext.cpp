@@ -10,9 +10,10 @@
 namespace psa {
 
 // snappy.hip
-void snappy_decompress_batch(torch::Tensor comp, torch::Tensor comp_offsets,
-                             torch::Tensor out, torch::Tensor out_offsets,
-                             torch::Tensor status);
+void snappy_decompress_batch(torch::Tensor comp, torch::Tensor comp_start,
+                             torch::Tensor comp_end, torch::Tensor out,
+                             torch::Tensor out_offsets,
+                             torch::Tensor out_len, torch::Tensor status);
 // parquet_decode.hip
 void rle_hybrid_decode_batch(torch::Tensor data, torch::Tensor start,
                              torch::Tensor end, torch::Tensor bit_width,
@@ -29,8 +30,10 @@ void plain_fixed_decode_batch(torch::Tensor page_buf,
                               torch::Tensor payload_start,
                               torch::Tensor payload_end,
                               torch::Tensor n_values, torch::Tensor row0,
-                              int64_t has_def, int64_t esize,
-                              int64_t fill_pattern, torch::Tensor out,
+                              int64_t def_mode, int64_t esize,
+                              int64_t fill_pattern, torch::Tensor def_buf,
+                              torch::Tensor def_start, torch::Tensor def_len,
+                              torch::Tensor out,
                               torch::Tensor valid_out, torch::Tensor status);
 void npy_payload_offsets(torch::Tensor data, torch::Tensor val_off,
                          torch::Tensor val_len, torch::Tensor pay_off,

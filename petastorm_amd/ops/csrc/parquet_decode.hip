@@ -349,13 +349,20 @@ __device__ __forceinline__ void fill_elem(uint8_t* dst, uint64_t pattern,
   for (int b = 0; b < esize; ++b) dst[b] = (uint8_t)(pattern >> (8 * (b & 7)));
 }
 
+// def_mode: 0 = required (no levels); 1 = v1 page (u32 length prefix then
+// RLE levels, then values, all at payload_start); 2 = v2 page (levels live
+// in def_buf[def_start..+def_len] — uncompressed — and payload_start points
+// directly at the values).
 __global__ void plain_fixed_decode_kernel(
     const uint8_t* __restrict__ page_buf,
     const int64_t* __restrict__ payload_start,  // per page
     const int64_t* __restrict__ payload_end,
     const int32_t* __restrict__ n_values,       // rows in page (incl nulls)
     const int64_t* __restrict__ row0,           // first output row of page
-    int32_t has_def, int32_t esize, uint64_t fill_pattern,
+    int32_t def_mode, int32_t esize, uint64_t fill_pattern,
+    const uint8_t* __restrict__ def_buf,        // v2 levels (may == page_buf)
+    const int64_t* __restrict__ def_start_arr,  // v2: per-page level offset
+    const int64_t* __restrict__ def_len_arr,    // v2: per-page level bytes
     uint8_t* __restrict__ out,                  // column base (row-major)
     uint8_t* __restrict__ valid_out,            // [total rows] or nullptr
     int32_t* __restrict__ status, int n_pages) {
@@ -370,7 +377,7 @@ __global__ void plain_fixed_decode_kernel(
   int64_t pos = payload_start[page];
   const int64_t hi = payload_end[page];
 
-  if (!has_def) {
+  if (!def_mode) {
     // required column: straight parallel copy
     const uint8_t* src = page_buf + pos;
     for (int32_t i = lane; i < want; i += PSA_WAVE)
@@ -381,13 +388,25 @@ __global__ void plain_fixed_decode_kernel(
     return;
   }
 
-  // definition levels: u32 length prefix then RLE hybrid at bit width 1
-  uint32_t dl_len = 0;
-  if (lane == 0) dl_len = load_u32_unaligned(page_buf + pos);
-  dl_len = wave_bcast(dl_len);
-  int64_t def_pos = pos + 4;
-  const int64_t def_end = def_pos + dl_len;
-  const uint8_t* values = page_buf + def_end;
+  // definition levels: RLE hybrid at bit width 1.
+  // v1: u32 length prefix at payload start, values follow the levels;
+  // v2: levels live in def_buf at an explicit per-page range.
+  const uint8_t* lev = page_buf;
+  int64_t def_pos, def_end;
+  const uint8_t* values;
+  if (def_mode == 1) {
+    uint32_t dl_len = 0;
+    if (lane == 0) dl_len = load_u32_unaligned(page_buf + pos);
+    dl_len = wave_bcast(dl_len);
+    def_pos = pos + 4;
+    def_end = def_pos + dl_len;
+    values = page_buf + def_end;
+  } else {
+    lev = def_buf;
+    def_pos = def_start_arr[page];
+    def_end = def_pos + def_len_arr[page];
+    values = page_buf + pos;
+  }
 
   int32_t row_cursor = 0;   // rows emitted
   int32_t val_cursor = 0;   // non-null values consumed
@@ -401,7 +420,7 @@ __global__ void plain_fixed_decode_kernel(
       if (def_pos >= def_end) {
         done = 1;
       } else {
-        header = read_varint_u32(page_buf, def_pos, def_end);
+        header = read_varint_u32(lev, def_pos, def_end);
         is_packed = header & 1;
         if (is_packed) {
           run_len = (int32_t)(header >> 1) * 8;
@@ -409,7 +428,7 @@ __global__ void plain_fixed_decode_kernel(
           def_pos += (int64_t)(header >> 1);  // bw=1: one byte per group
         } else {
           run_len = (int32_t)(header >> 1);
-          rle_value = page_buf[def_pos];
+          rle_value = lev[def_pos];
           def_pos += 1;
         }
       }
@@ -444,7 +463,7 @@ __global__ void plain_fixed_decode_kernel(
         int my_bit = 0;
         if (i < emit) {
           int64_t bit = (int64_t)i;  // bit index within this run
-          my_bit = (page_buf[payload + (bit >> 3)] >> (bit & 7)) & 1;
+          my_bit = (lev[payload + (bit >> 3)] >> (bit & 7)) & 1;
         }
         unsigned long long mask = __ballot(my_bit != 0);
         // lanes-below-me mask; lane 63 special-cased ((1ull<<64) is UB)
@@ -474,22 +493,30 @@ void plain_fixed_decode_batch(torch::Tensor page_buf,
                               torch::Tensor payload_start,
                               torch::Tensor payload_end,
                               torch::Tensor n_values, torch::Tensor row0,
-                              int64_t has_def, int64_t esize,
-                              int64_t fill_pattern, torch::Tensor out,
+                              int64_t def_mode, int64_t esize,
+                              int64_t fill_pattern, torch::Tensor def_buf,
+                              torch::Tensor def_start, torch::Tensor def_len,
+                              torch::Tensor out,
                               torch::Tensor valid_out, torch::Tensor status) {
   int n = (int)payload_start.numel();
   if (!n) return;
   const int WPB = 4;
   hipStream_t stream = c10::hip::getCurrentHIPStream();
   uint8_t* vptr = valid_out.numel() ? valid_out.data_ptr<uint8_t>() : nullptr;
+  const uint8_t* dbp = def_buf.numel() ? def_buf.data_ptr<uint8_t>()
+                                       : page_buf.data_ptr<uint8_t>();
+  const int64_t* dsp = def_start.numel() ? def_start.data_ptr<int64_t>()
+                                         : payload_start.data_ptr<int64_t>();
+  const int64_t* dlp = def_len.numel() ? def_len.data_ptr<int64_t>()
+                                       : payload_start.data_ptr<int64_t>();
   hipLaunchKernelGGL(plain_fixed_decode_kernel, dim3((n + WPB - 1) / WPB),
                      dim3(WPB * PSA_WAVE), 0, stream,
                      page_buf.data_ptr<uint8_t>(),
                      payload_start.data_ptr<int64_t>(),
                      payload_end.data_ptr<int64_t>(),
                      n_values.data_ptr<int32_t>(),
-                     row0.data_ptr<int64_t>(), (int32_t)has_def,
-                     (int32_t)esize, (uint64_t)fill_pattern,
+                     row0.data_ptr<int64_t>(), (int32_t)def_mode,
+                     (int32_t)esize, (uint64_t)fill_pattern, dbp, dsp, dlp,
                      out.data_ptr<uint8_t>(), vptr,
                      status.data_ptr<int32_t>(), n);
 }

@@ -44,18 +44,21 @@ __device__ __forceinline__ int64_t read_varint(const uint8_t* p, int64_t& pos,
 }
 
 __global__ void snappy_decompress_kernel(
-    const uint8_t* __restrict__ comp, const int64_t* __restrict__ comp_off,
+    const uint8_t* __restrict__ comp,
+    const int64_t* __restrict__ comp_start,
+    const int64_t* __restrict__ comp_end,
     uint8_t* __restrict__ out, const int64_t* __restrict__ out_off,
+    const int64_t* __restrict__ out_len,
     int32_t* __restrict__ status, int n_pages) {
   const int waves_per_block = blockDim.x / PSA_WAVE;
   const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
   if (page >= n_pages) return;
   const int lane = lane_id();
 
-  const uint8_t* in = comp + comp_off[page];
-  const int64_t in_len = comp_off[page + 1] - comp_off[page];
+  const uint8_t* in = comp + comp_start[page];
+  const int64_t in_len = comp_end[page] - comp_start[page];
   uint8_t* dst_base = out + out_off[page];
-  const int64_t expected = out_off[page + 1] - out_off[page];
+  const int64_t expected = out_len[page];
 
   // lane 0 state, broadcast each op
   int64_t in_pos = 0, out_pos = 0;
@@ -169,13 +172,14 @@ __global__ void snappy_decompress_kernel(
   if (lane == 0 && out_pos != total) status[page] = 3;  // truncated stream
 }
 
-void snappy_decompress_batch(torch::Tensor comp, torch::Tensor comp_offsets,
-                             torch::Tensor out, torch::Tensor out_offsets,
-                             torch::Tensor status) {
+void snappy_decompress_batch(torch::Tensor comp, torch::Tensor comp_start,
+                             torch::Tensor comp_end, torch::Tensor out,
+                             torch::Tensor out_offsets,
+                             torch::Tensor out_len, torch::Tensor status) {
   TORCH_CHECK(comp.is_cuda() && out.is_cuda(), "tensors must be on device");
   TORCH_CHECK(comp.scalar_type() == torch::kUInt8);
-  TORCH_CHECK(comp_offsets.scalar_type() == torch::kInt64);
-  int n_pages = (int)comp_offsets.numel() - 1;
+  TORCH_CHECK(comp_start.scalar_type() == torch::kInt64);
+  int n_pages = (int)comp_start.numel();
   if (n_pages <= 0) return;
   const int WPB = 4;  // waves per block
   int blocks = (n_pages + WPB - 1) / WPB;
@@ -183,9 +187,11 @@ void snappy_decompress_batch(torch::Tensor comp, torch::Tensor comp_offsets,
   hipLaunchKernelGGL(snappy_decompress_kernel, dim3(blocks),
                      dim3(WPB * PSA_WAVE), 0, stream,
                      comp.data_ptr<uint8_t>(),
-                     comp_offsets.data_ptr<int64_t>(),
+                     comp_start.data_ptr<int64_t>(),
+                     comp_end.data_ptr<int64_t>(),
                      out.data_ptr<uint8_t>(),
                      out_offsets.data_ptr<int64_t>(),
+                     out_len.data_ptr<int64_t>(),
                      status.data_ptr<int32_t>(), n_pages);
 }
 

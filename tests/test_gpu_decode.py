@@ -49,8 +49,12 @@ def test_snappy_roundtrip_random_and_repetitive(ext):
                               dtype=torch.uint8).to(dev)
     out = torch.zeros(int(u_off[-1]) + 16, dtype=torch.uint8, device=dev)
     status = torch.zeros(len(payloads), dtype=torch.int32, device=dev)
-    ext.snappy_decompress_batch(comp_t, torch.from_numpy(c_off).to(dev),
-                                out, torch.from_numpy(u_off).to(dev), status)
+    ext.snappy_decompress_batch(comp_t,
+                                torch.from_numpy(c_off[:-1]).to(dev),
+                                torch.from_numpy(c_off[1:]).to(dev),
+                                out, torch.from_numpy(u_off[:-1]).to(dev),
+                                torch.from_numpy(np.diff(u_off)).to(dev),
+                                status)
     torch.cuda.synchronize()
     assert status.cpu().tolist() == [0] * len(payloads)
     got = out[:int(u_off[-1])].cpu().numpy().tobytes()
@@ -705,3 +709,36 @@ def test_gzip_rowgroup_decode(ext, tmp_path):
         ids = torch.cat([b.id for b in r]).cpu().numpy()
         assert not r.diagnostics['cpu_assist_columns']
     np.testing.assert_array_equal(np.sort(ids), np.arange(3000))
+
+
+@pytest.mark.parametrize('compression', ['snappy', 'none', 'gzip'])
+def test_datapage_v2_decode(ext, tmp_path, compression):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd.gpu.decoder import GpuRowGroupDecoder
+    from petastorm_amd.unischema import Unischema
+    rng = np.random.RandomState(0)
+    vals = rng.rand(5000)
+    mask = rng.rand(5000) < 0.2
+    col = pa.array([None if m else float(v) for m, v in zip(mask, vals)])
+    path = str(tmp_path / ('v2_' + compression + '.parquet'))
+    pq.write_table(pa.table({'id': pa.array(np.arange(5000)), 'x': col}),
+                   path, compression=compression, use_dictionary=False,
+                   row_group_size=2000, data_page_version='2.0')
+    pf = pq.ParquetFile(path)
+    schema = Unischema.from_arrow_schema(pf.schema_arrow)
+    dec = GpuRowGroupDecoder('cuda')
+    for rg in range(pf.metadata.num_row_groups):
+        host, meta = dec.read_rowgroup_bytes(path, pf.metadata, pf.schema,
+                                             rg, ['id', 'x'])
+        out, _ = dec.decode(host, meta, schema)
+        dec.flush_status()
+        assert not dec.cpu_assist_columns, dec.cpu_assist_columns
+        oracle = pf.read_row_group(rg, columns=['id', 'x'])
+        got_x = out['x'].cpu().numpy()
+        exp_x = oracle.column('x').to_numpy(zero_copy_only=False)
+        np.testing.assert_array_equal(np.isnan(got_x), np.isnan(exp_x))
+        np.testing.assert_allclose(got_x[~np.isnan(got_x)],
+                                   exp_x[~np.isnan(exp_x)])
+        np.testing.assert_array_equal(out['id'].cpu().numpy(),
+                                      oracle.column('id').to_numpy())
