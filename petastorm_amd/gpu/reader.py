@@ -43,6 +43,26 @@ logger = logging.getLogger(__name__)
 _PREFETCH_DEPTH = 2
 
 
+class _TraceRange(object):
+    """rocTX range (torch.cuda.nvtx maps to roctx on ROCm) for rocprofv3
+    --marker-trace; enabled with PSA_TRACE=1 (replaces the
+    reference's per-worker cProfile as the pipeline-stage tracer)."""
+
+    enabled = os.environ.get('PSA_TRACE') == '1'
+
+    def __init__(self, name):
+        self._name = name
+
+    def __enter__(self):
+        if self.enabled:
+            torch.cuda.nvtx.range_push(self._name)
+        return self
+
+    def __exit__(self, *exc):
+        if self.enabled:
+            torch.cuda.nvtx.range_pop()
+
+
 class _PinnedPool(object):
     """Reusable pinned host buffers (rounded up to 1 MiB steps)."""
 
@@ -246,7 +266,8 @@ class GpuBatchReader(object):
                         self._inflight_hosts.append((ev, host))
                 self._reclaim_hosts()
                 t2 = time.perf_counter()
-                batch = self._postprocess(piece, columns)
+                with _TraceRange('psa.postprocess'):
+                    batch = self._postprocess(piece, columns)
                 self.stage_times['postprocess'] += time.perf_counter() - t2
                 dispatched += 1
                 if pending is not None:
@@ -272,8 +293,9 @@ class GpuBatchReader(object):
     # ------------------------------------------------------------------
     def _decode_piece(self, piece, host, meta, plan=None):
         t0 = time.perf_counter()
-        raw, dbuf = self._decoder.decode(host, meta, self._storage_schema,
-                                         plan)
+        with _TraceRange('psa.decode_rowgroup'):
+            raw, dbuf = self._decoder.decode(host, meta,
+                                             self._storage_schema, plan)
         self.stage_times['decode'] += time.perf_counter() - t0
         t0 = time.perf_counter()
         columns = {}

@@ -39,20 +39,51 @@ __global__ void nhwc_to_nchw_normalize_kernel(
   if (npix <= 0) return;
 
   const uint8_t* src = in + (n * hw + pix0) * c;
-  // coalesced byte read: thread t reads bytes t, t+B, ... of the tile
   const int64_t nbytes = npix * c;
-  for (int64_t i = threadIdx.x; i < nbytes; i += blockDim.x) {
-    int ch = (int)(i % c);
-    int64_t p = i / c;
-    tile[ch][p] = (float)src[i];
+  // vectorized read: 4 bytes per lane per iteration (guide G13 — scalar
+  // byte loads run ~2.5x slower); falls back to bytes when the tile base
+  // is not 4-aligned (odd H*W*C images)
+  if (((uintptr_t)src & 3) == 0) {
+    const uint32_t* src4 = (const uint32_t*)src;
+    const int64_t nwords = nbytes >> 2;
+    for (int64_t w = threadIdx.x; w < nwords; w += blockDim.x) {
+      uint32_t v = src4[w];
+      int64_t i = w << 2;
+#pragma unroll
+      for (int b = 0; b < 4; ++b) {
+        int64_t idx = i + b;
+        tile[(int)(idx % c)][idx / c] = (float)((v >> (8 * b)) & 0xFF);
+      }
+    }
+    for (int64_t i = (nwords << 2) + threadIdx.x; i < nbytes;
+         i += blockDim.x)
+      tile[(int)(i % c)][i / c] = (float)src[i];
+  } else {
+    for (int64_t i = threadIdx.x; i < nbytes; i += blockDim.x)
+      tile[(int)(i % c)][i / c] = (float)src[i];
   }
   __syncthreads();
   for (int ch = 0; ch < c; ++ch) {
     const float m = mean[ch], is = inv_std[ch];
     OutT* dst = out + (n * c + ch) * hw + pix0;
-    for (int64_t p = threadIdx.x; p < npix; p += blockDim.x) {
-      float v = (tile[ch][p] * scale - m) * is;
-      dst[p] = (OutT)v;
+    // vectorized write: 4 outputs per lane per iteration
+    const int64_t nv = npix & ~(int64_t)3;
+    if ((((uintptr_t)dst) & 15) == 0 && sizeof(OutT) == 4) {
+      float4* dst4 = (float4*)dst;
+      for (int64_t p4 = threadIdx.x; p4 < (nv >> 2); p4 += blockDim.x) {
+        int64_t p = p4 << 2;
+        float4 v;
+        v.x = (tile[ch][p + 0] * scale - m) * is;
+        v.y = (tile[ch][p + 1] * scale - m) * is;
+        v.z = (tile[ch][p + 2] * scale - m) * is;
+        v.w = (tile[ch][p + 3] * scale - m) * is;
+        dst4[p4] = v;
+      }
+      for (int64_t p = nv + threadIdx.x; p < npix; p += blockDim.x)
+        dst[p] = (OutT)((tile[ch][p] * scale - m) * is);
+    } else {
+      for (int64_t p = threadIdx.x; p < npix; p += blockDim.x)
+        dst[p] = (OutT)((tile[ch][p] * scale - m) * is);
     }
   }
 }
