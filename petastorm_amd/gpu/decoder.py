@@ -379,6 +379,12 @@ class GpuRowGroupDecoder(object):
                                     else comp_size, val_start, val_end,
                                     page_nval, nonnull_per_page, valid,
                                     n_rows, phys)
+        if data_enc in (_ENC_PLAIN_DICT, _ENC_RLE_DICT) and \
+                phys == 'BYTE_ARRAY' and dict_idx and valid is None:
+            return self._dict_byte_array(ext, dev, page_buf, page_start,
+                                         dict_idx, num_values,
+                                         uncomp_size if snappy else comp_size,
+                                         val_start, val_end, page_nval, ch)
         return self._cpu_assist_marker(ch['name'])
 
     # ------------------------------------------------------------------
@@ -519,6 +525,48 @@ class GpuRowGroupDecoder(object):
         col.jpeg_meta = plan.get('jpeg_meta')
         col.png_meta = plan.get('png_meta')
         return col
+
+    def _dict_byte_array(self, ext, dev, page_buf, page_start, dict_idx,
+                         num_values, size_arr, val_start, val_end, page_nval,
+                         ch):
+        """Dictionary-encoded binary column: decode the dictionary page's
+        value offsets once, decode the RLE indices, and gather per-row
+        (offset, length) — blob bytes are never copied.
+
+        The column comes back as a ByteArrayColumn WITHOUT host visibility
+        (indices only exist on device), so image codecs needing host header
+        parsing fall back to CPU; the ndarray codecs decode fully on-GPU.
+        """
+        di = dict_idx[0]
+        dict_n = int(num_values[di])
+        dstart = int(page_start[di])
+        dend = dstart + int(size_arr[di])
+        d_off = torch.empty(dict_n, dtype=torch.int64, device=dev)
+        d_len = torch.empty(dict_n, dtype=torch.int32, device=dev)
+        status = torch.zeros(1, dtype=torch.int32, device=dev)
+        ext.byte_array_offsets_batch(
+            page_buf, self._up(np.array([dstart], dtype=np.int64)),
+            self._up(np.array([dend], dtype=np.int64)),
+            self._up(np.array([dict_n], dtype=np.int32)),
+            self._up(np.array([0], dtype=np.int64)), d_off, d_len, status)
+        self._check(status, 'dictba-dict:' + ch['name'])
+
+        counts = page_nval
+        bw_t = torch.stack([page_buf[int(s)] for s in val_start]).cpu()
+        bw = bw_t.numpy().astype(np.int32)
+        i_off = np.zeros(len(counts) + 1, dtype=np.int64)
+        i_off[1:] = np.cumsum(counts)
+        indices = torch.empty(int(i_off[-1]), dtype=torch.int32, device=dev)
+        st2 = torch.zeros(len(counts), dtype=torch.int32, device=dev)
+        ext.rle_hybrid_decode_batch(
+            page_buf, self._up(val_start + 1), self._up(val_end),
+            self._up(bw), self._up(counts.astype(np.int32)),
+            self._up(i_off[:-1]), indices, st2)
+        self._check(st2, 'dictba-idx:' + ch['name'])
+        idx = indices.long()
+        return ByteArrayColumn(page_buf, d_off.index_select(0, idx),
+                               d_len.index_select(0, idx), None, None,
+                               int(i_off[-1]))
 
     def _dict_fixed(self, ext, dev, page_buf, page_start, dict_idx,
                     num_values, size_arr, val_start, val_end, page_nval,
