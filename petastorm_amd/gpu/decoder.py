@@ -75,19 +75,30 @@ class GpuRowGroupDecoder(object):
         # one per kernel
         self._pending_status = []
         self._pin_memory = torch.cuda.is_available()
+        # pinned staging buffers, recycled once flush_status() has proven
+        # their async H2D copies complete (hipHostMalloc per upload costs
+        # ~1ms; recycling makes _up() allocation-free in steady state)
+        self._staging_free = {}
+        self._staging_inuse = []
 
     def _up(self, arr):
         """Async host->device upload of a small numpy array / cpu tensor.
 
         A plain ``torch.from_numpy(x).to(dev)`` on unpinned memory is a
         BLOCKING copy that synchronizes the stream — profiled at >1ms per
-        call and the dominant cost of the scalar config.  Staging through
-        torch's cached pinned allocator keeps every upload asynchronous.
+        call and the dominant cost of the scalar config.  Staging goes
+        through a recycled pinned buffer so every upload is asynchronous
+        and allocation-free.
         """
         t = arr if isinstance(arr, torch.Tensor) else torch.from_numpy(arr)
         if self._pin_memory and not t.is_pinned():
-            pinned = torch.empty_like(t, pin_memory=True)
+            key = (t.dtype, t.numel())
+            free = self._staging_free.get(key)
+            pinned = free.pop() if free else torch.empty(
+                t.shape, dtype=t.dtype, pin_memory=True)
+            pinned = pinned.view(t.shape)
             pinned.copy_(t)
+            self._staging_inuse.append((key, pinned))
             t = pinned
         return t.to(self.device, non_blocking=True)
 
@@ -435,30 +446,45 @@ class GpuRowGroupDecoder(object):
         lev_start = (data_off[idx] + rl[idx]).astype(np.int64)
         lev_len = dl[idx].astype(np.int64)
         val_comp_start = (data_off[idx] + rl[idx] + dl[idx]).astype(np.int64)
-        compressed = ch['compression'] in ('SNAPPY', 'GZIP')
+        # V2 compresses only the values section, and a page may individually
+        # opt out via is_compressed (writers skip compression when it does
+        # not shrink the page)
+        v2c = pages['v2_is_compressed'].numpy()[idx].astype(bool)
+        chunk_compressed = ch['compression'] in ('SNAPPY', 'GZIP')
+        page_compressed = v2c & chunk_compressed
         n = len(idx)
-        if compressed:
+        if page_compressed.any():
             v_un = (uncomp_size[idx] - dl[idx] - rl[idx]).astype(np.int64)
             v_comp_end = (data_off[idx] + comp_size[idx]).astype(np.int64)
             u_off = np.zeros(n, dtype=np.int64)
             u_off[1:] = np.cumsum(v_un)[:-1]
             vbuf = torch.empty(int(v_un.sum()) + _SLACK, dtype=torch.uint8,
                                device=dev)
-            status = torch.zeros(n, dtype=torch.int32, device=dev)
+            ci = np.nonzero(page_compressed)[0]
+            status = torch.zeros(len(ci), dtype=torch.int32, device=dev)
             if ch['compression'] == 'SNAPPY':
                 ext.snappy_decompress_batch(
-                    dbuf, self._up(val_comp_start), self._up(v_comp_end),
-                    vbuf, self._up(u_off), self._up(v_un), status)
+                    dbuf, self._up(val_comp_start[ci]),
+                    self._up(v_comp_end[ci]),
+                    vbuf, self._up(u_off[ci]), self._up(v_un[ci]), status)
             else:
-                produced = torch.zeros(n, dtype=torch.int64, device=dev)
+                produced = torch.zeros(len(ci), dtype=torch.int64,
+                                       device=dev)
                 ext.inflate_batch(
-                    dbuf, self._up(val_comp_start),
-                    self._up((v_comp_end - val_comp_start)),
-                    torch.arange(n, dtype=torch.int32, device=dev),
-                    torch.ones(n, dtype=torch.int32, device=dev),
-                    vbuf, self._up(u_off), self._up(v_un), produced, 2,
-                    status)
+                    dbuf, self._up(val_comp_start[ci]),
+                    self._up((v_comp_end - val_comp_start)[ci]),
+                    torch.arange(len(ci), dtype=torch.int32, device=dev),
+                    torch.ones(len(ci), dtype=torch.int32, device=dev),
+                    vbuf, self._up(u_off[ci]), self._up(v_un[ci]), produced,
+                    2, status)
             self._check(status, 'v2-decompress:' + ch['name'])
+            ri = np.nonzero(~page_compressed)[0]
+            if len(ri):
+                # raw pages inside a compressed chunk: copy bytes into their
+                # vbuf slots so the decode kernel sees one buffer
+                ext.varlen_gather(dbuf, self._up(val_comp_start[ri]),
+                                  self._up(v_un[ri]), vbuf,
+                                  self._up(u_off[ri]))
             val_buf = vbuf
             val_start = u_off
             val_end = u_off + v_un
@@ -617,18 +643,30 @@ class GpuRowGroupDecoder(object):
         self._pending_status.append((what, status))
 
     def flush_status(self):
-        """One sync: verify every queued kernel status is clean."""
+        """One sync: verify every queued kernel status is clean.  The sync
+        also proves all prior async uploads landed, so the pinned staging
+        buffers go back to the free pool here."""
         if not self._pending_status:
+            if self._staging_inuse:
+                torch.cuda.synchronize(self.device)
+                self._recycle_staging()
             return
         pending, self._pending_status = self._pending_status, []
         total = torch.stack([s.abs().sum() for _, s in pending]).sum()
-        if int(total.item()) != 0:
+        bad = int(total.item()) != 0
+        self._recycle_staging()
+        if bad:
             for what, s in pending:
                 vals = s.cpu()
                 if int(vals.abs().sum()) != 0:
                     raise RuntimeError(
                         'GPU decode error in {}: status={}'
                         .format(what, vals.tolist()))
+
+    def _recycle_staging(self):
+        for key, buf in self._staging_inuse:
+            self._staging_free.setdefault(key, []).append(buf)
+        self._staging_inuse = []
 
     # ------------------------------------------------------------------
     # codec stages over ByteArrayColumn

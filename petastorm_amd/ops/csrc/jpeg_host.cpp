@@ -119,7 +119,15 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
   std::vector<int32_t> seg_img;
   std::vector<int64_t> seg_pos, seg_end;
   std::vector<int32_t> seg_mcu0, seg_nmcu;
+  seg_img.reserve(n * 64);
+  seg_pos.reserve(n * 64);
+  seg_end.reserve(n * 64);
+  seg_mcu0.reserve(n * 64);
+  seg_nmcu.reserve(n * 64);
 
+  // the scan below is pure C++ over the pinned buffer: release the GIL so
+  // the IO prefetch thread's parsing overlaps python work on other threads
+  py::gil_scoped_release nogil;
   for (int64_t i = 0; i < n; ++i) {
     const uint8_t* p = base + off[i];
     const int64_t len = vlen(i);
@@ -298,7 +306,8 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
     }
   }
 
-  // ---- flatten to tensors ----
+  // ---- flatten to tensors (GIL re-acquired for python object creation) ----
+  py::gil_scoped_acquire gil;
   auto i32 = torch::TensorOptions().dtype(torch::kInt32);
   auto i64 = torch::TensorOptions().dtype(torch::kInt64);
   auto f32 = torch::TensorOptions().dtype(torch::kFloat32);

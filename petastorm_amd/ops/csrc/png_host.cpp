@@ -134,15 +134,19 @@ py::dict byte_array_host_offsets(torch::Tensor buf, torch::Tensor val_start,
   int64_t* o = off.data_ptr<int64_t>();
   int64_t* l = len.data_ptr<int64_t>();
   int64_t k = 0;
-  for (int64_t j = 0; j < n_pages; ++j) {
-    int64_t pos = starts[j];
-    for (int64_t v = 0; v < cnt[j]; ++v) {
-      uint32_t ln;
-      std::memcpy(&ln, base + pos, 4);
-      o[k] = pos + 4;
-      l[k] = ln;
-      ++k;
-      pos += 4 + ln;
+  {
+    // pure C++ scan: release the GIL so IO-thread parsing overlaps python
+    py::gil_scoped_release nogil;
+    for (int64_t j = 0; j < n_pages; ++j) {
+      int64_t pos = starts[j];
+      for (int64_t v = 0; v < cnt[j]; ++v) {
+        uint32_t ln;
+        std::memcpy(&ln, base + pos, 4);
+        o[k] = pos + 4;
+        l[k] = ln;
+        ++k;
+        pos += 4 + ln;
+      }
     }
   }
   py::dict out;

@@ -199,7 +199,7 @@ py::dict parquet_walk_pages(torch::Tensor buf, torch::Tensor chunk_off,
   const int64_t nchunks = chunk_off.numel();
 
   std::vector<int64_t> p_chunk, p_type, p_off, p_comp, p_uncomp, p_nval,
-      p_enc, p_denc, p_dl, p_rl;
+      p_enc, p_denc, p_dl, p_rl, p_v2c;
   for (int64_t ci = 0; ci < nchunks; ++ci) {
     Cursor c{base, coff[ci], coff[ci] + clen[ci]};
     while (c.ok && c.pos < c.end) {
@@ -218,6 +218,7 @@ py::dict parquet_walk_pages(torch::Tensor buf, torch::Tensor chunk_off,
       p_denc.push_back(h.def_encoding);
       p_dl.push_back(h.dl_bytes);
       p_rl.push_back(h.rl_bytes);
+      p_v2c.push_back(h.v2_is_compressed ? 1 : 0);
       c.skip_bytes(h.compressed);
     }
   }
@@ -238,6 +239,7 @@ py::dict parquet_walk_pages(torch::Tensor buf, torch::Tensor chunk_off,
   out["def_encoding"] = mk(p_denc);
   out["dl_bytes"] = mk(p_dl);
   out["rl_bytes"] = mk(p_rl);
+  out["v2_is_compressed"] = mk(p_v2c);
   return out;
 }
 
