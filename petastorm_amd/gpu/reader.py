@@ -146,7 +146,7 @@ class GpuBatchReader(object):
         self._timing_enabled = os.environ.get('PSA_TIMING') == '1'
         self.stage_times = {'io_wait': 0.0, 'decode': 0.0, 'codec': 0.0,
                             'postprocess': 0.0, 'io_read': 0.0,
-                            'io_parse': 0.0}
+                            'io_parse': 0.0, 'flush': 0.0}
         self._gen = self._generate()
 
     # ------------------------------------------------------------------
@@ -277,7 +277,9 @@ class GpuBatchReader(object):
                     yield nt
                 if batch is None:
                     continue
+                t3 = time.perf_counter()
                 self._decoder.flush_status()
+                self.stage_times['flush'] += time.perf_counter() - t3
                 self._rows_epoch += len(next(iter(batch.values())))
                 # cursor value once THIS batch is consumed = pieces
                 # dispatched up to and including it
