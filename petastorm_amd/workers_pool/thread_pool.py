@@ -90,6 +90,10 @@ class ThreadPool(object):
         self._ventilated = 0
         self._processed = 0
         self._count_lock = threading.Lock()
+        # serializes concurrent consumers (reference supports multiple
+        # threads calling next(reader) on one Reader,
+        # tests/test_end_to_end.py:868-877)
+        self._consumer_lock = threading.Lock()
         self._started = False
 
     # ------------------------------------------------------------------
@@ -145,18 +149,19 @@ class ThreadPool(object):
                 raise EmptyResultError('Pool was stopped')
             if self._all_done():
                 raise EmptyResultError('No more work')
-            q = self._results_queues[self._rr_read]
-            try:
-                msg = q.get(timeout=_POLL_S)
-            except queue.Empty:
-                continue
-            if isinstance(msg, VentilatedItemProcessedMessage):
-                with self._count_lock:
-                    self._processed += 1
-                if self._ventilator is not None:
-                    self._ventilator.processed_item()
-                self._rr_read = (self._rr_read + 1) % self.workers_count
-                continue
+            with self._consumer_lock:
+                q = self._results_queues[self._rr_read]
+                try:
+                    msg = q.get(timeout=_POLL_S)
+                except queue.Empty:
+                    continue
+                if isinstance(msg, VentilatedItemProcessedMessage):
+                    with self._count_lock:
+                        self._processed += 1
+                    if self._ventilator is not None:
+                        self._ventilator.processed_item()
+                    self._rr_read = (self._rr_read + 1) % self.workers_count
+                    continue
             if isinstance(msg, WorkerExceptionMessage):
                 self.stop()
                 self.join()

@@ -275,3 +275,50 @@ def test_context_manager_stops_pool(test_dataset):
     # after exit, iteration raises StopIteration
     with pytest.raises(StopIteration):
         next(r)
+
+
+def test_concurrent_reads_from_one_reader(test_dataset):
+    """reference tests/test_end_to_end.py:868-877"""
+    import threading
+    results = []
+    lock = threading.Lock()
+    with make_reader(test_dataset['url'], reader_pool_type='thread',
+                     workers_count=3, shuffle_row_groups=False) as r:
+        def consume():
+            while True:
+                try:
+                    row = next(r)
+                except StopIteration:
+                    return
+                with lock:
+                    results.append(int(row.id))
+        threads = [threading.Thread(target=consume) for _ in range(3)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=60)
+    assert sorted(results) == sorted(int(s['id'])
+                                     for s in test_dataset['rows'])
+
+
+def test_infinite_epochs_and_manual_stop(test_dataset):
+    reader = make_reader(test_dataset['url'], reader_pool_type='thread',
+                         num_epochs=None, shuffle_row_groups=False)
+    n_rows = len(test_dataset['rows'])
+    rows = [next(reader) for _ in range(3 * n_rows)]  # several epochs
+    assert len(rows) == 3 * n_rows
+    reader.stop()
+    reader.join()
+    with pytest.raises(StopIteration):
+        next(reader)
+
+
+def test_unicode_dataset_path(tmp_path):
+    from petastorm_amd.test_util.dataset_gen import create_test_dataset
+    path = tmp_path / 'уникод-データ'
+    url = 'file://' + str(path)
+    rows = create_test_dataset(url, num_rows=10, num_files=1)
+    with make_reader(url, reader_pool_type='dummy',
+                     shuffle_row_groups=False) as r:
+        got = list(r)
+    assert len(got) == len(rows)
