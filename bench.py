@@ -82,10 +82,12 @@ def bench_imagenet(args, rank, world, device, dist):
 
     # >= 8 row-groups are required so every rank of an 8-GPU run gets data
     # (sharding is per row-group, reference reader.py:573-597)
+    # uniform 256-row row groups, 24 per 6144 rows: shards stay equal-sized
+    # so the RCCL epoch collectives stay in lock-step across ranks
     n_rows = args.rows or 6144
     url = _dataset_dir('imagenet_{}'.format(n_rows), rank, dist,
                        lambda u: create_imagenet_dataset(
-                           u, num_rows=n_rows, rowgroup_size_mb=16))
+                           u, num_rows=n_rows, rows_per_rowgroup=256))
 
     ext = ops.ext()
     mean = torch.tensor([0.485, 0.456, 0.406], device=device)
@@ -144,7 +146,7 @@ def bench_scalar(args, rank, world, device, dist):
     n_rows = args.rows or 2_000_000
     url = _dataset_dir('scalar_{}'.format(n_rows), rank, dist,
                        lambda u: create_scalar_dataset(
-                           u, num_rows=n_rows, rowgroup_size=65536))
+                           u, num_rows=n_rows, rowgroup_size=62500))
     reader = make_batch_reader(
         url, device=str(device), num_epochs=None, shuffle_row_groups=True,
         seed=7, schema_fields=['id', 'f0', 'f1', 'f2', 'f3', 'i0', 'i1'],
@@ -220,7 +222,7 @@ def bench_ngram(args, rank, world, device, dist):
     n_rows = args.rows or 100_000
     url = _dataset_dir('seq_{}'.format(n_rows), rank, dist,
                        lambda u: create_sequence_dataset(
-                           u, num_rows=n_rows, rowgroup_size_mb=32))
+                           u, num_rows=n_rows, rows_per_rowgroup=6250))
     pred = in_lambda(['source'], lambda v: v['source'] != 3)  # keep 3/4
     reader = make_batch_reader(
         url, device=str(device), num_epochs=None, shuffle_row_groups=True,

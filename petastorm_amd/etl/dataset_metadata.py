@@ -66,11 +66,13 @@ class DatasetWriter(object):
     """
 
     def __init__(self, fs, path, schema, rowgroup_size_mb=DEFAULT_ROWGROUP_SIZE_MB,
-                 compression='snappy', file_prefix='part'):
+                 compression='snappy', file_prefix='part',
+                 rows_per_rowgroup=None):
         self._fs = fs
         self._path = path
         self._schema = schema
         self._rowgroup_bytes = int(rowgroup_size_mb * (1 << 20))
+        self._rows_per_rowgroup = rows_per_rowgroup
         self._compression = compression
         self._file_prefix = file_prefix
 
@@ -88,7 +90,12 @@ class DatasetWriter(object):
         encoded = dict_to_encoded_row(self._schema, row_dict)
         self._buffer.append(encoded)
         self._buffer_bytes += self._estimate_row_bytes(encoded)
-        if self._buffer_bytes >= self._rowgroup_bytes:
+        if self._rows_per_rowgroup is not None:
+            # exact-rows mode: uniform row groups (equal shards for
+            # distributed readers; see parallel/epochs.py ordering contract)
+            if len(self._buffer) >= self._rows_per_rowgroup:
+                self._flush_row_group()
+        elif self._buffer_bytes >= self._rowgroup_bytes:
             self._flush_row_group()
 
     def write_rows(self, row_dicts):
@@ -166,7 +173,8 @@ class DatasetWriter(object):
 @contextmanager
 def materialize_dataset(dataset_url, schema,
                         rowgroup_size_mb=DEFAULT_ROWGROUP_SIZE_MB,
-                        compression='snappy', filesystem_factory=None):
+                        compression='snappy', filesystem_factory=None,
+                        rows_per_rowgroup=None):
     """Write a dataset: ``with materialize_dataset(url, schema) as writer: ...``
 
     On exit, schema metadata is persisted next to the data (reference
@@ -174,7 +182,8 @@ def materialize_dataset(dataset_url, schema,
     """
     fs, path = get_filesystem_and_path_or_paths(dataset_url)
     fs.makedirs(path, exist_ok=True)
-    writer = DatasetWriter(fs, path, schema, rowgroup_size_mb, compression)
+    writer = DatasetWriter(fs, path, schema, rowgroup_size_mb, compression,
+                           rows_per_rowgroup=rows_per_rowgroup)
     yield writer
     writer.close()
     _write_dataset_metadata(fs, path, schema)

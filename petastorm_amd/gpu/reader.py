@@ -162,6 +162,14 @@ class GpuBatchReader(object):
         perm = epoch_sync.epoch_permutation(
             n, epoch, self._seed, self._shuffle_row_groups)
         if self._shard_count is not None:
+            if epoch_sync._dist() is not None:
+                # With a live process group every rank runs the SAME number
+                # of epoch-boundary collectives, so shards must be equal:
+                # drop the remainder row-groups (DistributedSampler-style;
+                # which groups are dropped rotates with the epoch
+                # permutation, so coverage evens out across epochs).
+                n_even = n - (n % self._shard_count)
+                perm = perm[:n_even]
             perm = [p for pos, p in enumerate(perm)
                     if pos % self._shard_count == self._cur_shard]
         return [self._pieces[i] for i in perm]

@@ -15,6 +15,12 @@ with an explicit collective when a process group exists:
 Every function degrades gracefully to single-process semantics when
 torch.distributed is not initialized, and runs on the gloo backend for
 CPU-only tests (world_size > 1 multi-process tests run here without a GPU).
+
+Collective-ordering contract: every rank must execute the same sequence of
+collectives.  GpuBatchReader guarantees it by dropping the per-epoch
+row-group remainder so all shards are equal-sized (readers on different
+ranks then hit epoch_permutation / epoch_end_sync in lock-step regardless of
+relative progress within an epoch).
 """
 
 import numpy as np

@@ -126,14 +126,15 @@ ImageNetSchema = Unischema('ImageNetSchema', [
 
 
 def create_imagenet_dataset(url, num_rows=512, rowgroup_size_mb=32, seed=0,
-                            structured=True):
+                            structured=True, rows_per_rowgroup=None):
     """Random 224x224x3 jpegs + labels.
 
     ``structured`` images (smooth gradients + blobs) compress like photos;
     pure-noise images stress the Huffman decoder instead. Both are valid.
     """
     rng = np.random.RandomState(seed)
-    with materialize_dataset(url, ImageNetSchema, rowgroup_size_mb) as w:
+    with materialize_dataset(url, ImageNetSchema, rowgroup_size_mb,
+                             rows_per_rowgroup=rows_per_rowgroup) as w:
         for i in range(num_rows):
             if structured:
                 yy, xx = np.mgrid[0:224, 0:224].astype(np.float32)
@@ -186,9 +187,11 @@ SequenceSchema = Unischema('SequenceSchema', [
 ])
 
 
-def create_sequence_dataset(url, num_rows=200, rowgroup_size_mb=4, seed=0):
+def create_sequence_dataset(url, num_rows=200, rowgroup_size_mb=4, seed=0,
+                            rows_per_rowgroup=None):
     rng = np.random.RandomState(seed)
-    with materialize_dataset(url, SequenceSchema, rowgroup_size_mb) as w:
+    with materialize_dataset(url, SequenceSchema, rowgroup_size_mb,
+                             rows_per_rowgroup=rows_per_rowgroup) as w:
         for i in range(num_rows):
             w.write_row({
                 'timestamp': np.int64(i),
