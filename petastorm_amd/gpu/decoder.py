@@ -254,6 +254,16 @@ class GpuRowGroupDecoder(object):
         n_pages = len(page_type)
         snappy = ch['compression'] in ('SNAPPY', 'GZIP')
 
+        # V2 pages compress ONLY the values section, so they must never go
+        # through the whole-page decompression below — dispatch first
+        data_idx_early = [i for i in range(n_pages)
+                          if page_type[i] in (_PAGE_DATA_V1, _PAGE_DATA_V2)]
+        if not data_idx_early:
+            return torch.empty(0, device=dev)
+        if any(page_type[i] == _PAGE_DATA_V2 for i in data_idx_early):
+            return self._decode_v2_chunk(ext, dev, dbuf, ch, pages,
+                                         data_idx_early, n_rows)
+
         # 1) page payload location: either in dbuf directly, or in a
         #    decompressed scratch buffer
         if snappy:
@@ -297,9 +307,6 @@ class GpuRowGroupDecoder(object):
         if not data_idx:
             return torch.empty(0, device=dev)
         data_enc = encoding[data_idx[0]]
-        if any(page_type[i] == _PAGE_DATA_V2 for i in data_idx):
-            return self._decode_v2_chunk(ext, dev, dbuf, ch, pages, data_idx,
-                                         n_rows)
 
         max_def = ch['max_def']
         phys = ch['physical']
@@ -390,8 +397,11 @@ class GpuRowGroupDecoder(object):
                                     else comp_size, val_start, val_end,
                                     page_nval, nonnull_per_page, valid,
                                     n_rows, phys)
+        all_valid = valid is None or (
+            nonnull_per_page is not None and
+            bool((nonnull_per_page == page_nval).all()))
         if data_enc in (_ENC_PLAIN_DICT, _ENC_RLE_DICT) and \
-                phys == 'BYTE_ARRAY' and dict_idx and valid is None:
+                phys == 'BYTE_ARRAY' and dict_idx and all_valid:
             return self._dict_byte_array(ext, dev, page_buf, page_start,
                                          dict_idx, num_values,
                                          uncomp_size if snappy else comp_size,
