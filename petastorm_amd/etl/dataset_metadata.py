@@ -139,7 +139,10 @@ class DatasetWriter(object):
                 compression=self._column_compression(),
                 # one table write == one row group:
                 use_dictionary=False, write_statistics=True,
-                data_page_size=1 << 20)
+                # 256 KiB pages: the page is the GPU decode-parallelism unit
+                # (one wave per page), so smaller-than-arrow-default pages
+                # keep the 256-CU chip fed on few-column datasets
+                data_page_size=256 << 10)
         self._writer.write_table(table)
         self._buffer = []
         self._buffer_bytes = 0

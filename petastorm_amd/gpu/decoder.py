@@ -652,19 +652,29 @@ class GpuRowGroupDecoder(object):
         """Queue a status tensor for the end-of-row-group flush."""
         self._pending_status.append((what, status))
 
-    def flush_status(self):
-        """One sync: verify every queued kernel status is clean.  The sync
-        also proves all prior async uploads landed, so the pinned staging
-        buffers go back to the free pool here."""
-        if not self._pending_status:
-            if self._staging_inuse:
+    def take_pending(self):
+        """Snapshot-and-clear this row-group's queued kernel statuses and
+        in-flight pinned staging buffers; pass the snapshot to
+        :meth:`check_and_recycle` when the row-group is consumed.  Lets the
+        reader keep several row-groups of GPU work in flight."""
+        snap = (self._pending_status, self._staging_inuse)
+        self._pending_status = []
+        self._staging_inuse = []
+        return snap
+
+    def check_and_recycle(self, snapshot):
+        """One sync: verify the snapshot's kernel statuses are clean.  The
+        sync also proves its async uploads landed, so the pinned staging
+        buffers go back to the free pool."""
+        pending, staging = snapshot
+        if not pending:
+            if staging:
                 torch.cuda.synchronize(self.device)
-                self._recycle_staging()
+                self._recycle_staging(staging)
             return
-        pending, self._pending_status = self._pending_status, []
         total = torch.stack([s.abs().sum() for _, s in pending]).sum()
         bad = int(total.item()) != 0
-        self._recycle_staging()
+        self._recycle_staging(staging)
         if bad:
             for what, s in pending:
                 vals = s.cpu()
@@ -673,10 +683,13 @@ class GpuRowGroupDecoder(object):
                         'GPU decode error in {}: status={}'
                         .format(what, vals.tolist()))
 
-    def _recycle_staging(self):
-        for key, buf in self._staging_inuse:
+    def flush_status(self):
+        """Check everything queued so far (single-row-group convenience)."""
+        self.check_and_recycle(self.take_pending())
+
+    def _recycle_staging(self, staging):
+        for key, buf in staging:
             self._staging_free.setdefault(key, []).append(buf)
-        self._staging_inuse = []
 
     # ------------------------------------------------------------------
     # codec stages over ByteArrayColumn

@@ -89,7 +89,7 @@ class GpuBatchReader(object):
                  shuffle_row_groups=True, shuffle_rows=False, predicate=None,
                  num_epochs=1, cur_shard=None, shard_count=None, seed=None,
                  transform_spec=None, device='cuda',
-                 cache_type=None, cache_size_limit=None):
+                 cache_type=None, cache_size_limit=None, pipeline_depth=3):
         if isinstance(schema_fields, NGram):
             raise NotImplementedError('NGram is a make_reader feature; use '
                                       'the sequence reader path')
@@ -134,6 +134,7 @@ class GpuBatchReader(object):
         self._cache = HbmCache(cache_size_limit) \
             if cache_type == 'hbm' and cache_size_limit else None
         self._inflight_hosts = []
+        self._pipeline_depth = max(1, int(pipeline_depth))
 
         # per-file metadata handles (footer parse once per file)
         self._file_md = {}
@@ -239,12 +240,27 @@ class GpuBatchReader(object):
                                  daemon=True)
             t.start()
             self._rows_epoch = 0
-            # one-deep software pipeline: decode of row-group N+1 is LAUNCHED
-            # before row-group N's batch is yielded, so N+1's H2D + kernels
-            # overlap the consumer's work on N; each row-group's status sync
-            # happens just before ITS batch is yielded
-            pending = None
+            # software pipeline: decode of the next row-groups is LAUNCHED
+            # before earlier batches are yielded, keeping `pipeline_depth`
+            # row-groups of H2D + kernels in flight on the stream (fills the
+            # chip when a single row-group's decode launches few waves);
+            # each row-group's status sync happens just before ITS batch is
+            # yielded
+            from collections import deque
+            pending = deque()
             dispatched = self._piece_pos  # pieces fully processed so far
+
+            def emit(entry):
+                nt, snap, pos = entry
+                t3 = time.perf_counter()
+                self._decoder.check_and_recycle(snap)
+                self.stage_times['flush'] += time.perf_counter() - t3
+                if nt is None:
+                    return None
+                self._piece_pos = pos
+                self._rows_epoch += len(nt[0])
+                return nt
+
             while True:
                 t0 = time.perf_counter()
                 kind, piece, host, meta = q.get()
@@ -278,24 +294,17 @@ class GpuBatchReader(object):
                     batch = self._postprocess(piece, columns)
                 self.stage_times['postprocess'] += time.perf_counter() - t2
                 dispatched += 1
-                if pending is not None:
-                    nt, pos = pending
-                    pending = None
-                    self._piece_pos = pos
-                    yield nt
-                if batch is None:
-                    continue
-                t3 = time.perf_counter()
-                self._decoder.flush_status()
-                self.stage_times['flush'] += time.perf_counter() - t3
-                self._rows_epoch += len(next(iter(batch.values())))
-                # cursor value once THIS batch is consumed = pieces
-                # dispatched up to and including it
-                pending = (self.schema.make_namedtuple(**batch), dispatched)
-            if pending is not None:
-                nt, pos = pending
-                self._piece_pos = pos
-                yield nt
+                nt = self.schema.make_namedtuple(**batch) \
+                    if batch is not None else None
+                pending.append((nt, self._decoder.take_pending(), dispatched))
+                while len(pending) >= self._pipeline_depth:
+                    out = emit(pending.popleft())
+                    if out is not None:
+                        yield out
+            while pending:
+                out = emit(pending.popleft())
+                if out is not None:
+                    yield out
             t.join()
             epoch_sync.epoch_end_sync(self._rows_epoch)
             epoch += 1

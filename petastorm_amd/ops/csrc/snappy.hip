@@ -60,9 +60,43 @@ __global__ void snappy_decompress_kernel(
   uint8_t* dst_base = out + out_off[page];
   const int64_t expected = out_len[page];
 
-  // lane 0 state, broadcast each op
+  // Lane 0 parses the tag stream through a register FIFO: one 8-byte
+  // unaligned global load refills up to 8 tag bytes, so dependent
+  // ~100-cycle global byte loads happen once per 8 consumed bytes instead
+  // of per byte (profiled 10ms/page without this — the parse is the
+  // bottleneck on poorly-compressing data with many small ops).
   int64_t in_pos = 0, out_pos = 0;
   int64_t total = 0;
+  uint64_t fifo = 0;
+  int fifo_n = 0;        // valid bytes in fifo (low bytes first)
+  int64_t fifo_pos = 0;  // stream position of fifo byte 0
+
+#define FIFO_REFILL(need)                                              \
+  if (fifo_n < (need)) {                                               \
+    fifo_pos = in_pos;                                                 \
+    uint64_t w = 0;                                                    \
+    int64_t avail = in_len - in_pos;                                   \
+    if (avail >= 8) {                                                  \
+      w = (uint64_t)load_u32_unaligned(in + in_pos) |                  \
+          ((uint64_t)load_u32_unaligned(in + in_pos + 4) << 32);       \
+      fifo_n = 8;                                                      \
+    } else {                                                           \
+      fifo_n = (int)(avail > 0 ? avail : 0);                           \
+      for (int z = 0; z < fifo_n; ++z)                                 \
+        w |= (uint64_t)in[in_pos + z] << (8 * z);                      \
+    }                                                                  \
+    fifo = w;                                                          \
+  }
+
+#define FIFO_TAKE(nbytes, out_v)                                       \
+  do {                                                                 \
+    out_v = fifo & ((nbytes) >= 8 ? ~0ull                              \
+                                  : ((1ull << (8 * (nbytes))) - 1ull));\
+    fifo >>= 8 * (nbytes);                                             \
+    fifo_n -= (nbytes);                                                \
+    in_pos += (nbytes);                                                \
+  } while (0)
+
   if (lane == 0) {
     total = read_varint(in, in_pos, in_len);
   }
@@ -80,30 +114,32 @@ __global__ void snappy_decompress_kernel(
       if (out_pos >= total || in_pos >= in_len) {
         done = 1;
       } else {
-        uint8_t tag = in[in_pos++];
+        FIFO_REFILL(5);
+        uint64_t tagw;
+        uint64_t tag = fifo & 0xFF;
         switch (tag & 3) {
           case 0: {  // literal
-            int64_t len = (tag >> 2) + 1;
+            int64_t len = (int64_t)(tag >> 2) + 1;
             if (len > 60) {
-              int n_extra = (int)(len - 60);
-              len = 0;
-              for (int i = 0; i < n_extra; ++i)
-                len |= (int64_t)in[in_pos + i] << (8 * i);
-              len += 1;
-              in_pos += n_extra;
+              int n_extra = (int)(len - 60);  // 1..4 extra length bytes
+              FIFO_TAKE(1 + n_extra, tagw);
+              len = (int64_t)(tagw >> 8) + 1;
+            } else {
+              FIFO_TAKE(1, tagw);
             }
             op.is_copy = 0;
             op.src = in_pos;
             op.dst = out_pos;
             op.len = (int32_t)len;
             in_pos += len;
+            fifo_n = 0;  // literal bytes skipped: invalidate fifo
             out_pos += len;
             break;
           }
           case 1: {  // copy, 1-byte offset
-            int32_t len = ((tag >> 2) & 0x7) + 4;
-            int64_t off = ((int64_t)(tag >> 5) << 8) | in[in_pos];
-            in_pos += 1;
+            FIFO_TAKE(2, tagw);
+            int32_t len = (int32_t)((tag >> 2) & 0x7) + 4;
+            int64_t off = (int64_t)((tag >> 5) << 8) | ((tagw >> 8) & 0xFF);
             op.is_copy = 1;
             op.src = off;
             op.dst = out_pos;
@@ -112,9 +148,9 @@ __global__ void snappy_decompress_kernel(
             break;
           }
           case 2: {  // copy, 2-byte offset
-            int32_t len = (tag >> 2) + 1;
-            int64_t off = load_u16_unaligned(in + in_pos);
-            in_pos += 2;
+            FIFO_TAKE(3, tagw);
+            int32_t len = (int32_t)(tag >> 2) + 1;
+            int64_t off = (int64_t)((tagw >> 8) & 0xFFFF);
             op.is_copy = 1;
             op.src = off;
             op.dst = out_pos;
@@ -123,9 +159,9 @@ __global__ void snappy_decompress_kernel(
             break;
           }
           default: {  // copy, 4-byte offset
-            int32_t len = (tag >> 2) + 1;
-            int64_t off = load_u32_unaligned(in + in_pos);
-            in_pos += 4;
+            FIFO_TAKE(5, tagw);
+            int32_t len = (int32_t)(tag >> 2) + 1;
+            int64_t off = (int64_t)((tagw >> 8) & 0xFFFFFFFFull);
             op.is_copy = 1;
             op.src = off;
             op.dst = out_pos;

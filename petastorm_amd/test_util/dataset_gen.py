@@ -172,7 +172,7 @@ def create_scalar_dataset(url, num_rows=1000, num_float_cols=8,
     fs.makedirs(path, exist_ok=True)
     pq.write_table(table, path + '/data-00000.parquet',
                    row_group_size=rowgroup_size, compression=compression,
-                   use_dictionary=False)
+                   use_dictionary=False, data_page_size=128 << 10)
     return cols
 
 
