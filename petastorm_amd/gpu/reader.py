@@ -409,8 +409,20 @@ class GpuBatchReader(object):
 
     def _predicate_mask(self, columns):
         names = list(self._predicate.get_fields())
-        # predicate columns are typically small scalars: evaluate with the
-        # numpy vectorized path, apply the mask on-device
+        n = len(next(iter(columns.values())))
+        # 1) fully on-device: most predicates (in_lambda with tensor-
+        #    compatible expressions, in_set via torch.isin) evaluate
+        #    directly on the CUDA columns
+        try:
+            dev_cols = {f: columns[f] for f in names}
+            if all(isinstance(v, torch.Tensor) for v in dev_cols.values()):
+                res = self._predicate.do_include(dev_cols)
+                if isinstance(res, torch.Tensor) and \
+                        res.dtype == torch.bool and res.numel() == n:
+                    return res
+        except Exception:  # noqa: BLE001 - fall back to host evaluation
+            pass
+        # 2) host fallback: numpy vectorized path, mask uploaded
         host_cols = {}
         for f in names:
             v = columns[f]

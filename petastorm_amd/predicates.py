@@ -49,7 +49,16 @@ class in_set(PredicateBase):
         return {self._predicate_field}
 
     def do_include(self, values):
-        return values[self._predicate_field] in self._inclusion_values
+        v = values[self._predicate_field]
+        try:
+            import torch
+            if isinstance(v, torch.Tensor) and v.dim() > 0:
+                ref = torch.tensor(sorted(self._inclusion_values),
+                                   device=v.device)
+                return torch.isin(v, ref)
+        except ImportError:  # pragma: no cover
+            pass
+        return v in self._inclusion_values
 
     def do_include_vectorized(self, columns):
         return np.isin(np.asarray(columns[self._predicate_field]),
