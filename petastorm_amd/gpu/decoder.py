@@ -15,11 +15,14 @@ Replaces the Arrow C++ decode inside ``piece.read`` of the reference
   [n, *shape] tensor (reference np.load, petastorm/codecs.py:155-157)
 * CompressedImageCodec(jpeg): restart-parallel decode (ops csrc/jpeg.hip)
 
-Columns whose physical encoding falls outside the GPU fast path (strings for
-Python consumption, png/zlib payloads until the inflate kernel lands,
-exotic encodings) are decoded with the CPU codec path and uploaded — the
+* CompressedImageCodec(png) / CompressedNdarrayCodec: DEFLATE inflate +
+  unfilter kernels (ops csrc/inflate.hip)
+* DataPageV2, GZIP pages and dictionary-encoded columns are native too
+
+Columns outside the GPU fast path (strings for Python consumption,
+exotic encodings/types) decode on the CPU codec path and upload — the
 decoder reports which columns took the assist so benchmarks and tests can
-assert the hot path is native.
+assert the hot path stays native.
 """
 
 import numpy as np

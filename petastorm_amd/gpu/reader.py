@@ -75,7 +75,7 @@ class _PinnedPool(object):
         with self._lock:
             lst = self._free.get(size)
             if lst:
-                return lst.pop()[:nbytes] if False else lst.pop()
+                return lst.pop()
         return torch.empty(size, dtype=torch.uint8,
                            pin_memory=torch.cuda.is_available())
 
