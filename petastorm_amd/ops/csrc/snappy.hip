@@ -50,8 +50,23 @@ __global__ void snappy_decompress_kernel(
     uint8_t* __restrict__ out, const int64_t* __restrict__ out_off,
     const int64_t* __restrict__ out_len,
     int32_t* __restrict__ status, int n_pages) {
-  const int waves_per_block = blockDim.x / PSA_WAVE;
-  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  // Two-phase wave algorithm, LDS-staged:
+  //   refill: all 64 lanes stage a window of the compressed stream in LDS
+  //   parse:  lane 0 walks tags reading ~30-cycle LDS instead of ~400-cycle
+  //           global memory, writing op descriptors (also LDS)
+  //   exec:   all 64 lanes replay the op batch (coalesced global copies)
+  // Snappy back-references always point at completed output, so every copy
+  // parallelizes as out[d+i] = out[d-off + i%off].
+  constexpr int WIN = 2048;   // staged window bytes
+  constexpr int OPB = 128;    // op descriptors per batch
+  constexpr int WPB = 4;      // waves per block (must match launcher)
+  __shared__ uint8_t stage[WPB][WIN];
+  __shared__ int64_t op_src[WPB][OPB];
+  __shared__ int64_t op_dst[WPB][OPB];
+  __shared__ int32_t op_len[WPB][OPB];   // negative length = back-copy
+
+  const int wave = threadIdx.x / PSA_WAVE;
+  const int page = blockIdx.x * (blockDim.x / PSA_WAVE) + wave;
   if (page >= n_pages) return;
   const int lane = lane_id();
 
@@ -60,46 +75,9 @@ __global__ void snappy_decompress_kernel(
   uint8_t* dst_base = out + out_off[page];
   const int64_t expected = out_len[page];
 
-  // Lane 0 parses the tag stream through a register FIFO: one 8-byte
-  // unaligned global load refills up to 8 tag bytes, so dependent
-  // ~100-cycle global byte loads happen once per 8 consumed bytes instead
-  // of per byte (profiled 10ms/page without this — the parse is the
-  // bottleneck on poorly-compressing data with many small ops).
   int64_t in_pos = 0, out_pos = 0;
   int64_t total = 0;
-  uint64_t fifo = 0;
-  int fifo_n = 0;        // valid bytes in fifo (low bytes first)
-  int64_t fifo_pos = 0;  // stream position of fifo byte 0
-
-#define FIFO_REFILL(need)                                              \
-  if (fifo_n < (need)) {                                               \
-    fifo_pos = in_pos;                                                 \
-    uint64_t w = 0;                                                    \
-    int64_t avail = in_len - in_pos;                                   \
-    if (avail >= 8) {                                                  \
-      w = (uint64_t)load_u32_unaligned(in + in_pos) |                  \
-          ((uint64_t)load_u32_unaligned(in + in_pos + 4) << 32);       \
-      fifo_n = 8;                                                      \
-    } else {                                                           \
-      fifo_n = (int)(avail > 0 ? avail : 0);                           \
-      for (int z = 0; z < fifo_n; ++z)                                 \
-        w |= (uint64_t)in[in_pos + z] << (8 * z);                      \
-    }                                                                  \
-    fifo = w;                                                          \
-  }
-
-#define FIFO_TAKE(nbytes, out_v)                                       \
-  do {                                                                 \
-    out_v = fifo & ((nbytes) >= 8 ? ~0ull                              \
-                                  : ((1ull << (8 * (nbytes))) - 1ull));\
-    fifo >>= 8 * (nbytes);                                             \
-    fifo_n -= (nbytes);                                                \
-    in_pos += (nbytes);                                                \
-  } while (0)
-
-  if (lane == 0) {
-    total = read_varint(in, in_pos, in_len);
-  }
+  if (lane == 0) total = read_varint(in, in_pos, in_len);
   total = wave_bcast(total);
   in_pos = wave_bcast(in_pos);
   if (total != expected) {
@@ -108,101 +86,117 @@ __global__ void snappy_decompress_kernel(
   }
 
   while (true) {
-    SnappyOp op;
-    int done = 0;
+    // ---- refill window at the current stream position ----
+    int64_t win_base = wave_bcast(in_pos);
+    if (win_base >= in_len || wave_bcast(out_pos) >= total) break;
+    for (int i = lane; i < WIN; i += PSA_WAVE)
+      stage[wave][i] = (win_base + i < in_len) ? in[win_base + i] : 0;
+    // lane 0 reads other lanes' LDS writes: needs a wave-wide LDS fence
+    __builtin_amdgcn_wave_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // ---- parse phase (lane 0 only) ----
+    int nops = 0;
+    int bad = 0;
     if (lane == 0) {
-      if (out_pos >= total || in_pos >= in_len) {
-        done = 1;
-      } else {
-        FIFO_REFILL(5);
-        uint64_t tagw;
-        uint64_t tag = fifo & 0xFF;
+      const uint8_t* w = stage[wave];
+      while (nops < OPB && out_pos < total) {
+        int64_t rel = in_pos - win_base;
+        // headers are <= 5 bytes; never read stage[] past WIN — break to
+        // refill the window at the new position instead
+        if (rel + 5 > WIN) break;
+        if (in_pos >= in_len) { bad = 3; break; }          // truncated
+        uint8_t tag = w[rel];
         switch (tag & 3) {
           case 0: {  // literal
             int64_t len = (int64_t)(tag >> 2) + 1;
+            int adv = 1;
             if (len > 60) {
-              int n_extra = (int)(len - 60);  // 1..4 extra length bytes
-              FIFO_TAKE(1 + n_extra, tagw);
-              len = (int64_t)(tagw >> 8) + 1;
-            } else {
-              FIFO_TAKE(1, tagw);
+              int n_extra = (int)(len - 60);
+              len = 0;
+              for (int i = 0; i < n_extra; ++i)
+                len |= (int64_t)w[rel + 1 + i] << (8 * i);
+              len += 1;
+              adv = 1 + n_extra;
             }
-            op.is_copy = 0;
-            op.src = in_pos;
-            op.dst = out_pos;
-            op.len = (int32_t)len;
-            in_pos += len;
-            fifo_n = 0;  // literal bytes skipped: invalidate fifo
+            op_src[wave][nops] = in_pos + adv;   // absolute input offset
+            op_dst[wave][nops] = out_pos;
+            op_len[wave][nops] = (int32_t)len;
+            in_pos += adv + len;                 // skip literal payload
             out_pos += len;
             break;
           }
-          case 1: {  // copy, 1-byte offset
-            FIFO_TAKE(2, tagw);
-            int32_t len = (int32_t)((tag >> 2) & 0x7) + 4;
-            int64_t off = (int64_t)((tag >> 5) << 8) | ((tagw >> 8) & 0xFF);
-            op.is_copy = 1;
-            op.src = off;
-            op.dst = out_pos;
-            op.len = len;
+          case 1: {
+            int32_t len = ((tag >> 2) & 0x7) + 4;
+            int64_t off = ((int64_t)(tag >> 5) << 8) | w[rel + 1];
+            op_src[wave][nops] = off;
+            op_dst[wave][nops] = out_pos;
+            op_len[wave][nops] = -len;
+            in_pos += 2;
             out_pos += len;
+            if (off <= 0 || off > op_dst[wave][nops]) bad = 2;
             break;
           }
-          case 2: {  // copy, 2-byte offset
-            FIFO_TAKE(3, tagw);
+          case 2: {
             int32_t len = (int32_t)(tag >> 2) + 1;
-            int64_t off = (int64_t)((tagw >> 8) & 0xFFFF);
-            op.is_copy = 1;
-            op.src = off;
-            op.dst = out_pos;
-            op.len = len;
+            int64_t off = (int64_t)w[rel + 1] | ((int64_t)w[rel + 2] << 8);
+            op_src[wave][nops] = off;
+            op_dst[wave][nops] = out_pos;
+            op_len[wave][nops] = -len;
+            in_pos += 3;
             out_pos += len;
+            if (off <= 0 || off > op_dst[wave][nops]) bad = 2;
             break;
           }
-          default: {  // copy, 4-byte offset
-            FIFO_TAKE(5, tagw);
+          default: {
             int32_t len = (int32_t)(tag >> 2) + 1;
-            int64_t off = (int64_t)((tagw >> 8) & 0xFFFFFFFFull);
-            op.is_copy = 1;
-            op.src = off;
-            op.dst = out_pos;
-            op.len = len;
+            int64_t off = (int64_t)w[rel + 1] | ((int64_t)w[rel + 2] << 8) |
+                          ((int64_t)w[rel + 3] << 16) |
+                          ((int64_t)w[rel + 4] << 24);
+            op_src[wave][nops] = off;
+            op_dst[wave][nops] = out_pos;
+            op_len[wave][nops] = -len;
+            in_pos += 5;
             out_pos += len;
+            if (off <= 0 || off > op_dst[wave][nops]) bad = 2;
             break;
           }
         }
-        if (op.is_copy && (op.src <= 0 || op.src > op.dst)) {
-          status[page] = 2;  // corrupt back-reference
-          done = 1;
-        }
+        if (bad) break;
+        ++nops;
       }
+      if (bad) status[page] = bad;
     }
-    done = wave_bcast(done);
-    if (done) break;
-    op.src = wave_bcast(op.src);
-    op.dst = wave_bcast(op.dst);
-    op.len = wave_bcast(op.len);
-    op.is_copy = wave_bcast(op.is_copy);
+    bad = wave_bcast(bad);
+    if (bad) return;
+    nops = wave_bcast(nops);
+    __builtin_amdgcn_wave_barrier();
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-    if (op.is_copy) {
-      const int64_t off = op.src;
-      uint8_t* d = dst_base + op.dst;
-      const uint8_t* s = d - off;
-      for (int32_t i = lane; i < op.len; i += PSA_WAVE)
-        d[i] = s[i % off];
-    } else {
-      const uint8_t* s = in + op.src;
-      uint8_t* d = dst_base + op.dst;
-      // vectorize the common large-literal case
-      int32_t len = op.len;
-      int32_t vec = len & ~3;
-      for (int32_t i = lane * 4; i < vec; i += PSA_WAVE * 4) {
-        uint32_t w = load_u32_unaligned(s + i);
-        d[i + 0] = (uint8_t)(w);
-        d[i + 1] = (uint8_t)(w >> 8);
-        d[i + 2] = (uint8_t)(w >> 16);
-        d[i + 3] = (uint8_t)(w >> 24);
+    // ---- execute phase (all lanes) ----
+    for (int k = 0; k < nops; ++k) {
+      int32_t len = op_len[wave][k];
+      int64_t d0 = op_dst[wave][k];
+      if (len >= 0) {  // literal: copy from input
+        const uint8_t* s = in + op_src[wave][k];
+        uint8_t* d = dst_base + d0;
+        int32_t vec = len & ~3;
+        for (int32_t i = lane * 4; i < vec; i += PSA_WAVE * 4) {
+          uint32_t v = load_u32_unaligned(s + i);
+          d[i + 0] = (uint8_t)(v);
+          d[i + 1] = (uint8_t)(v >> 8);
+          d[i + 2] = (uint8_t)(v >> 16);
+          d[i + 3] = (uint8_t)(v >> 24);
+        }
+        for (int32_t i = vec + lane; i < len; i += PSA_WAVE) d[i] = s[i];
+      } else {         // back-copy (possibly overlapping)
+        len = -len;
+        const int64_t off = op_src[wave][k];
+        uint8_t* d = dst_base + d0;
+        const uint8_t* s = d - off;
+        for (int32_t i = lane; i < len; i += PSA_WAVE)
+          d[i] = s[i % off];
       }
-      for (int32_t i = vec + lane; i < len; i += PSA_WAVE) d[i] = s[i];
     }
   }
   if (lane == 0 && out_pos != total) status[page] = 3;  // truncated stream
