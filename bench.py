@@ -151,7 +151,8 @@ def bench_scalar(args, rank, world, device, dist):
         url, device=str(device), num_epochs=None, shuffle_row_groups=True,
         seed=7, schema_fields=['id', 'f0', 'f1', 'f2', 'f3', 'i0', 'i1'],
         cur_shard=rank if world > 1 else None,
-        shard_count=world if world > 1 else None)
+        shard_count=world if world > 1 else None,
+        gpu_options=dict(pipeline_depth=6))
     loader = BatchedDataLoader(reader, batch_size=args.batch_size * 64)
     it = iter(loader)
 

@@ -170,9 +170,11 @@ def create_scalar_dataset(url, num_rows=1000, num_float_cols=8,
     table = pa.table(cols)
     fs, path = get_filesystem_and_path_or_paths(url)
     fs.makedirs(path, exist_ok=True)
+    # 32 KiB pages: the page is the decompression-parallelism unit on the
+    # GPU (one wave per snappy stream); small pages keep 256 CUs fed
     pq.write_table(table, path + '/data-00000.parquet',
                    row_group_size=rowgroup_size, compression=compression,
-                   use_dictionary=False, data_page_size=128 << 10)
+                   use_dictionary=False, data_page_size=32 << 10)
     return cols
 
 
