@@ -93,3 +93,11 @@ def test_shuffling_analysis(test_dataset):
         test_dataset['url'], 'id', {'shuffle_row_groups': False},
         num_corr_samples=2)
     assert corr_shuffled.mean() < corr_ordered.mean()
+
+
+def test_reader_throughput_spawned_process(test_dataset):
+    """reference throughput.py:144-149 self-respawn for clean RSS"""
+    result = reader_throughput(test_dataset['url'], warmup_cycles_count=2,
+                               measure_cycles_count=5, loaders_count=2,
+                               spawn_new_process=True)
+    assert result.samples_per_second > 0

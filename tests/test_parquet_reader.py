@@ -77,3 +77,31 @@ def test_shuffle_rows_within_rowgroup(scalar_dataset):
         b = next(iter(r))
     assert b.id.tolist() != sorted(b.id.tolist())
     assert sorted(b.id.tolist()) == list(range(len(b.id)))
+
+
+def test_cache_and_shuffle_across_epochs(scalar_dataset, tmp_path):
+    """Cached row-groups re-shuffle correctly across epochs (reference
+    tests/test_parquet_reader.py shuffle+cache interaction)."""
+    kwargs = dict(reader_pool_type='thread', workers_count=2,
+                  shuffle_row_groups=True, seed=4, num_epochs=2,
+                  cache_type='local-disk',
+                  cache_location=str(tmp_path / 'cache'),
+                  cache_size_limit=200 << 20, cache_row_size_estimate=1024,
+                  schema_fields=['id'])
+    with make_batch_reader(scalar_dataset['url'], **kwargs) as r:
+        ids = _collect(list(r))['id']
+    assert len(ids) == 1000
+    np.testing.assert_array_equal(np.sort(ids), np.repeat(np.arange(500), 2))
+    # second reader: cache hits must yield identical content
+    with make_batch_reader(scalar_dataset['url'], **kwargs) as r:
+        ids2 = _collect(list(r))['id']
+    np.testing.assert_array_equal(ids, ids2)
+
+
+def test_zero_rows_after_filter_cached(scalar_dataset, tmp_path):
+    from petastorm_amd.predicates import in_lambda
+    pred = in_lambda(['id'], lambda v: v['id'] < 0)  # matches nothing
+    kwargs = dict(reader_pool_type='dummy', shuffle_row_groups=False,
+                  predicate=pred)
+    with make_batch_reader(scalar_dataset['url'], **kwargs) as r:
+        assert list(r) == []
