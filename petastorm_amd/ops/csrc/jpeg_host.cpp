@@ -18,6 +18,7 @@
 #include <cstdint>
 #include <cstring>
 #include <map>
+#include <thread>
 #include <vector>
 
 namespace psa {
@@ -255,7 +256,21 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
     }
     TORCH_CHECK(got_sos, "image ", i, ": no scan found");
 
-    // ---- restart-segment scan over the entropy-coded data ----
+  }
+
+  // ---- restart-segment scan (parallel over images) ----
+  // memchr-driven: jump 0xFF to 0xFF instead of walking every byte; scans
+  // are independent per image, so they fan out over host threads (this is
+  // the heaviest host stage at fine restart intervals).
+  struct SegList {
+    std::vector<int64_t> pos, end;
+    std::vector<int32_t> mcu0, nmcu;
+  };
+  std::vector<SegList> per_img(n);
+  auto scan_image = [&](int64_t i) {
+    ImgInfo& im = imgs[i];
+    const uint8_t* p = base + off[i];
+    const int64_t len = vlen(i);
     int hmax = 1, vmax = 1;
     for (int c = 0; c < im.ncomp; ++c) {
       hmax = std::max(hmax, im.comp_h[c]);
@@ -265,14 +280,16 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
     int mcus_y = (im.h + 8 * vmax - 1) / (8 * vmax);
     int total_mcus = mcus_x * mcus_y;
     int ri = im.restart_interval > 0 ? im.restart_interval : total_mcus;
+    SegList& sl = per_img[i];
+    sl.pos.reserve(total_mcus / std::max(ri, 1) + 2);
+    sl.end.reserve(sl.pos.capacity());
+    sl.mcu0.reserve(sl.pos.capacity());
+    sl.nmcu.reserve(sl.pos.capacity());
 
     int64_t sp = im.scan_start;
     int64_t seg_begin = sp;
     int mcu_done = 0;
     const int64_t abs0 = off[i];
-    // memchr-driven scan: jump 0xFF to 0xFF instead of walking every byte
-    // (the scan data is high-entropy; stuffed 0xFF bytes appear ~1/256, so
-    // this runs at memchr speed rather than 1 byte/iteration)
     while (sp + 1 < len) {
       const void* hit = memchr(p + sp, 0xFF, (size_t)(len - sp - 1));
       if (hit == nullptr) { sp = len; break; }
@@ -283,11 +300,10 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
         continue;
       }
       if (m >= 0xD0 && m <= 0xD7) {  // RSTn
-        seg_img.push_back((int32_t)i);
-        seg_pos.push_back(abs0 + seg_begin);
-        seg_end.push_back(abs0 + sp);
-        seg_mcu0.push_back(mcu_done);
-        seg_nmcu.push_back(std::min(ri, total_mcus - mcu_done));
+        sl.pos.push_back(abs0 + seg_begin);
+        sl.end.push_back(abs0 + sp);
+        sl.mcu0.push_back(mcu_done);
+        sl.nmcu.push_back(std::min(ri, total_mcus - mcu_done));
         mcu_done += ri;
         sp += 2;
         seg_begin = sp;
@@ -298,11 +314,35 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
     }
     im.scan_end = sp;
     if (mcu_done < total_mcus) {
+      sl.pos.push_back(abs0 + seg_begin);
+      sl.end.push_back(abs0 + sp);
+      sl.mcu0.push_back(mcu_done);
+      sl.nmcu.push_back(total_mcus - mcu_done);
+    }
+  };
+  {
+    unsigned hw = std::thread::hardware_concurrency();
+    int n_threads = (int)std::min<int64_t>(std::max(1u, hw / 2), n);
+    if (n_threads <= 1 || n < 8) {
+      for (int64_t i = 0; i < n; ++i) scan_image(i);
+    } else {
+      std::vector<std::thread> pool;
+      pool.reserve(n_threads);
+      for (int t = 0; t < n_threads; ++t)
+        pool.emplace_back([&, t]() {
+          for (int64_t i = t; i < n; i += n_threads) scan_image(i);
+        });
+      for (auto& th : pool) th.join();
+    }
+  }
+  for (int64_t i = 0; i < n; ++i) {
+    SegList& sl = per_img[i];
+    for (size_t k2 = 0; k2 < sl.pos.size(); ++k2) {
       seg_img.push_back((int32_t)i);
-      seg_pos.push_back(abs0 + seg_begin);
-      seg_end.push_back(abs0 + sp);
-      seg_mcu0.push_back(mcu_done);
-      seg_nmcu.push_back(total_mcus - mcu_done);
+      seg_pos.push_back(sl.pos[k2]);
+      seg_end.push_back(sl.end[k2]);
+      seg_mcu0.push_back(sl.mcu0[k2]);
+      seg_nmcu.push_back(sl.nmcu[k2]);
     }
   }
 
