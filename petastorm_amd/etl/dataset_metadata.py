@@ -314,3 +314,77 @@ def infer_or_load_unischema(fs, path_or_paths):
         arrow_schema = pq.ParquetFile(fs.open(files[0], 'rb')).schema_arrow
         return _with_partition_fields(
             Unischema.from_arrow_schema(arrow_schema)), False
+
+
+def select_pieces_by_filters(fs, pieces, filters):
+    """Statistics/partition-based row-group pruning for pyarrow-style
+    ``filters`` (the reference forwards these to pq.ParquetDataset,
+    reference reader.py:431-433 ``filters`` kwarg).
+
+    ``filters`` is a list of ``(column, op, value)`` tuples (ANDed), or a
+    list of such lists (ORed DNF).  Ops: ==, =, !=, <, <=, >, >=, in,
+    not in.  A row group is kept unless its column-chunk min/max statistics
+    (or hive partition value) PROVE no row can match — unknown statistics
+    keep the group (pruning is an optimization, never a correctness filter).
+    """
+    import pyarrow.parquet as pq
+    if not filters:
+        return pieces
+    if filters and isinstance(filters[0], tuple):
+        dnf = [filters]
+    else:
+        dnf = filters
+
+    stats_cache = {}
+
+    def col_range(piece, name):
+        if name in piece.partitions:
+            v = piece.partitions[name]
+            return v, v
+        key = piece.path
+        if key not in stats_cache:
+            stats_cache[key] = pq.ParquetFile(fs.open(key, 'rb')).metadata
+        md = stats_cache[key].row_group(piece.row_group)
+        for ci in range(md.num_columns):
+            col = md.column(ci)
+            if col.path_in_schema == name:
+                st = col.statistics
+                if st is not None and st.has_min_max:
+                    return st.min, st.max
+        return None, None
+
+    def clause_may_match(piece, clause):
+        for name, op, value in clause:
+            lo, hi = col_range(piece, name)
+            if lo is None:
+                continue  # unknown -> can't prune on this term
+            try:
+                if op in ('==', '='):
+                    if not (lo <= value <= hi):
+                        return False
+                elif op == '<':
+                    if not (lo < value):
+                        return False
+                elif op == '<=':
+                    if not (lo <= value):
+                        return False
+                elif op == '>':
+                    if not (hi > value):
+                        return False
+                elif op == '>=':
+                    if not (hi >= value):
+                        return False
+                elif op == 'in':
+                    if not any(lo <= v <= hi for v in value):
+                        return False
+                elif op in ('!=', 'not in'):
+                    # can only prune when the whole range is one value
+                    excluded = [value] if op == '!=' else list(value)
+                    if lo == hi and lo in excluded:
+                        return False
+            except TypeError:
+                continue  # incomparable types: keep
+        return True
+
+    return [p for p in pieces
+            if any(clause_may_match(p, clause) for clause in dnf)]

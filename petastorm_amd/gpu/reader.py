@@ -88,7 +88,7 @@ class GpuBatchReader(object):
     def __init__(self, fs, path_or_paths, schema_fields=None,
                  shuffle_row_groups=True, shuffle_rows=False, predicate=None,
                  num_epochs=1, cur_shard=None, shard_count=None, seed=None,
-                 transform_spec=None, device='cuda',
+                 transform_spec=None, filters=None, device='cuda',
                  cache_type=None, cache_size_limit=None, pipeline_depth=3):
         if isinstance(schema_fields, NGram):
             raise NotImplementedError('NGram is a make_reader feature; use '
@@ -116,6 +116,9 @@ class GpuBatchReader(object):
         self.last_row_consumed = False
 
         self._pieces = dsm.load_row_groups(fs, path_or_paths)
+        if filters:
+            self._pieces = dsm.select_pieces_by_filters(fs, self._pieces,
+                                                        filters)
         if not self._pieces:
             raise NoDataAvailableError('Dataset has no row groups')
         cur_shard, shard_count = epoch_sync.shard_for_rank(cur_shard,

@@ -91,7 +91,7 @@ def make_reader(dataset_url,
                 cur_shard=None, shard_count=None, seed=None,
                 cache_type='null', cache_location=None, cache_size_limit=None,
                 cache_row_size_estimate=None, cache_extra_settings=None,
-                transform_spec=None,
+                transform_spec=None, filters=None,
                 storage_options=None):
     """Row-oriented reader over a petastorm_amd dataset (reference :60-206).
 
@@ -121,7 +121,7 @@ def make_reader(dataset_url,
                   num_epochs=num_epochs,
                   cur_shard=cur_shard, shard_count=shard_count, seed=seed,
                   cache=cache,
-                  transform_spec=transform_spec,
+                  transform_spec=transform_spec, filters=filters,
                   batched_output=False)
 
 
@@ -138,7 +138,7 @@ def make_batch_reader(dataset_url_or_urls,
                       cache_type='null', cache_location=None,
                       cache_size_limit=None, cache_row_size_estimate=None,
                       cache_extra_settings=None,
-                      transform_spec=None,
+                      transform_spec=None, filters=None,
                       decode_codecs=True,
                       storage_options=None,
                       device=None, gpu_options=None):
@@ -165,6 +165,7 @@ def make_batch_reader(dataset_url_or_urls,
                               num_epochs=num_epochs,
                               cur_shard=cur_shard, shard_count=shard_count,
                               seed=seed, transform_spec=transform_spec,
+                              filters=filters,
                               device=device, **(gpu_options or {}))
     cache = _make_cache(cache_type, cache_location, cache_size_limit,
                         cache_row_size_estimate, cache_extra_settings)
@@ -181,7 +182,7 @@ def make_batch_reader(dataset_url_or_urls,
                   num_epochs=num_epochs,
                   cur_shard=cur_shard, shard_count=shard_count, seed=seed,
                   cache=cache,
-                  transform_spec=transform_spec,
+                  transform_spec=transform_spec, filters=filters,
                   decode_codecs=decode_codecs,
                   batched_output=True)
 
@@ -200,8 +201,8 @@ class Reader(object):
                  shuffle_row_groups=True, shuffle_row_drop_partitions=1,
                  shuffle_rows=False, predicate=None, rowgroup_selector=None,
                  num_epochs=1, cur_shard=None, shard_count=None, seed=None,
-                 cache=None, transform_spec=None, decode_codecs=True,
-                 batched_output=False):
+                 cache=None, transform_spec=None, filters=None,
+                 decode_codecs=True, batched_output=False):
         if (cur_shard is None) != (shard_count is None):
             raise ValueError('cur_shard and shard_count must be used together')
         if num_epochs is not None and (not isinstance(num_epochs, int) or num_epochs < 1):
@@ -246,7 +247,14 @@ class Reader(object):
         self._pieces = dsm.load_row_groups(self._fs, self._paths)
         if not self._pieces:
             raise NoDataAvailableError('Dataset has no row groups')
+        if filters:
+            kept = {p.index for p in dsm.select_pieces_by_filters(
+                self._fs, self._pieces, filters)}
+        else:
+            kept = None
         selected = self._apply_row_group_selector(rowgroup_selector)
+        if kept is not None:
+            selected = [i for i in selected if i in kept]
         predicate = self._push_down_partition_predicate(predicate, selected)
         if isinstance(predicate, tuple):  # (pushed_down_selection, None)
             selected, predicate = predicate

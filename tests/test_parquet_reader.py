@@ -105,3 +105,26 @@ def test_zero_rows_after_filter_cached(scalar_dataset, tmp_path):
                   predicate=pred)
     with make_batch_reader(scalar_dataset['url'], **kwargs) as r:
         assert list(r) == []
+
+
+def test_filters_statistics_pruning(scalar_dataset):
+    """pyarrow-style `filters`: row groups pruned by min/max statistics
+    (reference forwards filters to pq.ParquetDataset, reader.py:431-433)."""
+    with make_batch_reader(scalar_dataset['url'], reader_pool_type='dummy',
+                           shuffle_row_groups=False,
+                           schema_fields=['id'],
+                           filters=[('id', '>=', 400)]) as r:
+        ids = _collect(list(r))['id']
+        # only the last row group (ids 400..499) should be ventilated
+        assert r.diagnostics['items_ventilated'] == 1
+    assert set(ids) == set(range(400, 500))
+
+
+def test_filters_dnf_or(scalar_dataset):
+    dnf = [[('id', '<', 100)], [('id', '>=', 400)]]
+    with make_batch_reader(scalar_dataset['url'], reader_pool_type='dummy',
+                           shuffle_row_groups=False, schema_fields=['id'],
+                           filters=dnf) as r:
+        ids = _collect(list(r))['id']
+        assert r.diagnostics['items_ventilated'] == 2
+    assert set(ids) == set(range(0, 100)) | set(range(400, 500))
