@@ -43,6 +43,25 @@ logger = logging.getLogger(__name__)
 _PREFETCH_DEPTH = 2
 
 
+class _Ended(object):
+    """Placeholder for an exhausted IO queue in the round-robin scan."""
+
+    def get(self):
+        raise RuntimeError('queue already ended')
+
+
+_ENDED_QUEUE = _Ended()
+
+
+def _next_live(qs, rr):
+    n = len(qs)
+    for step in range(1, n + 1):
+        cand = (rr + step) % n
+        if qs[cand] is not _ENDED_QUEUE:
+            return cand
+    return rr
+
+
 class _TraceRange(object):
     """rocTX range (torch.cuda.nvtx maps to roctx on ROCm) for rocprofv3
     --marker-trace; enabled with PSA_TRACE=1 (replaces the
@@ -89,7 +108,8 @@ class GpuBatchReader(object):
                  shuffle_row_groups=True, shuffle_rows=False, predicate=None,
                  num_epochs=1, cur_shard=None, shard_count=None, seed=None,
                  transform_spec=None, filters=None, device='cuda',
-                 cache_type=None, cache_size_limit=None, pipeline_depth=3):
+                 cache_type=None, cache_size_limit=None, pipeline_depth=3,
+                 io_threads=2):
         if isinstance(schema_fields, NGram):
             raise NotImplementedError('NGram is a make_reader feature; use '
                                       'the sequence reader path')
@@ -138,6 +158,7 @@ class GpuBatchReader(object):
             if cache_type == 'hbm' and cache_size_limit else None
         self._inflight_hosts = []
         self._pipeline_depth = max(1, int(pipeline_depth))
+        self._io_threads = max(1, int(io_threads))
 
         # per-file metadata handles (footer parse once per file)
         self._file_md = {}
@@ -180,6 +201,14 @@ class GpuBatchReader(object):
 
     # ------------------------------------------------------------------
     def _io_worker(self, pieces, out_q):
+        """Read + host-parse the given pieces in order, into ``out_q``.
+
+        Several IO workers run concurrently (round-robin piece assignment,
+        round-robin consumption) — a single thread's pread + native parse
+        caps the pipeline at a few hundred row-groups/s on a contended host.
+        Ordering stays deterministic because the consumer reads the
+        per-thread queues in the same round-robin order.
+        """
         columns = list(self._view_schema.fields.keys())
         try:
             for piece in pieces:
@@ -196,8 +225,8 @@ class GpuBatchReader(object):
                     self._pin_pool)
                 t1 = time.perf_counter()
                 # host-only parse work (page walk, offset scans, image
-                # headers) runs HERE so it overlaps GPU decode of the
-                # previous row-group
+                # headers) runs HERE so it overlaps GPU decode of other
+                # row-groups
                 plan = self._decoder.prepare_host(host, meta,
                                                   self._storage_schema)
                 self.stage_times['io_read'] += t1 - t0
@@ -238,10 +267,14 @@ class GpuBatchReader(object):
                 skip = 0
             else:
                 self._piece_pos = 0
-            q = queue.Queue(maxsize=_PREFETCH_DEPTH)
-            t = threading.Thread(target=self._io_worker, args=(pieces, q),
-                                 daemon=True)
-            t.start()
+            n_io = min(self._io_threads, max(1, len(pieces)))
+            qs = [queue.Queue(maxsize=_PREFETCH_DEPTH) for _ in range(n_io)]
+            threads = [
+                threading.Thread(target=self._io_worker,
+                                 args=(pieces[t::n_io], qs[t]), daemon=True)
+                for t in range(n_io)]
+            for t in threads:
+                t.start()
             self._rows_epoch = 0
             # software pipeline: decode of the next row-groups is LAUNCHED
             # before earlier batches are yielded, keeping `pipeline_depth`
@@ -264,14 +297,20 @@ class GpuBatchReader(object):
                 self._rows_epoch += len(nt[0])
                 return nt
 
-            while True:
+            rr = 0
+            ended = 0
+            while ended < n_io:
                 t0 = time.perf_counter()
-                kind, piece, host, meta = q.get()
+                kind, piece, host, meta = qs[rr].get()
                 self.stage_times['io_wait'] += time.perf_counter() - t0
                 if kind == 'end':
-                    break
+                    ended += 1
+                    qs[rr] = _ENDED_QUEUE
+                    rr = _next_live(qs, rr)
+                    continue
                 if kind == 'error':
                     raise piece
+                rr = _next_live(qs, rr)
                 if kind == 'cached':
                     columns = self._cache._store[self._cache_key(piece)]
                     self._cache.get(self._cache_key(piece), lambda: columns)
@@ -308,7 +347,8 @@ class GpuBatchReader(object):
                 out = emit(pending.popleft())
                 if out is not None:
                     yield out
-            t.join()
+            for t in threads:
+                t.join()
             epoch_sync.epoch_end_sync(self._rows_epoch)
             epoch += 1
 
