@@ -85,7 +85,8 @@ def bench_imagenet(args, rank, world, device, dist):
     # uniform 256-row row groups, 24 per 6144 rows: shards stay equal-sized
     # so the RCCL epoch collectives stay in lock-step across ranks
     n_rows = args.rows or 6144
-    url = _dataset_dir('imagenet_{}'.format(n_rows), rank, dist,
+    rst = os.environ.get('PSA_JPEG_RST_BLOCKS', '4')
+    url = _dataset_dir('imagenet_{}_r{}'.format(n_rows, rst), rank, dist,
                        lambda u: create_imagenet_dataset(
                            u, num_rows=n_rows, rows_per_rowgroup=256))
 

@@ -255,14 +255,17 @@ class CompressedImageCodec(DataframeColumnCodec):
         if self._image_codec == 'jpeg':
             if value.dtype != np.uint8:
                 raise ValueError('jpeg requires uint8 images')
-            # One RSTn every 4 MCUs (~0.3% size): each restart segment is
-            # an independent bitstream, so a 224px image decodes as ~49
+            # One RSTn every few MCUs (~0.3% size): each restart segment
+            # is an independent bitstream, so a 224px image decodes as ~50+
             # parallel segments on the MI355X Huffman kernel (vs 14 with
             # row-level markers) while keeping the host-side segment scan
-            # cheap.  Measured sweet spot: blocks=1 made the host parse the
-            # pipeline bottleneck; rows=1 underfilled the GPU.
+            # cheap.  Sweep-measured default: 4 MCUs (blocks=1 doubled host
+            # parse; rows=1 underfilled the GPU).  Override with
+            # PSA_JPEG_RST_BLOCKS for experiments.
+            import os
+            rst = int(os.environ.get('PSA_JPEG_RST_BLOCKS', '4'))
             img.save(buf, format='JPEG', quality=self.quality,
-                     restart_marker_blocks=4)
+                     restart_marker_blocks=rst)
         else:
             img.save(buf, format='PNG')
         return buf.getvalue()
