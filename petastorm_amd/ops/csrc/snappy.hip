@@ -119,6 +119,8 @@ __global__ void snappy_decompress_kernel(
               len += 1;
               adv = 1 + n_extra;
             }
+            if (in_pos + adv + len > in_len ||
+                out_pos + len > total) { bad = 2; break; }
             op_src[wave][nops] = in_pos + adv;   // absolute input offset
             op_dst[wave][nops] = out_pos;
             op_len[wave][nops] = (int32_t)len;
@@ -133,6 +135,7 @@ __global__ void snappy_decompress_kernel(
             op_dst[wave][nops] = out_pos;
             op_len[wave][nops] = -len;
             in_pos += 2;
+            if (out_pos + len > total) { bad = 2; break; }
             out_pos += len;
             if (off <= 0 || off > op_dst[wave][nops]) bad = 2;
             break;
@@ -144,6 +147,7 @@ __global__ void snappy_decompress_kernel(
             op_dst[wave][nops] = out_pos;
             op_len[wave][nops] = -len;
             in_pos += 3;
+            if (out_pos + len > total) { bad = 2; break; }
             out_pos += len;
             if (off <= 0 || off > op_dst[wave][nops]) bad = 2;
             break;
@@ -157,6 +161,7 @@ __global__ void snappy_decompress_kernel(
             op_dst[wave][nops] = out_pos;
             op_len[wave][nops] = -len;
             in_pos += 5;
+            if (out_pos + len > total) { bad = 2; break; }
             out_pos += len;
             if (off <= 0 || off > op_dst[wave][nops]) bad = 2;
             break;
