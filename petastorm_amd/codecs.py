@@ -255,15 +255,14 @@ class CompressedImageCodec(DataframeColumnCodec):
         if self._image_codec == 'jpeg':
             if value.dtype != np.uint8:
                 raise ValueError('jpeg requires uint8 images')
-            # One RSTn every few MCUs (~0.3% size): each restart segment
-            # is an independent bitstream, so a 224px image decodes as ~50+
-            # parallel segments on the MI355X Huffman kernel (vs 14 with
-            # row-level markers) while keeping the host-side segment scan
-            # cheap.  Sweep-measured default: 4 MCUs (blocks=1 doubled host
-            # parse; rows=1 underfilled the GPU).  Override with
-            # PSA_JPEG_RST_BLOCKS for experiments.
+            # One RSTn every 2 MCUs (~0.5% size): each restart segment is
+            # an independent bitstream, so a 224px image decodes as ~100
+            # parallel segments on the MI355X Huffman kernel.  Sweep-measured
+            # on MI355X (profiles/RESULTS.md): blocks=2 = 181k img/s vs
+            # blocks=4 = 148k and blocks=1 = 16k (host segment handling
+            # dominates at 1).  Override with PSA_JPEG_RST_BLOCKS.
             import os
-            rst = int(os.environ.get('PSA_JPEG_RST_BLOCKS', '4'))
+            rst = int(os.environ.get('PSA_JPEG_RST_BLOCKS', '2'))
             img.save(buf, format='JPEG', quality=self.quality,
                      restart_marker_blocks=rst)
         else:
