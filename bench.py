@@ -84,11 +84,13 @@ def bench_imagenet(args, rank, world, device, dist):
     # (sharding is per row-group, reference reader.py:573-597)
     # uniform 256-row row groups, 24 per 6144 rows: shards stay equal-sized
     # so the RCCL epoch collectives stay in lock-step across ranks
-    n_rows = args.rows or 6144
-    rst = os.environ.get('PSA_JPEG_RST_BLOCKS', '4')
-    url = _dataset_dir('imagenet_{}_r{}'.format(n_rows, rst), rank, dist,
+    rpg = int(os.environ.get('PSA_IMAGENET_RPG', '256'))
+    n_rows = args.rows or 24 * rpg  # 24 uniform row-groups (equal 8-GPU shards)
+    rst = os.environ.get('PSA_JPEG_RST_BLOCKS', '2')
+    url = _dataset_dir('imagenet_{}_r{}_g{}'.format(n_rows, rst, rpg), rank,
+                       dist,
                        lambda u: create_imagenet_dataset(
-                           u, num_rows=n_rows, rows_per_rowgroup=256))
+                           u, num_rows=n_rows, rows_per_rowgroup=rpg))
 
     ext = ops.ext()
     mean = torch.tensor([0.485, 0.456, 0.406], device=device)
