@@ -43,7 +43,9 @@ def _init_dist(world):
     import torch.distributed as dist
     if world <= 1 or dist.is_initialized():
         return None
-    backend = 'nccl' if torch.cuda.is_available() else 'gloo'
+    backend = os.environ.get(
+        'PSA_DIST_BACKEND',
+        'nccl' if torch.cuda.is_available() else 'gloo')
     dist.init_process_group(backend=backend)
     return dist
 
@@ -310,6 +312,9 @@ def main():
     rank, world, local = _dist_env()
     dist = _init_dist(world)
     if torch.cuda.is_available():
+        # validation aid: multiple ranks can share one GPU (with a gloo
+        # process group) on single-GPU test boxes
+        local = min(local, torch.cuda.device_count() - 1)
         torch.cuda.set_device(local)
         device = torch.device('cuda', local)
     else:
