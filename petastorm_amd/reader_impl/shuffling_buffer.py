@@ -180,12 +180,19 @@ class BatchedNoopShufflingBuffer(ShufflingBufferBase):
 
 
 class BatchedRandomShufflingBuffer(ShufflingBufferBase):
-    """Column-tensor shuffling pool: maintains a single dict of column
-    tensors up to capacity; batches are gathered with a presampled random
-    permutation (reference pytorch_shuffling_buffer.py:137-279).
+    """Column-tensor shuffling pool over a PREALLOCATED buffer.
 
-    Works for CPU and CUDA tensors alike — on CUDA the index gather is the
-    HBM-resident ``gpu_shuffle_gather``.
+    Semantics follow reference pytorch_shuffling_buffer.py:137-279 (random
+    batches out of a bounded mixing pool), but the implementation is the
+    vectorized analog of the reference's O(1) swap-with-last single-row
+    buffer (reference shuffling_buffer.py:158-167): a fixed
+    ``[capacity + extra, ...]`` tensor per column, appends copy into free
+    slots, and a retrieve gathers ``batch_size`` random rows then back-fills
+    their slots from the tail.  Per-operation cost is O(rows moved) — the
+    previous implementation re-``torch.cat``-ed the whole pool on every add.
+
+    Works for CPU and CUDA tensors alike — on CUDA the gathers/scatters stay
+    HBM-resident.
     """
 
     def __init__(self, shuffling_buffer_capacity, min_after_retrieve,
@@ -197,7 +204,8 @@ class BatchedRandomShufflingBuffer(ShufflingBufferBase):
         self._min_after_retrieve = min_after_retrieve
         self._batch_size = batch_size
         self._extra_capacity = extra_capacity
-        self._columns = None
+        self._pool = None           # name -> [cap_total, ...] tensor
+        self._cap_total = None
         self._size = 0
         self._done = False
         self._generator = None
@@ -215,28 +223,45 @@ class BatchedRandomShufflingBuffer(ShufflingBufferBase):
         n = _num_rows(columns)
         if n == 0:
             return
-        if self._columns is None:
-            self._columns = dict(columns)
-        else:
-            self._columns = {k: torch.cat([self._columns[k], columns[k]])
-                             for k in self._columns}
-        self._size += n
-        if self._size > self._capacity + self._extra_capacity:
+        if self._size + n > self._capacity + self._extra_capacity:
             raise RuntimeError('Buffer exceeded capacity+extra_capacity')
+        if self._pool is None:
+            self._cap_total = self._capacity + self._extra_capacity
+            self._pool = {
+                k: torch.empty((self._cap_total,) + tuple(v.shape[1:]),
+                               dtype=v.dtype, device=v.device)
+                for k, v in columns.items()}
+        for k, v in columns.items():
+            self._pool[k][self._size:self._size + n] = v
+        self._size += n
 
     def retrieve(self):
         import torch
         want = min(self._batch_size, self._size)
-        any_col = next(iter(self._columns.values()))
-        device = any_col.device
+        device = next(iter(self._pool.values())).device
         perm = torch.randperm(self._size, device=device,
                               generator=self._torch_gen(device))
-        batch_idx, keep_idx = perm[:want], perm[want:]
+        batch_idx = perm[:want]
         batch = {k: v.index_select(0, batch_idx)
-                 for k, v in self._columns.items()}
-        self._columns = {k: v.index_select(0, keep_idx)
-                         for k, v in self._columns.items()}
-        self._size -= want
+                 for k, v in self._pool.items()}
+        # back-fill the removed slots from the tail (vectorized
+        # swap-with-last).  A tail row chosen for the batch would be
+        # clobbered by the scatter, so only relocate tail rows that SURVIVE,
+        # into surviving holes below the new size.
+        new_size = self._size - want
+        in_tail = batch_idx >= new_size
+        holes = batch_idx[~in_tail]                  # empty slots below cut
+        if holes.numel():
+            # surviving tail rows = tail slots NOT picked into the batch;
+            # their count equals the number of holes by construction
+            tail = torch.arange(new_size, self._size, device=device)
+            picked_tail = batch_idx[in_tail] - new_size
+            tail_mask = torch.ones(want, dtype=torch.bool, device=device)
+            tail_mask[picked_tail] = False
+            movers = tail[tail_mask]
+            for k, v in self._pool.items():
+                v[holes] = v.index_select(0, movers)
+        self._size = new_size
         return batch
 
     def can_add(self):
