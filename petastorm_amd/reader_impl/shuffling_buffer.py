@@ -226,11 +226,22 @@ class BatchedRandomShufflingBuffer(ShufflingBufferBase):
         if self._size + n > self._capacity + self._extra_capacity:
             raise RuntimeError('Buffer exceeded capacity+extra_capacity')
         if self._pool is None:
-            self._cap_total = self._capacity + self._extra_capacity
+            # size for capacity + one add-chunk of headroom; grows on demand
+            # (extra_capacity is an upper BOUND, not a preallocation)
+            self._cap_total = self._capacity + n
             self._pool = {
                 k: torch.empty((self._cap_total,) + tuple(v.shape[1:]),
                                dtype=v.dtype, device=v.device)
                 for k, v in columns.items()}
+        elif self._size + n > self._cap_total:
+            new_cap = min(self._capacity + self._extra_capacity,
+                          max(self._cap_total * 2, self._size + n))
+            for k, v in self._pool.items():
+                grown = torch.empty((new_cap,) + tuple(v.shape[1:]),
+                                    dtype=v.dtype, device=v.device)
+                grown[:self._size] = v[:self._size]
+                self._pool[k] = grown
+            self._cap_total = new_cap
         for k, v in columns.items():
             self._pool[k][self._size:self._size + n] = v
         self._size += n
