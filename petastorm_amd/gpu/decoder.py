@@ -84,6 +84,25 @@ class GpuRowGroupDecoder(object):
         self._staging_free = {}
         self._staging_inuse = []
 
+    def _up_many(self, tensors):
+        """Upload a dict of CPU tensors with ONE staged copy per dtype:
+        tensors are flattened and concatenated per dtype, uploaded once, and
+        returned as reshaped views of the device buffer.  Collapses the ~26
+        per-row-group jpeg metadata uploads into 3 copies."""
+        by_dtype = {}
+        for k, t in tensors.items():
+            by_dtype.setdefault(t.dtype, []).append(k)
+        out = {}
+        for dtype, keys in by_dtype.items():
+            flat = torch.cat([tensors[k].reshape(-1) for k in keys])
+            dev_flat = self._up(flat)
+            pos = 0
+            for k in keys:
+                n = tensors[k].numel()
+                out[k] = dev_flat[pos:pos + n].view(tensors[k].shape)
+                pos += n
+        return out
+
     def _up(self, arr):
         """Async host->device upload of a small numpy array / cpu tensor.
 
@@ -847,10 +866,11 @@ class GpuRowGroupDecoder(object):
         widths = meta['width'].numpy()
         heights = meta['height'].numpy()
         ncomp = meta['ncomp'].numpy()
-        # move every tensor to device
-        meta_dev = {}
-        for k, v in meta.items():
-            meta_dev[k] = self._up(v) if isinstance(v, torch.Tensor) else v
+        # move every tensor to device (batched: one copy per dtype)
+        tensors = {k: v for k, v in meta.items()
+                   if isinstance(v, torch.Tensor)}
+        meta_dev = dict(meta)
+        meta_dev.update(self._up_many(tensors))
         block_total = int(meta['block_total'])
         samp_total = int(meta['samp_total'])
         coef = torch.zeros(block_total * 64, dtype=torch.float32, device=dev)
