@@ -216,8 +216,47 @@ class GpuRowGroupDecoder(object):
                                     host_buf, ho['off'], ho['len'])
                     except RuntimeError:
                         pass  # unsupported flavor -> device/CPU path decides
+            if ch['compression'] == 'LZ4':
+                entry['lz4_blocks'] = self._lz4_parse_framing(
+                    host_buf, pages)
             plan[name] = entry
         return plan
+
+    @staticmethod
+    def _lz4_parse_framing(host_buf, pages):
+        """Parquet codec LZ4 is ambiguous: the deprecated Hadoop framing
+        ([4B BE dlen][4B BE clen][lz4 block])* or a single raw LZ4 block
+        (LZ4_RAW).  Detect per page the way Arrow does — accept the framing
+        only if it parses EXACTLY (consumes the page, produces uncomp_size)
+        — and emit a flat raw-block list for the kernel:
+        (page_idx, src_off, src_len, dst_rel, dst_len) arrays."""
+        hb = host_buf.numpy()
+        offs = pages['data_off'].numpy()
+        csz = pages['comp_size'].numpy()
+        usz = pages['uncomp_size'].numpy()
+        pg, src, slen, dst, dlen_a = [], [], [], [], []
+        for i in range(len(offs)):
+            off, clen, ulen = int(offs[i]), int(csz[i]), int(usz[i])
+            blocks = []
+            pos, out = 0, 0
+            while pos + 8 <= clen:
+                dl = int.from_bytes(hb[off + pos:off + pos + 4], 'big')
+                cl = int.from_bytes(hb[off + pos + 4:off + pos + 8], 'big')
+                if cl <= 0 or pos + 8 + cl > clen or out + dl > ulen:
+                    break
+                blocks.append((off + pos + 8, cl, out, dl))
+                pos += 8 + cl
+                out += dl
+            if not (pos == clen and out == ulen):
+                blocks = [(off, clen, 0, ulen)]  # raw single block
+            for s, sl, dr, dl in blocks:
+                pg.append(i); src.append(s); slen.append(sl)
+                dst.append(dr); dlen_a.append(dl)
+        return {'page': np.asarray(pg, dtype=np.int64),
+                'src': np.asarray(src, dtype=np.int64),
+                'src_len': np.asarray(slen, dtype=np.int64),
+                'dst_rel': np.asarray(dst, dtype=np.int64),
+                'dst_len': np.asarray(dlen_a, dtype=np.int64)}
 
     def decode(self, host_buf, chunk_meta, schema, host_plan=None):
         """Decode the requested columns.  Returns dict name ->
@@ -233,7 +272,7 @@ class GpuRowGroupDecoder(object):
         for ch in chunk_meta['chunks']:
             name = ch['name']
             comp = ch['compression']
-            if comp not in ('UNCOMPRESSED', 'SNAPPY', 'GZIP'):
+            if comp not in ('UNCOMPRESSED', 'SNAPPY', 'GZIP', 'LZ4'):
                 out[name] = self._cpu_assist_marker(name)
                 continue
             col = self._decode_chunk(ext, dev, dbuf, host_buf, ch,
@@ -257,7 +296,7 @@ class GpuRowGroupDecoder(object):
         num_values = pages['num_values'].numpy()
         encoding = pages['encoding'].numpy()
         n_pages = len(page_type)
-        snappy = ch['compression'] in ('SNAPPY', 'GZIP')
+        snappy = ch['compression'] in ('SNAPPY', 'GZIP', 'LZ4')
 
         # V2 pages compress ONLY the values section, so they must never go
         # through the whole-page decompression below — dispatch first
@@ -266,6 +305,10 @@ class GpuRowGroupDecoder(object):
         if not data_idx_early:
             return torch.empty(0, device=dev)
         if any(page_type[i] == _PAGE_DATA_V2 for i in data_idx_early):
+            if ch['compression'] == 'LZ4':
+                # V2+LZ4 (compressed values section with possible Hadoop
+                # framing) is rare enough to stay on the assist path
+                return self._cpu_assist_marker(ch['name'])
             return self._decode_v2_chunk(ext, dev, dbuf, ch, pages,
                                          data_idx_early, n_rows)
 
@@ -285,6 +328,20 @@ class GpuRowGroupDecoder(object):
                     ubuf, self._up(u_off[:-1]),
                     self._up(uncomp_size.astype(np.int64)), status)
                 self._check(status, 'snappy:' + ch['name'])
+            elif ch['compression'] == 'LZ4':
+                blocks = plan_entry.get('lz4_blocks')
+                if blocks is None:
+                    blocks = self._lz4_parse_framing(host_buf, pages)
+                blk_src = blocks['src']
+                blk_dst = u_off[blocks['page']] + blocks['dst_rel']
+                bstatus = torch.zeros(len(blk_src), dtype=torch.int32,
+                                      device=dev)
+                ext.lz4_decompress_batch(
+                    dbuf, self._up(blk_src),
+                    self._up(blk_src + blocks['src_len']),
+                    ubuf, self._up(blk_dst),
+                    self._up(blocks['dst_len']), bstatus)
+                self._check(bstatus, 'lz4:' + ch['name'])
             else:  # GZIP: each page is one gzip member -> inflate kernel
                 produced = torch.zeros(n_pages, dtype=torch.int64,
                                        device=dev)

@@ -14,6 +14,12 @@ void snappy_decompress_batch(torch::Tensor comp, torch::Tensor comp_start,
                              torch::Tensor comp_end, torch::Tensor out,
                              torch::Tensor out_offsets,
                              torch::Tensor out_len, torch::Tensor status);
+// lz4.hip
+void lz4_decompress_batch(torch::Tensor comp, torch::Tensor blk_start,
+                          torch::Tensor blk_end, torch::Tensor out,
+                          torch::Tensor out_offsets, torch::Tensor out_len,
+                          torch::Tensor status);
+
 // parquet_decode.hip
 void rle_hybrid_decode_batch(torch::Tensor data, torch::Tensor start,
                              torch::Tensor end, torch::Tensor bit_width,
@@ -76,6 +82,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "petastorm_amd MI355X (gfx950) decode kernels";
   m.def("snappy_decompress_batch", &psa::snappy_decompress_batch,
         "Batched snappy page decompression (wave-per-page)");
+  m.def("lz4_decompress_batch", &psa::lz4_decompress_batch,
+        "batch LZ4 block decompression, one wave per block");
   m.def("rle_hybrid_decode_batch", &psa::rle_hybrid_decode_batch,
         "Parquet RLE/bit-packed hybrid decode (levels & dict indices)");
   m.def("byte_array_offsets_batch", &psa::byte_array_offsets_batch,

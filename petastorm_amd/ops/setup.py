@@ -19,6 +19,7 @@ SRC = [
     'csrc/thrift_pages.cpp',
     'csrc/png_host.cpp',
     'csrc/snappy.hip',
+    'csrc/lz4.hip',
     'csrc/parquet_decode.hip',
     'csrc/jpeg.hip',
     'csrc/transforms.hip',

@@ -26,6 +26,7 @@ class _StubExt(object):
                 'byte_array_offsets_batch', 'varlen_gather',
                 'npy_payload_offsets', 'plain_fixed_decode_batch',
                 'nhwc_to_nchw_normalize', 'jpeg_decode_batch',
+                'lz4_decompress_batch',
                 'inflate_batch', 'png_unfilter_batch', 'bswap16'}
 
     def __init__(self):
@@ -130,3 +131,45 @@ def test_dryrun_compressed_ndarray(stub_decoder, tmp_path):
     z = dec.decode_compressed_ndarray_column(out['mat'], sch.fields['mat'])
     assert z.shape == (20, 8, 4)
     assert 'inflate_batch' in stub.calls
+
+
+def test_dryrun_lz4_scalar(stub_decoder, tmp_path):
+    """LZ4 chunk: framing parse + python orchestration (kernel stubbed)."""
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    dec, stub = stub_decoder
+    url = 'file://' + str(tmp_path / 'lz4ds')
+    create_scalar_dataset(url, num_rows=2000, rowgroup_size=1000,
+                          compression='lz4')
+    out = _decode_all(dec, url, ['id', 'f0'])
+    assert 'lz4_decompress_batch' in stub.calls
+    assert not dec.cpu_assist_columns
+
+
+def test_lz4_parse_framing_hadoop_and_raw():
+    """Host-side framing detection: exact Hadoop framing is split into
+    blocks; anything else is a single raw LZ4 block."""
+    from petastorm_amd.gpu.decoder import GpuRowGroupDecoder
+    blk1 = b'\x12' * 40          # opaque "compressed" payload
+    blk2 = b'\x34' * 24
+    framed = (len(blk1) + 4).to_bytes(4, 'big') + \
+        len(blk1).to_bytes(4, 'big') + blk1 + \
+        (60 - (len(blk1) + 4)).to_bytes(4, 'big') + \
+        len(blk2).to_bytes(4, 'big') + blk2
+    raw = b'\xAB' * 50
+    buf = torch.from_numpy(
+        np.frombuffer(framed + raw, dtype=np.uint8).copy())
+    pages = {
+        'data_off': torch.tensor([0, len(framed)], dtype=torch.int64),
+        'comp_size': torch.tensor([len(framed), len(raw)],
+                                  dtype=torch.int64),
+        'uncomp_size': torch.tensor([60, 777], dtype=torch.int64),
+    }
+    b = GpuRowGroupDecoder._lz4_parse_framing(buf, pages)
+    # page 0: two hadoop blocks; page 1: raw fallback
+    np.testing.assert_array_equal(b['page'], [0, 0, 1])
+    np.testing.assert_array_equal(b['src'], [8, 8 + len(blk1) + 8,
+                                             len(framed)])
+    np.testing.assert_array_equal(b['src_len'], [len(blk1), len(blk2), 50])
+    np.testing.assert_array_equal(b['dst_rel'], [0, len(blk1) + 4, 0])
+    np.testing.assert_array_equal(b['dst_len'], [len(blk1) + 4,
+                                                 60 - (len(blk1) + 4), 777])

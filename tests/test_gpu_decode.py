@@ -190,7 +190,7 @@ def test_varlen_gather_misaligned(ext):
 # full rowgroup decode vs pyarrow (the CPU oracle)
 # ---------------------------------------------------------------------------
 
-@pytest.mark.parametrize('compression', ['snappy', 'none'])
+@pytest.mark.parametrize('compression', ['snappy', 'none', 'lz4'])
 def test_scalar_rowgroup_decode_vs_pyarrow(ext, tmp_path, compression):
     import pyarrow.parquet as pq
     from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
@@ -782,3 +782,60 @@ def test_dictionary_byte_array_ndarray_decode(ext, tmp_path):
         oracle = pf.read_row_group(rg, columns=['mat']).column('mat')
         exp = np.stack([np.load(_io.BytesIO(v.as_py())) for v in oracle])
         np.testing.assert_array_equal(decoded.cpu().numpy(), exp)
+
+
+def test_lz4_kernel_direct(ext):
+    """Handwritten LZ4 blocks: literals, a long run-replicating match, and
+    multi-byte length extensions; plus a short-offset (off < lanes) match."""
+    dev = torch.device('cuda')
+    rng = np.random.RandomState(7)
+
+    # block A: 8 literals + matchlen-84 run copy (offset 8) + 3 literals
+    expected_a = b'abcdefgh' * 11 + b'XYZ'   # 8 + 80 + 3
+    blk_a = bytes([0x8F]) + b'abcdefgh' + b'\x08\x00' + bytes([80 - 4 - 15])
+    blk_a += bytes([0x30]) + b'XYZ'
+
+    # block B: 300 literals only (length extension 15+255+30)
+    lits = rng.randint(0, 256, 300, dtype=np.uint8).tobytes()
+    blk_b = bytes([0xF0, 255, 30]) + lits
+    expected_b = lits
+
+    # block C: offset 3 < wave width -> i % off replication path
+    expected_c = b'abc' * 40 + b'Q'          # 3 + 117 + 1
+    blk_c = bytes([0x3F]) + b'abc' + b'\x03\x00' + bytes([117 - 4 - 15])
+    blk_c += bytes([0x10]) + b'Q'
+
+    comp = torch.from_numpy(np.frombuffer(
+        blk_a + blk_b + blk_c, dtype=np.uint8).copy()).to(dev)
+    starts = torch.tensor([0, len(blk_a), len(blk_a) + len(blk_b)],
+                          dtype=torch.int64, device=dev)
+    ends = torch.tensor([len(blk_a), len(blk_a) + len(blk_b),
+                         len(blk_a) + len(blk_b) + len(blk_c)],
+                        dtype=torch.int64, device=dev)
+    lens = [len(expected_a), len(expected_b), len(expected_c)]
+    out_off = torch.tensor([0, lens[0], lens[0] + lens[1]],
+                           dtype=torch.int64, device=dev)
+    out = torch.empty(sum(lens) + 16, dtype=torch.uint8, device=dev)
+    status = torch.zeros(3, dtype=torch.int32, device=dev)
+    ext.lz4_decompress_batch(comp, starts, ends, out,
+                             out_off, torch.tensor(lens, dtype=torch.int64,
+                                                   device=dev), status)
+    torch.cuda.synchronize()
+    assert status.cpu().tolist() == [0, 0, 0]
+    got = bytes(out.cpu().numpy().tobytes())
+    assert got[:lens[0]] == expected_a
+    assert got[lens[0]:lens[0] + lens[1]] == expected_b
+    assert got[lens[0] + lens[1]:sum(lens)] == expected_c
+
+
+def test_lz4_reader_end_to_end(ext, tmp_path):
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    from petastorm_amd import make_batch_reader
+    url = 'file://' + str(tmp_path / 'lz4')
+    create_scalar_dataset(url, num_rows=3000, rowgroup_size=1000,
+                          compression='lz4')
+    with make_batch_reader(url, device='cuda', shuffle_row_groups=False,
+                           schema_fields=['id', 'f0']) as r:
+        ids = torch.cat([b.id for b in r]).cpu().numpy()
+        assert not r.diagnostics['cpu_assist_columns']
+    np.testing.assert_array_equal(np.sort(ids), np.arange(3000))
