@@ -85,6 +85,9 @@ class GpuRowGroupDecoder(object):
         # ~1ms; recycling makes _up() allocation-free in steady state)
         self._staging_free = {}
         self._staging_inuse = []
+        # pinned scalar verdicts for take_pending()'s dispatch-time status
+        # reduction (recycled)
+        self._host_scalar_free = []
 
     def _up(self, arr):
         """Async host->device upload of a small numpy array / cpu tensor.
@@ -718,25 +721,50 @@ class GpuRowGroupDecoder(object):
         """Snapshot-and-clear this row-group's queued kernel statuses and
         in-flight pinned staging buffers; pass the snapshot to
         :meth:`check_and_recycle` when the row-group is consumed.  Lets the
-        reader keep several row-groups of GPU work in flight."""
-        snap = (self._pending_status, self._staging_inuse)
+        reader keep several row-groups of GPU work in flight.
+
+        The status verdict is REDUCED AND COPIED TO PINNED HOST MEMORY
+        *now*, behind this row-group's kernels on the stream, with an event
+        recorded after it.  Checking later only waits on the event — issuing
+        the D2H at check time instead would enqueue it behind every
+        later-dispatched row-group (the stream is FIFO) and was measured at
+        ~0.8 ms of sync wait per step."""
+        pending = self._pending_status
         self._pending_status = []
+        staging = self._staging_inuse
         self._staging_inuse = []
-        return snap
+        ev = host_total = None
+        if self.device.type == 'cuda' and torch.cuda.is_available():
+            if pending:
+                total = torch.stack(
+                    [s.abs().sum() for _, s in pending]).sum()
+                host_total = (self._host_scalar_free.pop()
+                              if self._host_scalar_free else
+                              torch.empty((), dtype=torch.int64,
+                                          pin_memory=True))
+                host_total.copy_(total, non_blocking=True)
+            if pending or staging:
+                ev = torch.cuda.Event()
+                ev.record()
+        return (pending, staging, ev, host_total)
 
     def check_and_recycle(self, snapshot):
-        """One sync: verify the snapshot's kernel statuses are clean.  The
-        sync also proves its async uploads landed, so the pinned staging
-        buffers go back to the free pool."""
-        pending, staging = snapshot
-        if not pending:
-            if staging:
-                torch.cuda.synchronize(self.device)
-                self._recycle_staging(staging)
-            return
-        total = torch.stack([s.abs().sum() for _, s in pending]).sum()
-        bad = int(total.item()) != 0
+        """Wait for the snapshot's pre-recorded event (completes right after
+        ITS kernels, independent of later dispatches) and verify its pinned
+        status verdict.  The event also proves the async uploads landed, so
+        the pinned staging buffers go back to the free pool."""
+        pending, staging, ev, host_total = snapshot
+        if ev is not None:
+            ev.synchronize()
+        elif pending:  # cpu decoder (dry runs): check synchronously
+            host_total = torch.stack(
+                [s.abs().sum() for _, s in pending]).sum()
         self._recycle_staging(staging)
+        if host_total is None:
+            return
+        bad = int(host_total.item()) != 0
+        if ev is not None:
+            self._host_scalar_free.append(host_total)
         if bad:
             for what, s in pending:
                 vals = s.cpu()
