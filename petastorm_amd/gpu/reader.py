@@ -157,6 +157,7 @@ class GpuBatchReader(object):
         self._cache = HbmCache(cache_size_limit) \
             if cache_type == 'hbm' and cache_size_limit else None
         self._inflight_hosts = []
+        self._cnt_free = []  # pinned row-count scalars (async predicate)
         self._pipeline_depth = max(1, int(pipeline_depth))
         self._io_threads = max(1, int(io_threads))
 
@@ -287,12 +288,18 @@ class GpuBatchReader(object):
             dispatched = self._piece_pos  # pieces fully processed so far
 
             def emit(entry):
-                nt, snap, pos = entry
+                piece, columns, pmeta, snap, pos = entry
                 t3 = time.perf_counter()
                 self._decoder.check_and_recycle(snap)
                 self.stage_times['flush'] += time.perf_counter() - t3
-                if nt is None:
+                if columns is None:
                     return None
+                t3 = time.perf_counter()
+                batch = self._postprocess_emit(piece, columns, pmeta)
+                self.stage_times['postprocess'] += time.perf_counter() - t3
+                if batch is None:
+                    return None
+                nt = self.schema.make_namedtuple(**batch)
                 self._piece_pos = pos
                 self._rows_epoch += len(nt[0])
                 return nt
@@ -333,12 +340,11 @@ class GpuBatchReader(object):
                 self._reclaim_hosts()
                 t2 = time.perf_counter()
                 with _TraceRange('psa.postprocess'):
-                    batch = self._postprocess(piece, columns)
+                    cols, pmeta = self._postprocess(piece, columns)
                 self.stage_times['postprocess'] += time.perf_counter() - t2
                 dispatched += 1
-                nt = self.schema.make_namedtuple(**batch) \
-                    if batch is not None else None
-                pending.append((nt, self._decoder.take_pending(), dispatched))
+                pending.append((piece, cols, pmeta,
+                                self._decoder.take_pending(), dispatched))
                 while len(pending) >= self._pipeline_depth:
                     out = emit(pending.popleft())
                     if out is not None:
@@ -415,23 +421,57 @@ class GpuBatchReader(object):
 
     # ------------------------------------------------------------------
     def _postprocess(self, piece, columns):
+        """Dispatch-time half of postprocessing.  The predicate compaction
+        is enqueued fully ASYNC: a stable argsort of the (negated) mask
+        permutes kept rows to the front while the row count lands in pinned
+        memory behind this row-group's kernels; ``_postprocess_emit`` does
+        the final slice after the pipeline's event wait.  ``torch.nonzero``
+        here would sync the stream and serialize the whole pipeline
+        (measured 21 ms/row-group on the ngram config).
+
+        Returns ``(columns, pinned_count_or_None)``; ``(None, None)`` for
+        empty row-groups."""
         columns = dict(columns)
-        n = None
+        n = 0
         for v in columns.values():
             n = len(v)
             break
         if not n:
-            return None
+            return None, None
+        host_cnt = None
         if self._predicate is not None:
             mask = self._predicate_mask(columns)
             if mask is not None:
-                idx = torch.nonzero(mask, as_tuple=False).squeeze(1)
-                if idx.numel() == 0:
-                    return None
-                columns = {k: (v.index_select(0, idx)
-                               if isinstance(v, torch.Tensor)
-                               else v[idx.cpu().numpy()])
-                           for k, v in columns.items()}
+                if isinstance(mask, torch.Tensor) and mask.is_cuda and \
+                        all(isinstance(v, torch.Tensor)
+                            for v in columns.values()):
+                    host_cnt = (self._cnt_free.pop() if self._cnt_free
+                                else torch.empty((), dtype=torch.int64,
+                                                 pin_memory=True))
+                    host_cnt.copy_(mask.sum(), non_blocking=True)
+                    order = torch.argsort(
+                        mask.logical_not().to(torch.uint8), stable=True)
+                    columns = {k: v.index_select(0, order)
+                               for k, v in columns.items()}
+                else:  # host mask / non-tensor columns: sync path
+                    idx = torch.nonzero(mask, as_tuple=False).squeeze(1)
+                    if idx.numel() == 0:
+                        return None, None
+                    columns = {k: (v.index_select(0, idx)
+                                   if isinstance(v, torch.Tensor)
+                                   else v[idx.cpu().numpy()])
+                               for k, v in columns.items()}
+        return columns, host_cnt
+
+    def _postprocess_emit(self, piece, columns, host_cnt):
+        """Emit-time half: slice the predicate-kept rows (count now valid
+        — check_and_recycle waited the event), then shuffle/transform."""
+        if host_cnt is not None:
+            k = int(host_cnt.item())
+            self._cnt_free.append(host_cnt)
+            if k == 0:
+                return None
+            columns = {name: v[:k] for name, v in columns.items()}
         if self._shuffle_rows:
             n2 = len(next(iter(columns.values())))
             g = None
