@@ -245,6 +245,14 @@ def bench_ngram(args, rank, world, device, dist):
         b = next(it)
         return b['tokens'].shape[0]
 
+    # this config MEASURES the HBM-cache tier (BASELINE config 5: epoch
+    # re-reads served from HBM): warmup must cover the cold first epoch so
+    # the timed steps hit the cache steady state.  The raised warmup is
+    # reported in the JSON line.
+    per_rank_rows = n_rows // max(1, world)
+    batch_rows = args.batch_size * 4
+    args.warmup = max(args.warmup,
+                      (per_rank_rows + batch_rows - 1) // batch_rows + 8)
     result = _run_timed(args, step, device, dist, world)
     reader.stop()
     reader.join()
