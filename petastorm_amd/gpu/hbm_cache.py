@@ -38,7 +38,18 @@ class HbmCache(CacheBase):
         if key in self._store:
             self.hits += 1
             self._store.move_to_end(key)
-            return self._store[key]
+            value = self._store[key]
+            # multi-stream readers gather from cached tensors on their own
+            # stream: mark that use so a later LRU eviction cannot hand the
+            # blocks back to the allocator while the gather is in flight
+            import torch
+            if torch.cuda.is_available():
+                cur = torch.cuda.current_stream()
+                for v in (value.values() if isinstance(value, dict)
+                          else [value]):
+                    if isinstance(v, torch.Tensor) and v.is_cuda:
+                        v.record_stream(cur)
+            return value
         self.misses += 1
         value = fill_cache_func()
         nb = _nbytes(value)

@@ -19,6 +19,7 @@ boundaries all-gather per-rank row counts (petastorm_amd.parallel.epochs).
 """
 
 import logging
+import contextlib
 import os
 import queue
 import threading
@@ -109,7 +110,7 @@ class GpuBatchReader(object):
                  num_epochs=1, cur_shard=None, shard_count=None, seed=None,
                  transform_spec=None, filters=None, device='cuda',
                  cache_type=None, cache_size_limit=None, pipeline_depth=3,
-                 io_threads=2):
+                 io_threads=2, decode_streams=3):
         if isinstance(schema_fields, NGram):
             raise NotImplementedError('NGram is a make_reader feature; use '
                                       'the sequence reader path')
@@ -160,6 +161,18 @@ class GpuBatchReader(object):
         self._cnt_free = []  # pinned row-count scalars (async predicate)
         self._pipeline_depth = max(1, int(pipeline_depth))
         self._io_threads = max(1, int(io_threads))
+        # row-groups are independent, so each one decodes on its own HIP
+        # stream (round-robin): row-group N+1's Huffman/snappy waves overlap
+        # N's IDCT/color/transform and fill the chip — a single in-order
+        # stream serializes them (jpeg_huffman alone launches only ~390
+        # waves on a 256-CU part).  Emit order is unchanged; the per-row-
+        # group event (take_pending) is recorded on the owning stream.
+        n_streams = max(1, int(decode_streams))
+        if self.device.type == 'cuda' and torch.cuda.is_available():
+            self._streams = [torch.cuda.Stream(device=self.device)
+                             for _ in range(n_streams)]
+        else:
+            self._streams = None
 
         # per-file metadata handles (footer parse once per file)
         self._file_md = {}
@@ -294,6 +307,15 @@ class GpuBatchReader(object):
                 self.stage_times['flush'] += time.perf_counter() - t3
                 if columns is None:
                     return None
+                if self._streams:
+                    # tensors were produced on the row-group's own stream;
+                    # the event wait above proves the DATA ready, but the
+                    # caching allocator must also not hand their blocks to
+                    # that stream before the consumer stream is done
+                    cur = torch.cuda.current_stream(self.device)
+                    for v in columns.values():
+                        if isinstance(v, torch.Tensor) and v.is_cuda:
+                            v.record_stream(cur)
                 t3 = time.perf_counter()
                 batch = self._postprocess_emit(piece, columns, pmeta)
                 self.stage_times['postprocess'] += time.perf_counter() - t3
@@ -318,33 +340,14 @@ class GpuBatchReader(object):
                 if kind == 'error':
                     raise piece
                 rr = _next_live(qs, rr)
-                if kind == 'cached':
-                    columns = self._cache._store[self._cache_key(piece)]
-                    self._cache.get(self._cache_key(piece), lambda: columns)
-                else:
-                    meta, plan = meta
-                    if self._cache is not None:
-                        columns = self._cache.get(
-                            self._cache_key(piece),
-                            lambda: self._decode_piece(piece, host, meta,
-                                                       plan))
-                    else:
-                        columns = self._decode_piece(piece, host, meta, plan)
-                    if host is not None:
-                        # the async H2D of `host` may still be in flight;
-                        # reclaim the pinned buffer only after an event
-                        # recorded behind it completes
-                        ev = torch.cuda.Event()
-                        ev.record()
-                        self._inflight_hosts.append((ev, host))
-                self._reclaim_hosts()
-                t2 = time.perf_counter()
-                with _TraceRange('psa.postprocess'):
-                    cols, pmeta = self._postprocess(piece, columns)
-                self.stage_times['postprocess'] += time.perf_counter() - t2
+                stream_ctx = (
+                    torch.cuda.stream(
+                        self._streams[dispatched % len(self._streams)])
+                    if self._streams else contextlib.nullcontext())
+                with stream_ctx:
+                    entry = self._dispatch_one(piece, kind, host, meta)
                 dispatched += 1
-                pending.append((piece, cols, pmeta,
-                                self._decoder.take_pending(), dispatched))
+                pending.append(entry + (dispatched,))
                 while len(pending) >= self._pipeline_depth:
                     out = emit(pending.popleft())
                     if out is not None:
@@ -357,6 +360,34 @@ class GpuBatchReader(object):
                 t.join()
             epoch_sync.epoch_end_sync(self._rows_epoch)
             epoch += 1
+
+    def _dispatch_one(self, piece, kind, host, meta):
+        """Decode + dispatch-phase postprocess of one row-group on the
+        CURRENT stream; returns a pending-pipeline entry (sans position)."""
+        if kind == 'cached':
+            columns = self._cache._store[self._cache_key(piece)]
+            self._cache.get(self._cache_key(piece), lambda: columns)
+        else:
+            meta, plan = meta
+            if self._cache is not None:
+                columns = self._cache.get(
+                    self._cache_key(piece),
+                    lambda: self._decode_piece(piece, host, meta, plan))
+            else:
+                columns = self._decode_piece(piece, host, meta, plan)
+            if host is not None:
+                # the async H2D of `host` may still be in flight; reclaim
+                # the pinned buffer only after an event recorded behind it
+                # (on this row-group's stream) completes
+                ev = torch.cuda.Event()
+                ev.record()
+                self._inflight_hosts.append((ev, host))
+        self._reclaim_hosts()
+        t2 = time.perf_counter()
+        with _TraceRange('psa.postprocess'):
+            cols, pmeta = self._postprocess(piece, columns)
+        self.stage_times['postprocess'] += time.perf_counter() - t2
+        return (piece, cols, pmeta, self._decoder.take_pending())
 
     # ------------------------------------------------------------------
     def _decode_piece(self, piece, host, meta, plan=None):
