@@ -149,7 +149,13 @@ class GpuBatchReader(object):
             raise NoDataAvailableError(
                 'Number of row-groups ({}) < shard_count ({})'
                 .format(len(self._pieces), shard_count))
-        self._seed = seed
+        # agree a base seed ONCE (rank-0 broadcast if distributed and the
+        # user gave none): every epoch permutation is then derived locally,
+        # keeping the data path collective-free (see parallel/epochs.py)
+        self._seed = epoch_sync.agree_seed(seed) \
+            if (shuffle_row_groups and
+                (seed is not None or epoch_sync._dist() is not None)) \
+            else seed
         self._shuffle_row_groups = shuffle_row_groups
         self._shuffle_rows = shuffle_rows
         self._predicate = predicate
@@ -202,11 +208,11 @@ class GpuBatchReader(object):
             n, epoch, self._seed, self._shuffle_row_groups)
         if self._shard_count is not None:
             if epoch_sync._dist() is not None:
-                # With a live process group every rank runs the SAME number
-                # of epoch-boundary collectives, so shards must be equal:
-                # drop the remainder row-groups (DistributedSampler-style;
-                # which groups are dropped rotates with the epoch
-                # permutation, so coverage evens out across epochs).
+                # Equal shards (DistributedSampler-style remainder drop)
+                # keep per-rank epochs the same LENGTH, so explicit
+                # epoch_stats() calls and checkpoint cursors line up across
+                # ranks; which groups are dropped rotates with the epoch
+                # permutation, so coverage evens out across epochs.
                 n_even = n - (n % self._shard_count)
                 perm = perm[:n_even]
             perm = [p for pos, p in enumerate(perm)
@@ -358,7 +364,6 @@ class GpuBatchReader(object):
                     yield out
             for t in threads:
                 t.join()
-            epoch_sync.epoch_end_sync(self._rows_epoch)
             epoch += 1
 
     def _dispatch_one(self, piece, kind, host, meta):
@@ -571,6 +576,16 @@ class GpuBatchReader(object):
     # process group, where rank 0's broadcast makes epochs reproducible
     # within one job but NOT across restarts — so persist with seed set).
     # ------------------------------------------------------------------
+    def epoch_stats(self):
+        """All-gather per-rank rows consumed in the current epoch.
+
+        COLLECTIVE: call from the application at a point where every rank
+        calls it together (e.g. the training loop's epoch boundary) — it is
+        deliberately not part of the reader's generator, whose progress is
+        consumption-driven and not synchronized across ranks
+        (parallel/epochs.py module docstring)."""
+        return epoch_sync.epoch_end_sync(self._rows_epoch)
+
     def state_dict(self):
         return {'epoch': self._epoch, 'piece_pos': self._piece_pos,
                 'seed': self._seed}

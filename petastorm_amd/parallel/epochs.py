@@ -3,24 +3,31 @@
 Replaces the reference's "same seed everywhere" convention for sharded
 readers (reference petastorm/reader.py:573-597 — each rank permutes row
 groups with an identical seed and takes ``index % shard_count == cur_shard``)
-with an explicit collective when a process group exists:
+with a design that is collective-free on the data path:
 
-* epoch start: rank 0 samples the epoch's row-group permutation and
-  broadcasts it (1 small message over xGMI), so sharding is consistent even
-  without a user-provided seed.
-* epoch end: all-gather of per-rank consumed row counts + barrier, replacing
-  the reference's implicit per-rank ventilator completion
-  (reference ventilator.py:124-126).
+* reader construction: when no user seed is given, rank 0 samples a base
+  seed and broadcasts it ONCE (``agree_seed`` — all ranks construct the
+  reader together, so this is the only point where lock-step is guaranteed).
+* every epoch's row-group permutation is then derived LOCALLY from
+  (seed, epoch) — identical on every rank with zero messages.
+* per-rank epoch statistics (``epoch_end_sync``) are an EXPLICIT
+  application-called collective, not part of the reader's generator.
+
+Why not a broadcast/all-gather at every epoch boundary (the first design)?
+Ranks do not cross epoch boundaries after the same number of consumer
+steps once anything makes per-row-group yields uneven — a predicate that
+filters different row counts per shard, shuffle-buffer thresholds, or
+uneven storage.  A rank blocking in a boundary collective that its peers
+will only reach k steps later (or never, if their step budget runs out
+first) deadlocks the job; collectives inside a demand-driven generator are
+ordered by CONSUMPTION, which nothing synchronizes.  The data path is
+therefore collective-free by construction, matching the reference's
+convention while fixing its "hope the seeds match" weakness via the
+init-time agreement.
 
 Every function degrades gracefully to single-process semantics when
 torch.distributed is not initialized, and runs on the gloo backend for
 CPU-only tests (world_size > 1 multi-process tests run here without a GPU).
-
-Collective-ordering contract: every rank must execute the same sequence of
-collectives.  GpuBatchReader guarantees it by dropping the per-epoch
-row-group remainder so all shards are equal-sized (readers on different
-ranks then hit epoch_permutation / epoch_end_sync in lock-step regardless of
-relative progress within an epoch).
 """
 
 import numpy as np
@@ -31,34 +38,49 @@ def _dist():
     return dist if dist.is_available() and dist.is_initialized() else None
 
 
-def epoch_permutation(n_items, epoch, seed=None, shuffle=True):
-    """The epoch's row-group order, identical on every rank.
+def agree_seed(seed=None):
+    """A base seed identical on every rank.
 
-    With a process group: rank 0 samples (seeded or not) and broadcasts.
-    Without: seeded local RNG (reference behavior).
+    A user-provided seed is already rank-consistent and returned as-is.
+    Otherwise rank 0 samples one and broadcasts it — called ONCE from
+    reader construction, the only naturally lock-step point.  Without a
+    process group, a locally sampled seed is returned.
     """
     import torch
+    if seed is not None:
+        return int(seed)
+    dist = _dist()
+    local = int(np.random.randint(0, 2 ** 31 - 1))
+    if dist is None:
+        return local
+    device = _collective_device(dist)
+    t = torch.tensor([local if dist.get_rank() == 0 else 0],
+                     dtype=torch.int64, device=device)
+    dist.broadcast(t, src=0)
+    return int(t.cpu().item())
+
+
+def epoch_permutation(n_items, epoch, seed=None, shuffle=True):
+    """The epoch's row-group order — derived locally from (seed, epoch) so
+    every rank computes the same order with no communication.
+
+    ``seed=None`` gives an unseeded local permutation and is only meaningful
+    single-process; distributed readers pass a seed from :func:`agree_seed`.
+    """
     if not shuffle:
         return np.arange(n_items, dtype=np.int64)
-    dist = _dist()
-    if dist is None:
-        rng = np.random.RandomState(
-            None if seed is None else (seed + epoch) % (2 ** 31))
-        return rng.permutation(n_items).astype(np.int64)
-    if dist.get_rank() == 0:
-        rng = np.random.RandomState(
-            None if seed is None else (seed + epoch) % (2 ** 31))
-        perm = torch.from_numpy(rng.permutation(n_items).astype(np.int64))
-    else:
-        perm = torch.empty(n_items, dtype=torch.int64)
-    device = _collective_device(dist)
-    perm = perm.to(device)
-    dist.broadcast(perm, src=0)
-    return perm.cpu().numpy()
+    rng = np.random.RandomState(
+        None if seed is None else (seed + epoch) % (2 ** 31))
+    return rng.permutation(n_items).astype(np.int64)
 
 
 def epoch_end_sync(rows_consumed):
-    """All-gather per-rank consumed-row counts + barrier at an epoch boundary.
+    """All-gather per-rank consumed-row counts + barrier.
+
+    COLLECTIVE — every rank must call it together.  It is deliberately NOT
+    called from the reader's generator (see module docstring); applications
+    call it (or ``GpuBatchReader.epoch_stats()``) at points where all ranks
+    are known to be synchronized, e.g. a training-loop epoch boundary.
 
     Returns the list of per-rank counts (len == world_size), or
     ``[rows_consumed]`` when not distributed.

@@ -15,10 +15,13 @@ def _worker_perm(rank, world, file_store, out_q):
     import torch.distributed as dist
     dist.init_process_group('gloo', init_method='file://' + file_store,
                             rank=rank, world_size=world)
-    from petastorm_amd.parallel.epochs import (epoch_end_sync,
+    from petastorm_amd.parallel.epochs import (agree_seed, epoch_end_sync,
                                                epoch_permutation,
                                                shard_for_rank)
-    perms = [epoch_permutation(20, e, seed=None).tolist() for e in range(3)]
+    # no user seed: ranks agree a base seed ONCE (rank-0 broadcast), then
+    # derive every epoch permutation locally — no per-epoch collective
+    seed = agree_seed(None)
+    perms = [epoch_permutation(20, e, seed=seed).tolist() for e in range(3)]
     counts = epoch_end_sync(100 + rank)
     shard = shard_for_rank()
     out_q.put((rank, perms, counts, shard))
@@ -27,8 +30,10 @@ def _worker_perm(rank, world, file_store, out_q):
 
 @pytest.mark.timeout(120)
 def test_epoch_permutation_broadcast_consistency(tmp_path):
-    """Unseeded permutations must still agree across ranks (rank-0
-    broadcast replaces the reference's same-seed convention)."""
+    """Unseeded runs must still agree across ranks: a single init-time
+    seed agreement (rank-0 broadcast) replaces the reference's same-seed
+    convention; epochs themselves are collective-free (a rank blocking in a
+    per-epoch collective its peers reach later would deadlock)."""
     ctx = mp.get_context('spawn')
     q = ctx.Queue()
     store = str(tmp_path / 'store')
@@ -102,3 +107,42 @@ def test_epoch_permutation_single_process():
 def test_epoch_end_sync_single_process():
     from petastorm_amd.parallel.epochs import epoch_end_sync
     assert epoch_end_sync(42) == [42]
+
+
+def _worker_uneven_steps(rank, world, file_store, url, out_q):
+    """Ranks stop consuming after DIFFERENT step counts mid-epoch: with a
+    collective-free data path this must not deadlock."""
+    import torch.distributed as dist
+    dist.init_process_group('gloo', init_method='file://' + file_store,
+                            rank=rank, world_size=world)
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.predicates import in_lambda
+    pred = in_lambda(['id'], lambda v: (v['id'] % (2 + rank)) != 0)
+    got = 0
+    with make_batch_reader(url, reader_pool_type='dummy', predicate=pred,
+                           shuffle_row_groups=True, num_epochs=None,
+                           cur_shard=rank, shard_count=world) as r:
+        it = iter(r)
+        for _ in range(3 + rank * 4):   # uneven consumption across ranks
+            got += len(next(it).id)
+    out_q.put((rank, got))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_uneven_rank_progress_does_not_deadlock(tmp_path, scalar_dataset):
+    ctx = mp.get_context('spawn')
+    q = ctx.Queue()
+    store = str(tmp_path / 'store3')
+    procs = [ctx.Process(target=_worker_uneven_steps,
+                         args=(r, 2, store, scalar_dataset['url'], q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, got = q.get(timeout=150)
+        results[rank] = got
+    for p in procs:
+        p.join(timeout=30)
+    assert results[0] > 0 and results[1] > 0
