@@ -26,6 +26,8 @@ def _hbm_mb():
 
 def run_config(name, seconds):
     sys.argv = ['bench.py']
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
     import bench
     import argparse as ap
     args = ap.Namespace(gpus=1, steps=10, warmup=2, batch_size=256,
