@@ -267,10 +267,12 @@ class GpuRowGroupDecoder(object):
         only if it parses EXACTLY (consumes the page, produces uncomp_size)
         — and emit a flat raw-block list for the kernel:
         (page_idx, src_off, src_len, dst_rel, dst_len) arrays."""
-        hb = host_buf.numpy()
-        offs = pages['data_off'].numpy()
-        csz = pages['comp_size'].numpy()
-        usz = pages['uncomp_size'].numpy()
+        return GpuRowGroupDecoder._lz4_spans(
+            host_buf.numpy(), pages['data_off'].numpy(),
+            pages['comp_size'].numpy(), pages['uncomp_size'].numpy())
+
+    @staticmethod
+    def _lz4_spans(hb, offs, csz, usz):
         pg, src, slen, dst, dlen_a = [], [], [], [], []
         for i in range(len(offs)):
             off, clen, ulen = int(offs[i]), int(csz[i]), int(usz[i])
@@ -342,12 +344,8 @@ class GpuRowGroupDecoder(object):
         if not data_idx_early:
             return torch.empty(0, device=dev)
         if any(page_type[i] == _PAGE_DATA_V2 for i in data_idx_early):
-            if ch['compression'] == 'LZ4':
-                # V2+LZ4 (compressed values section with possible Hadoop
-                # framing) is rare enough to stay on the assist path
-                return self._cpu_assist_marker(ch['name'])
-            return self._decode_v2_chunk(ext, dev, dbuf, ch, pages,
-                                         data_idx_early, n_rows)
+            return self._decode_v2_chunk(ext, dev, dbuf, host_buf, ch,
+                                         pages, data_idx_early, n_rows)
 
         # 1) page payload location: either in dbuf directly, or in a
         #    decompressed scratch buffer
@@ -532,7 +530,8 @@ class GpuRowGroupDecoder(object):
         self._check(status, 'plainfixed:' + name)
         return out[:total * esize].view(dtype)
 
-    def _decode_v2_chunk(self, ext, dev, dbuf, ch, pages, data_idx, n_rows):
+    def _decode_v2_chunk(self, ext, dev, dbuf, host_buf, ch, pages,
+                         data_idx, n_rows):
         """DataPageV2: levels are stored uncompressed with explicit byte
         lengths; only the values section is compressed (Parquet format
         spec).  Supported for PLAIN fixed-width columns; everything else
@@ -559,7 +558,7 @@ class GpuRowGroupDecoder(object):
         # opt out via is_compressed (writers skip compression when it does
         # not shrink the page)
         v2c = pages['v2_is_compressed'].numpy()[idx].astype(bool)
-        chunk_compressed = ch['compression'] in ('SNAPPY', 'GZIP')
+        chunk_compressed = ch['compression'] in ('SNAPPY', 'GZIP', 'LZ4')
         page_compressed = v2c & chunk_compressed
         n = len(idx)
         if page_compressed.any():
@@ -576,6 +575,20 @@ class GpuRowGroupDecoder(object):
                     dbuf, self._up(val_comp_start[ci]),
                     self._up(v_comp_end[ci]),
                     vbuf, self._up(u_off[ci]), self._up(v_un[ci]), status)
+            elif ch['compression'] == 'LZ4':
+                # values sections may carry the Hadoop framing (see
+                # _lz4_spans); parse on host, decompress raw blocks
+                blocks = self._lz4_spans(
+                    host_buf.numpy(), val_comp_start[ci],
+                    (v_comp_end - val_comp_start)[ci], v_un[ci])
+                status = torch.zeros(len(blocks['src']), dtype=torch.int32,
+                                     device=dev)
+                blk_dst = u_off[ci][blocks['page']] + blocks['dst_rel']
+                ext.lz4_decompress_batch(
+                    dbuf, self._up(blocks['src']),
+                    self._up(blocks['src'] + blocks['src_len']),
+                    vbuf, self._up(blk_dst), self._up(blocks['dst_len']),
+                    status)
             else:
                 produced = torch.zeros(len(ci), dtype=torch.int64,
                                        device=dev)
