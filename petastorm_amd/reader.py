@@ -299,6 +299,12 @@ class Reader(object):
                                  ventilator=self._ventilator)
         self._cache = cache
         self._row_buffer = []
+        # checkpoint/resume bookkeeping (state_dict); deterministic iteration
+        # requires a seed whenever anything shuffles
+        self._seed = seed
+        self._any_shuffle = bool(shuffle_row_groups or shuffle_rows or
+                                 (shuffle_row_drop_partitions or 1) > 1)
+        self._rows_consumed = 0
 
     # ------------------------------------------------------------------
     def _push_down_partition_predicate(self, predicate, selected):
@@ -380,10 +386,12 @@ class Reader(object):
         try:
             if self.batched_output:
                 columns = self._workers_pool.get_results()
+                self._rows_consumed += 1
                 return self.schema.make_namedtuple(**columns)
             while not self._row_buffer:
                 self._row_buffer = list(self._workers_pool.get_results())
             row = self._row_buffer.pop(0)
+            self._rows_consumed += 1
             if self.ngram is not None:
                 return row  # already {timestep: namedtuple}
             return self.schema.make_namedtuple(**row)
@@ -401,6 +409,46 @@ class Reader(object):
                            'produce duplicate or dropped rows')
         self._ventilator.reset()
         self.last_row_consumed = False
+
+    def _check_deterministic(self):
+        if self._any_shuffle and self._seed is None:
+            raise NotImplementedError(
+                'state_dict requires deterministic iteration: pass seed= '
+                'or disable shuffling (shuffle_row_groups/shuffle_rows/'
+                'shuffle_row_drop_partitions)')
+
+    def state_dict(self):
+        """Checkpoint the iterator position (units consumed since
+        construction: rows for make_reader, batches for make_batch_reader).
+
+        The reference has no reader checkpointing; the GPU
+        reader keeps an exact row-group cursor, while this CPU pool path
+        restores by deterministic fast-forward — valid for any seeded (or
+        shuffle-free) configuration, with any pool, because the thread
+        pool's seeded readout is strict round-robin
+        (workers_pool/thread_pool.py) and the dummy pool is inline."""
+        self._check_deterministic()
+        return {'rows_consumed': self._rows_consumed, 'seed': self._seed,
+                'version': 1}
+
+    def load_state_dict(self, state):
+        """Fast-forward a FRESH reader (same constructor arguments) to a
+        :meth:`state_dict` position by consuming and discarding.  Costs the
+        re-decode of the skipped span — the price of exactness with
+        worker-pool parallelism; prefer the GPU reader's O(1) cursor resume
+        for large skips."""
+        self._check_deterministic()
+        if self._rows_consumed:
+            raise RuntimeError('load_state_dict requires a fresh reader')
+        if state.get('seed') != self._seed:
+            raise ValueError('state was captured with a different seed')
+        target = int(state['rows_consumed'])
+        for _ in range(target):
+            try:
+                next(self)
+            except StopIteration:
+                break
+        return self
 
     def stop(self):
         self._stopped = True
