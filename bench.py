@@ -113,7 +113,10 @@ def bench_imagenet(args, rank, world, device, dist):
         url, device=str(device), num_epochs=None, shuffle_row_groups=True,
         seed=1234, transform_spec=ts,
         cur_shard=rank if world > 1 else None,
-        shard_count=world if world > 1 else None)
+        shard_count=world if world > 1 else None,
+        gpu_options=dict(
+            pipeline_depth=int(os.environ.get('PSA_PIPELINE_DEPTH', '3')),
+            decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '3'))))
     loader = BatchedDataLoader(reader, batch_size=args.batch_size)
 
     it = iter(loader)
