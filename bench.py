@@ -115,8 +115,8 @@ def bench_imagenet(args, rank, world, device, dist):
         cur_shard=rank if world > 1 else None,
         shard_count=world if world > 1 else None,
         gpu_options=dict(
-            pipeline_depth=int(os.environ.get('PSA_PIPELINE_DEPTH', '3')),
-            decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '3'))))
+            pipeline_depth=int(os.environ.get('PSA_PIPELINE_DEPTH', '6')),
+            decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '4'))))
     loader = BatchedDataLoader(reader, batch_size=args.batch_size)
 
     it = iter(loader)
