@@ -225,9 +225,21 @@ class GpuRowGroupDecoder(object):
                 host_buf, torch.tensor([ch['offset']], dtype=torch.int64),
                 torch.tensor([ch['length']], dtype=torch.int64))
             pages = entry['pages']
+            # ZSTD chunks decompress HERE (IO thread, native threads, GIL
+            # released) into a pinned buffer; downstream everything —
+            # host offset scans, jpeg parse, the device decode path —
+            # sees an ordinary UNCOMPRESSED chunk view
+            eff_comp = ch['compression']
+            eff_buf = host_buf
+            if ch['compression'] == 'ZSTD' and \
+                    not (pages['page_type'].numpy() == _PAGE_DATA_V2).any():
+                entry['zstd'] = self._zstd_decompress_pages(host_buf, pages)
+                pages = entry['zstd']['pages']
+                eff_buf = entry['zstd']['host_buf']
+                eff_comp = 'UNCOMPRESSED'
             # host-visible PLAIN byte-array of a REQUIRED column: offsets and
             # image headers can be parsed before any GPU work
-            if ch['compression'] == 'UNCOMPRESSED' and \
+            if eff_comp == 'UNCOMPRESSED' and \
                     ch['physical'] == 'BYTE_ARRAY' and ch['max_def'] == 0:
                 ptype = pages['page_type'].numpy()
                 enc = pages['encoding'].numpy()
@@ -237,7 +249,7 @@ class GpuRowGroupDecoder(object):
                     counts = pages['num_values'].numpy()[didx] \
                         .astype(np.int64)
                     ho = ext.byte_array_host_offsets(
-                        host_buf, torch.from_numpy(starts),
+                        eff_buf, torch.from_numpy(starts),
                         torch.from_numpy(counts))
                     entry['host_off'] = ho['off'].numpy()
                     entry['host_len'] = ho['len'].numpy()
@@ -247,10 +259,10 @@ class GpuRowGroupDecoder(object):
                         if isinstance(codec, CompressedImageCodec):
                             if codec.image_codec == 'jpeg':
                                 entry['jpeg_meta'] = ext.jpeg_parse_batch(
-                                    host_buf, ho['off'], ho['len'])
+                                    eff_buf, ho['off'], ho['len'])
                             else:
                                 entry['png_meta'] = ext.png_parse_batch(
-                                    host_buf, ho['off'], ho['len'])
+                                    eff_buf, ho['off'], ho['len'])
                     except RuntimeError:
                         pass  # unsupported flavor -> device/CPU path decides
             if ch['compression'] == 'LZ4':
@@ -258,6 +270,32 @@ class GpuRowGroupDecoder(object):
                     host_buf, pages)
             plan[name] = entry
         return plan
+
+    def _zstd_decompress_pages(self, host_buf, pages):
+        """Decompress every page of a ZSTD chunk (one zstd frame per page)
+        into a fresh pinned buffer using the from-scratch RFC 8878 decoder
+        (ops/csrc/zstd_core.h) on native host threads.  Returns
+        {'host_buf', 'pages'} where 'pages' is the chunk's page table
+        rebased onto the decompressed buffer (an UNCOMPRESSED view)."""
+        offs = pages['data_off'].numpy().astype(np.int64)
+        csz = pages['comp_size'].numpy().astype(np.int64)
+        usz = pages['uncomp_size'].numpy().astype(np.int64)
+        u_off = np.zeros(len(offs), dtype=np.int64)
+        if len(offs) > 1:
+            u_off[1:] = np.cumsum(usz)[:-1]
+        hbuf = torch.empty(int(usz.sum()) + _SLACK, dtype=torch.uint8,
+                           pin_memory=self._pin_memory)
+        status = torch.zeros(len(offs), dtype=torch.int32)
+        self._ext.zstd_decompress_host(
+            host_buf, torch.from_numpy(offs), torch.from_numpy(csz),
+            hbuf, torch.from_numpy(u_off), torch.from_numpy(usz), status)
+        if int(status.abs().sum()):
+            raise RuntimeError('zstd decode error: status={}'
+                               .format(status.tolist()))
+        pages2 = dict(pages)
+        pages2['data_off'] = torch.from_numpy(u_off)
+        pages2['comp_size'] = torch.from_numpy(usz)
+        return {'host_buf': hbuf, 'pages': pages2}
 
     @staticmethod
     def _lz4_parse_framing(host_buf, pages):
@@ -311,7 +349,8 @@ class GpuRowGroupDecoder(object):
         for ch in chunk_meta['chunks']:
             name = ch['name']
             comp = ch['compression']
-            if comp not in ('UNCOMPRESSED', 'SNAPPY', 'GZIP', 'LZ4'):
+            if comp not in ('UNCOMPRESSED', 'SNAPPY', 'GZIP',
+                            'LZ4', 'ZSTD'):
                 out[name] = self._cpu_assist_marker(name)
                 continue
             col = self._decode_chunk(ext, dev, dbuf, host_buf, ch,
@@ -327,6 +366,14 @@ class GpuRowGroupDecoder(object):
     def _decode_chunk(self, ext, dev, dbuf, host_buf, ch, plan_entry, n_rows,
                       schema):
         pages = plan_entry['pages']
+        if ch['compression'] == 'ZSTD':
+            z = plan_entry.get('zstd')
+            if z is None:  # V2+ZSTD (values-section framing): assist path
+                return self._cpu_assist_marker(ch['name'])
+            host_buf = z['host_buf']
+            dbuf = host_buf.to(dev, non_blocking=True)
+            pages = z['pages']
+            ch = dict(ch, compression='UNCOMPRESSED')
         self._plan_entry = plan_entry
         page_type = pages['page_type'].numpy()
         data_off = pages['data_off'].numpy()       # relative to chunk walk

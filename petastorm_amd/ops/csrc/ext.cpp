@@ -20,6 +20,12 @@ void lz4_decompress_batch(torch::Tensor comp, torch::Tensor blk_start,
                           torch::Tensor out_offsets, torch::Tensor out_len,
                           torch::Tensor status);
 
+// zstd_host.cpp
+void zstd_decompress_host(torch::Tensor src, torch::Tensor src_off,
+                          torch::Tensor src_len, torch::Tensor dst,
+                          torch::Tensor dst_off, torch::Tensor dst_len,
+                          torch::Tensor status);
+
 // parquet_decode.hip
 void rle_hybrid_decode_batch(torch::Tensor data, torch::Tensor start,
                              torch::Tensor end, torch::Tensor bit_width,
@@ -84,6 +90,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Batched snappy page decompression (wave-per-page)");
   m.def("lz4_decompress_batch", &psa::lz4_decompress_batch,
         "batch LZ4 block decompression, one wave per block");
+  m.def("zstd_decompress_host", &psa::zstd_decompress_host,
+        "batch ZSTD frame decompression on host threads (from-scratch "
+        "RFC 8878 decoder, GIL released)");
   m.def("rle_hybrid_decode_batch", &psa::rle_hybrid_decode_batch,
         "Parquet RLE/bit-packed hybrid decode (levels & dict indices)");
   m.def("byte_array_offsets_batch", &psa::byte_array_offsets_batch,

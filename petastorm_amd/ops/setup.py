@@ -20,6 +20,7 @@ SRC = [
     'csrc/png_host.cpp',
     'csrc/snappy.hip',
     'csrc/lz4.hip',
+    'csrc/zstd_host.cpp',
     'csrc/parquet_decode.hip',
     'csrc/jpeg.hip',
     'csrc/transforms.hip',

@@ -173,3 +173,60 @@ def test_lz4_parse_framing_hadoop_and_raw():
     np.testing.assert_array_equal(b['dst_rel'], [0, len(blk1) + 4, 0])
     np.testing.assert_array_equal(b['dst_len'], [len(blk1) + 4,
                                                  60 - (len(blk1) + 4), 777])
+
+
+def test_zstd_host_decompress_matches_libzstd_oracle():
+    """From-scratch RFC 8878 decoder vs data compressed by libzstd
+    (ctypes oracle) across levels and data shapes."""
+    import ctypes
+    from petastorm_amd import ops
+    ext = ops.ext()
+    z = ctypes.CDLL('libzstd.so.1')
+    z.ZSTD_compress.restype = ctypes.c_size_t
+    z.ZSTD_compressBound.restype = ctypes.c_size_t
+    rng = np.random.RandomState(3)
+    datas = [
+        bytes(rng.randint(0, 256, 5000, dtype=np.uint8)),
+        bytes(rng.randint(0, 7, 60000, dtype=np.uint8)),
+        (b'abcdef' * 40000),                      # multi-block, matches
+        rng.rand(20000).tobytes(),                # float pages
+        b'x',
+        bytes(200000),                            # zeros / RLE blocks
+    ]
+    comps = []
+    for d, lvl in zip(datas, (1, 3, 9, 19, 1, 5)):
+        buf = ctypes.create_string_buffer(z.ZSTD_compressBound(len(d)))
+        n = z.ZSTD_compress(buf, len(buf.raw), d, len(d), lvl)
+        comps.append(buf.raw[:n])
+    blob = b''.join(comps)
+    src = torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy())
+    src_off, pos = [], 0
+    for c in comps:
+        src_off.append(pos)
+        pos += len(c)
+    src_len = [len(c) for c in comps]
+    dst_len = [len(d) for d in datas]
+    dst_off, pos = [], 0
+    for n in dst_len:
+        dst_off.append(pos)
+        pos += n
+    dst = torch.empty(pos, dtype=torch.uint8)
+    status = torch.zeros(len(datas), dtype=torch.int32)
+    ext.zstd_decompress_host(
+        src, torch.tensor(src_off), torch.tensor(src_len),
+        dst, torch.tensor(dst_off), torch.tensor(dst_len), status)
+    assert status.tolist() == [0] * len(datas)
+    out = dst.numpy().tobytes()
+    for d, o, n in zip(datas, dst_off, dst_len):
+        assert out[o:o + n] == d
+
+
+def test_dryrun_zstd_scalar(stub_decoder, tmp_path):
+    """ZSTD chunk: host decompression (real, native) + stubbed kernels."""
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    dec, stub = stub_decoder
+    url = 'file://' + str(tmp_path / 'zstdds')
+    create_scalar_dataset(url, num_rows=2000, rowgroup_size=1000,
+                          compression='zstd')
+    out = _decode_all(dec, url, ['id', 'f0'])
+    assert not dec.cpu_assist_columns

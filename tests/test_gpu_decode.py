@@ -190,7 +190,7 @@ def test_varlen_gather_misaligned(ext):
 # full rowgroup decode vs pyarrow (the CPU oracle)
 # ---------------------------------------------------------------------------
 
-@pytest.mark.parametrize('compression', ['snappy', 'none', 'lz4'])
+@pytest.mark.parametrize('compression', ['snappy', 'none', 'lz4', 'zstd'])
 def test_scalar_rowgroup_decode_vs_pyarrow(ext, tmp_path, compression):
     import pyarrow.parquet as pq
     from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
@@ -885,3 +885,16 @@ def test_gpu_ngram_reader_matches_cpu(ext, tmp_path):
         t1g, s0g, s1g, tokg = gpu_windows[k]
         assert (t1c, s0c, s1c) == (t1g, s0g, s1g)
         np.testing.assert_array_equal(tokc, tokg)
+
+
+def test_zstd_reader_end_to_end(ext, tmp_path):
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    from petastorm_amd import make_batch_reader
+    url = 'file://' + str(tmp_path / 'zstd')
+    create_scalar_dataset(url, num_rows=3000, rowgroup_size=1000,
+                          compression='zstd')
+    with make_batch_reader(url, device='cuda', shuffle_row_groups=False,
+                           schema_fields=['id', 'f0']) as r:
+        ids = torch.cat([b.id for b in r]).cpu().numpy()
+        assert not r.diagnostics['cpu_assist_columns']
+    np.testing.assert_array_equal(np.sort(ids), np.arange(3000))
