@@ -26,6 +26,13 @@ void zstd_decompress_host(torch::Tensor src, torch::Tensor src_off,
                           torch::Tensor dst_off, torch::Tensor dst_len,
                           torch::Tensor status);
 
+// zstd.hip
+int64_t zstd_work_bytes();
+void zstd_decompress_batch(torch::Tensor src, torch::Tensor src_off,
+                           torch::Tensor src_len, torch::Tensor dst,
+                           torch::Tensor dst_off, torch::Tensor dst_len,
+                           torch::Tensor work, torch::Tensor status);
+
 // parquet_decode.hip
 void rle_hybrid_decode_batch(torch::Tensor data, torch::Tensor start,
                              torch::Tensor end, torch::Tensor bit_width,
@@ -93,6 +100,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("zstd_decompress_host", &psa::zstd_decompress_host,
         "batch ZSTD frame decompression on host threads (from-scratch "
         "RFC 8878 decoder, GIL released)");
+  m.def("zstd_work_bytes", &psa::zstd_work_bytes,
+        "per-frame device workspace size for zstd_decompress_batch");
+  m.def("zstd_decompress_batch", &psa::zstd_decompress_batch,
+        "batch ZSTD frame decompression on device (thread per frame)");
   m.def("rle_hybrid_decode_batch", &psa::rle_hybrid_decode_batch,
         "Parquet RLE/bit-packed hybrid decode (levels & dict indices)");
   m.def("byte_array_offsets_batch", &psa::byte_array_offsets_batch,

@@ -21,6 +21,7 @@ SRC = [
     'csrc/snappy.hip',
     'csrc/lz4.hip',
     'csrc/zstd_host.cpp',
+    'csrc/zstd.hip',
     'csrc/parquet_decode.hip',
     'csrc/jpeg.hip',
     'csrc/transforms.hip',

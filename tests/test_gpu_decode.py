@@ -898,3 +898,42 @@ def test_zstd_reader_end_to_end(ext, tmp_path):
         ids = torch.cat([b.id for b in r]).cpu().numpy()
         assert not r.diagnostics['cpu_assist_columns']
     np.testing.assert_array_equal(np.sort(ids), np.arange(3000))
+
+
+def test_zstd_device_kernel_matches_host(ext, tmp_path):
+    """Thread-per-frame GPU zstd kernel vs the host decoder on
+    libzstd-compressed data."""
+    import ctypes
+    z = ctypes.CDLL('libzstd.so.1')
+    z.ZSTD_compress.restype = ctypes.c_size_t
+    z.ZSTD_compressBound.restype = ctypes.c_size_t
+    rng = np.random.RandomState(11)
+    datas = [bytes(rng.randint(0, 256, 4000, dtype=np.uint8)),
+             bytes(rng.randint(0, 9, 150000, dtype=np.uint8)),
+             (b'seq-' * 50000),
+             rng.rand(30000).tobytes()]
+    comps = []
+    for d, lvl in zip(datas, (1, 3, 9, 19)):
+        buf = ctypes.create_string_buffer(z.ZSTD_compressBound(len(d)))
+        n = z.ZSTD_compress(buf, len(buf.raw), d, len(d), lvl)
+        comps.append(buf.raw[:n])
+    dev = torch.device('cuda')
+    blob = b''.join(comps)
+    src = torch.from_numpy(np.frombuffer(blob, dtype=np.uint8).copy()).to(dev)
+    soff = np.cumsum([0] + [len(c) for c in comps])[:-1]
+    doff = np.cumsum([0] + [len(d) for d in datas])[:-1]
+    dlen = [len(d) for d in datas]
+    dst = torch.empty(int(sum(dlen)), dtype=torch.uint8, device=dev)
+    work = torch.empty(len(datas) * int(ext.zstd_work_bytes()),
+                       dtype=torch.uint8, device=dev)
+    status = torch.zeros(len(datas), dtype=torch.int32, device=dev)
+    ext.zstd_decompress_batch(
+        src, torch.tensor(soff, dtype=torch.int64, device=dev),
+        torch.tensor([len(c) for c in comps], dtype=torch.int64, device=dev),
+        dst, torch.tensor(doff, dtype=torch.int64, device=dev),
+        torch.tensor(dlen, dtype=torch.int64, device=dev), work, status)
+    torch.cuda.synchronize()
+    assert status.cpu().tolist() == [0] * len(datas)
+    out = dst.cpu().numpy().tobytes()
+    for d, o, n in zip(datas, doff, dlen):
+        assert out[o:o + n] == d
