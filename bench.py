@@ -152,9 +152,11 @@ def bench_scalar(args, rank, world, device, dist):
     from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
 
     n_rows = args.rows or 2_000_000
-    url = _dataset_dir('scalar_{}'.format(n_rows), rank, dist,
+    comp = os.environ.get('PSA_SCALAR_COMPRESSION', 'snappy')
+    url = _dataset_dir('scalar_{}_{}'.format(n_rows, comp), rank, dist,
                        lambda u: create_scalar_dataset(
-                           u, num_rows=n_rows, rowgroup_size=62500))
+                           u, num_rows=n_rows, rowgroup_size=62500,
+                           compression=comp))
     reader = make_batch_reader(
         url, device=str(device), num_epochs=None, shuffle_row_groups=True,
         seed=7, schema_fields=['id', 'f0', 'f1', 'f2', 'f3', 'i0', 'i1'],
