@@ -245,7 +245,7 @@ class GpuBatchReader(object):
                     break
                 if self._cache is not None and \
                         self._cache_key(piece) in self._cache._store:
-                    out_q.put(('cached', piece, None, None))
+                    self._q_put(out_q, ('cached', piece, None, None))
                     continue
                 md, pschema = self._metadata(piece.path)
                 t0 = time.perf_counter()
@@ -260,10 +260,25 @@ class GpuBatchReader(object):
                                                   self._storage_schema)
                 self.stage_times['io_read'] += t1 - t0
                 self.stage_times['io_parse'] += time.perf_counter() - t1
-                out_q.put(('data', piece, host, (meta, plan)))
-            out_q.put(('end', None, None, None))
+                if not self._q_put(out_q, ('data', piece, host,
+                                           (meta, plan))):
+                    return
+            self._q_put(out_q, ('end', None, None, None))
         except Exception as e:  # noqa: BLE001 - forwarded to consumer
-            out_q.put(('error', e, None, None))
+            self._q_put(out_q, ('error', e, None, None))
+
+    def _q_put(self, q, item):
+        """Stop-aware bounded put: a blocked IO thread must wake when the
+        reader stops, or process exit can finalize the interpreter while
+        the thread is inside GIL-released native code (observed as a
+        SIGABRT at teardown with slow host decompression in flight)."""
+        while not self._stopped:
+            try:
+                q.put(item, timeout=0.1)
+                return True
+            except queue.Full:
+                continue
+        return False
 
     @staticmethod
     def _cache_key(piece):
@@ -302,6 +317,7 @@ class GpuBatchReader(object):
                 threading.Thread(target=self._io_worker,
                                  args=(pieces[t::n_io], qs[t]), daemon=True)
                 for t in range(n_io)]
+            self._live_io_threads = threads
             for t in threads:
                 t.start()
             self._rows_epoch = 0
@@ -624,6 +640,9 @@ class GpuBatchReader(object):
         self._stopped = True
 
     def join(self):
+        # IO threads must be OUT of native code before interpreter exit
+        for t in getattr(self, '_live_io_threads', []):
+            t.join(timeout=10.0)
         if self._cache is not None:
             self._cache.cleanup()
 
