@@ -231,8 +231,7 @@ class GpuRowGroupDecoder(object):
             # sees an ordinary UNCOMPRESSED chunk view
             eff_comp = ch['compression']
             eff_buf = host_buf
-            if ch['compression'] == 'ZSTD' and \
-                    not (pages['page_type'].numpy() == _PAGE_DATA_V2).any():
+            if ch['compression'] == 'ZSTD':
                 entry['zstd'] = self._zstd_decompress_pages(host_buf, pages)
                 pages = entry['zstd']['pages']
                 eff_buf = entry['zstd']['host_buf']
@@ -280,21 +279,50 @@ class GpuRowGroupDecoder(object):
         offs = pages['data_off'].numpy().astype(np.int64)
         csz = pages['comp_size'].numpy().astype(np.int64)
         usz = pages['uncomp_size'].numpy().astype(np.int64)
+        ptype = pages['page_type'].numpy()
+        dlb = pages['dl_bytes'].numpy().astype(np.int64)
+        rlb = pages['rl_bytes'].numpy().astype(np.int64)
+        v2c = pages['v2_is_compressed'].numpy().astype(bool)
         u_off = np.zeros(len(offs), dtype=np.int64)
         if len(offs) > 1:
             u_off[1:] = np.cumsum(usz)[:-1]
         hbuf = torch.empty(int(usz.sum()) + _SLACK, dtype=torch.uint8,
                            pin_memory=self._pin_memory)
-        status = torch.zeros(len(offs), dtype=torch.int32)
-        self._ext.zstd_decompress_host(
-            host_buf, torch.from_numpy(offs), torch.from_numpy(csz),
-            hbuf, torch.from_numpy(u_off), torch.from_numpy(usz), status)
-        if int(status.abs().sum()):
-            raise RuntimeError('zstd decode error: status={}'
-                               .format(status.tolist()))
+        hb_src = host_buf.numpy()
+        hb_dst = hbuf.numpy()
+        f_src, f_slen, f_dst, f_dlen = [], [], [], []
+        for i in range(len(offs)):
+            # V2 pages store def/rep levels UNCOMPRESSED ahead of the
+            # (optionally) compressed values section; copy the prefix and
+            # decompress only the values.  V1/dict pages are whole-page
+            # frames.
+            pre = int(dlb[i] + rlb[i]) if ptype[i] == _PAGE_DATA_V2 else 0
+            if pre:
+                hb_dst[u_off[i]:u_off[i] + pre] = \
+                    hb_src[offs[i]:offs[i] + pre]
+            if ptype[i] == _PAGE_DATA_V2 and not v2c[i]:
+                hb_dst[u_off[i] + pre:u_off[i] + usz[i]] = \
+                    hb_src[offs[i] + pre:offs[i] + csz[i]]
+                continue
+            f_src.append(offs[i] + pre)
+            f_slen.append(csz[i] - pre)
+            f_dst.append(u_off[i] + pre)
+            f_dlen.append(usz[i] - pre)
+        if f_src:
+            status = torch.zeros(len(f_src), dtype=torch.int32)
+            self._ext.zstd_decompress_host(
+                host_buf, torch.tensor(f_src, dtype=torch.int64),
+                torch.tensor(f_slen, dtype=torch.int64),
+                hbuf, torch.tensor(f_dst, dtype=torch.int64),
+                torch.tensor(f_dlen, dtype=torch.int64), status)
+            if int(status.abs().sum()):
+                raise RuntimeError('zstd decode error: status={}'
+                                   .format(status.tolist()))
         pages2 = dict(pages)
         pages2['data_off'] = torch.from_numpy(u_off)
         pages2['comp_size'] = torch.from_numpy(usz)
+        pages2['v2_is_compressed'] = torch.zeros_like(
+            pages['v2_is_compressed'])
         return {'host_buf': hbuf, 'pages': pages2}
 
     @staticmethod
