@@ -83,20 +83,35 @@ struct BitReader {
   int64_t pos, end;
   uint64_t buf;   // left-aligned: next bit is bit 63
   int cnt;
+  uint32_t stash;       // prefetched NEXT 4 bytes (at stash_pos)
+  int64_t stash_pos;    // -1: invalid
 
   __device__ void init(const uint8_t* data, int64_t lo, int64_t hi) {
-    p = data; pos = lo; end = hi; buf = 0; cnt = 0;
+    p = data; pos = lo; end = hi; buf = 0; cnt = 0; stash_pos = -1;
   }
   // JPEG entropy data marks every 0xFF with a stuffed 0x00.  The common
-  // case (no 0xFF in the next 4 bytes, ~98%) refills 32 bits with ONE
-  // global load instead of four dependent ~400-cycle byte loads — the
-  // refill was 70% of the decode kernel's time before this.
+  // case (no 0xFF in the next 4 bytes, ~98%) refills 32 bits from a word
+  // load; the kernel is L2-LATENCY bound (PMC: 61% wave cycles waiting at
+  // 92% L2 hit), so each refill trip also PREFETCHES the following word —
+  // the two loads issue together and every other refill costs no memory
+  // trip at all.
   __device__ __forceinline__ void fill() {
     while (cnt <= 32) {
       if (pos + 4 <= end) {
-        uint32_t w = (uint32_t)p[pos] | ((uint32_t)p[pos + 1] << 8) |
-                     ((uint32_t)p[pos + 2] << 16) |
-                     ((uint32_t)p[pos + 3] << 24);
+        uint32_t w;
+        if (stash_pos == pos) {
+          w = stash;
+          stash_pos = -1;
+        } else {
+          w = (uint32_t)p[pos] | ((uint32_t)p[pos + 1] << 8) |
+              ((uint32_t)p[pos + 2] << 16) | ((uint32_t)p[pos + 3] << 24);
+          if (pos + 8 <= end) {  // issues alongside w's loads (ILP)
+            stash = (uint32_t)p[pos + 4] | ((uint32_t)p[pos + 5] << 8) |
+                    ((uint32_t)p[pos + 6] << 16) |
+                    ((uint32_t)p[pos + 7] << 24);
+            stash_pos = pos + 4;
+          }
+        }
         // detect any 0xFF byte: a byte of ~w is zero iff the byte is 0xFF
         uint32_t inv = ~w;
         if (!((inv - 0x01010101u) & ~inv & 0x80808080u)) {
