@@ -65,7 +65,7 @@ def train(url, epochs=2, batch_size=64):
                 opt.zero_grad()
                 loss.backward()
                 opt.step()
-                total += float(loss)
+                total += loss.item()
                 n += 1
             print('epoch {} mean loss {:.4f}'.format(epoch, total / n))
 
