@@ -230,3 +230,26 @@ def test_dryrun_zstd_scalar(stub_decoder, tmp_path):
                           compression='zstd')
     out = _decode_all(dec, url, ['id', 'f0'])
     assert not dec.cpu_assist_columns
+
+
+def test_zstd_corrupt_page_raises(stub_decoder, tmp_path):
+    """Bit-flipped ZSTD page payload must raise loudly (status contract),
+    not return garbage."""
+    import pyarrow.parquet as pq
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    dec, stub = stub_decoder
+    url = 'file://' + str(tmp_path / 'zc')
+    create_scalar_dataset(url, num_rows=2000, rowgroup_size=1000,
+                          compression='zstd')
+    path = [str(p) for p in (tmp_path / 'zc').iterdir()
+            if p.suffix == '.parquet'][0]
+    pf = pq.ParquetFile(path)
+    col = pf.metadata.row_group(0).column(0)  # 'id': compressible int64
+    # +200: clear of the thrift header (statistics make it ~100 bytes)
+    off = col.data_page_offset + 200
+    raw = bytearray(open(path, 'rb').read())
+    for i in range(24):
+        raw[off + i] ^= 0xA5
+    open(path, 'wb').write(bytes(raw))
+    with pytest.raises(RuntimeError, match='zstd'):
+        _decode_all(dec, url, [col.path_in_schema])

@@ -937,3 +937,28 @@ def test_zstd_device_kernel_matches_host(ext, tmp_path):
     out = dst.cpu().numpy().tobytes()
     for d, o, n in zip(datas, doff, dlen):
         assert out[o:o + n] == d
+
+
+def test_corrupt_lz4_page_raises(ext, tmp_path):
+    """Corrupt LZ4 page -> kernel status propagates as a loud RuntimeError
+    at the reader's status check."""
+    import pyarrow.parquet as pq
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    from petastorm_amd import make_batch_reader
+    url = 'file://' + str(tmp_path / 'lc')
+    create_scalar_dataset(url, num_rows=2000, rowgroup_size=1000,
+                          compression='lz4')
+    path = [str(p) for p in (tmp_path / 'lc').iterdir()
+            if p.suffix == '.parquet'][0]
+    pf = pq.ParquetFile(path)
+    col = pf.metadata.row_group(0).column(2)
+    raw = bytearray(open(path, 'rb').read())
+    off = col.data_page_offset + 40
+    for i in range(16):
+        raw[off + i] ^= 0x5A
+    open(path, 'wb').write(bytes(raw))
+    with pytest.raises(RuntimeError):
+        with make_batch_reader(url, device='cuda',
+                               shuffle_row_groups=False) as r:
+            for _ in r:
+                pass
