@@ -70,3 +70,14 @@ def test_in_pseudorandom_split_partitions_and_determinism():
     # fractions are approximately honored
     frac0 = assigned.count(0) / len(assigned)
     assert 0.4 < frac0 < 0.6
+
+
+def test_pseudorandom_split_vectorized_matches_scalar():
+    import numpy as np
+    from petastorm_amd.predicates import in_pseudorandom_split
+    p = in_pseudorandom_split([0.6, 0.4], 0, 'id')
+    ids = np.arange(500)
+    vec = p.do_include_vectorized({'id': ids})
+    scalar = np.array([p.do_include({'id': i}) for i in ids])
+    np.testing.assert_array_equal(vec, scalar)
+    assert 200 < vec.sum() < 400  # roughly the 60% fraction
