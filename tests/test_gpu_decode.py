@@ -711,7 +711,7 @@ def test_gzip_rowgroup_decode(ext, tmp_path):
     np.testing.assert_array_equal(np.sort(ids), np.arange(3000))
 
 
-@pytest.mark.parametrize('compression', ['snappy', 'none', 'gzip', 'zstd'])
+@pytest.mark.parametrize('compression', ['snappy', 'none', 'gzip', 'zstd', 'lz4'])
 def test_datapage_v2_decode(ext, tmp_path, compression):
     import pyarrow as pa
     import pyarrow.parquet as pq
