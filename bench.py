@@ -322,6 +322,9 @@ def main():
     ap.add_argument('--rows', type=int, default=None)
     args = ap.parse_args()
 
+    si = os.environ.get('PSA_SWITCH_INTERVAL')
+    if si:
+        sys.setswitchinterval(float(si))
     rank, world, local = _dist_env()
     dist = _init_dist(world)
     if torch.cuda.is_available():
