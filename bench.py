@@ -132,7 +132,10 @@ def bench_imagenet(args, rank, world, device, dist):
         print('WARNING: cpu-assist columns: {}'.format(
             reader.diagnostics['cpu_assist_columns']), file=sys.stderr)
     if os.environ.get('PSA_TIMING') == '1' and rank == 0:
-        print('stage_times:', reader.diagnostics.get('stage_times'),
+        diag = reader.diagnostics
+        print('stage_times:', diag.get('stage_times'),
+              'staging_allocs:', diag.get('staging_allocs'),
+              'staging_copy_s:', diag.get('staging_copy_s'),
               file=sys.stderr)
     return result, {
         'model': 'ImageNetSchema(224x224x3 jpeg CompressedImageCodec + '

@@ -85,6 +85,8 @@ class GpuRowGroupDecoder(object):
         # ~1ms; recycling makes _up() allocation-free in steady state)
         self._staging_free = {}
         self._staging_inuse = []
+        self.staging_allocs = 0
+        self.staging_copy_s = 0.0
         # pinned scalar verdicts for take_pending()'s dispatch-time status
         # reduction (recycled)
         self._host_scalar_free = []
@@ -136,10 +138,17 @@ class GpuRowGroupDecoder(object):
         if self._pin_memory and not t.is_pinned():
             key = (t.dtype, t.numel())
             free = self._staging_free.get(key)
-            pinned = free.pop() if free else torch.empty(
-                t.shape, dtype=t.dtype, pin_memory=True)
+            if free:
+                pinned = free.pop()
+            else:
+                pinned = torch.empty(t.shape, dtype=t.dtype,
+                                     pin_memory=True)
+                self.staging_allocs += 1
             pinned = pinned.view(t.shape)
+            import time as _time
+            _t0 = _time.perf_counter()
             pinned.copy_(t)
+            self.staging_copy_s += _time.perf_counter() - _t0
             self._staging_inuse.append((key, pinned))
             t = pinned
         return t.to(self.device, non_blocking=True)
