@@ -136,6 +136,7 @@ def bench_imagenet(args, rank, world, device, dist):
         print('stage_times:', diag.get('stage_times'),
               'staging_allocs:', diag.get('staging_allocs'),
               'staging_copy_s:', diag.get('staging_copy_s'),
+              'by_key:', diag.get('staging_copy_by_key'),
               file=sys.stderr)
     return result, {
         'model': 'ImageNetSchema(224x224x3 jpeg CompressedImageCodec + '

@@ -87,6 +87,7 @@ class GpuRowGroupDecoder(object):
         self._staging_inuse = []
         self.staging_allocs = 0
         self.staging_copy_s = 0.0
+        self.staging_copy_by_key = {}
         # pinned scalar verdicts for take_pending()'s dispatch-time status
         # reduction (recycled)
         self._host_scalar_free = []
@@ -148,7 +149,11 @@ class GpuRowGroupDecoder(object):
             import time as _time
             _t0 = _time.perf_counter()
             pinned.copy_(t)
-            self.staging_copy_s += _time.perf_counter() - _t0
+            _dt = _time.perf_counter() - _t0
+            self.staging_copy_s += _dt
+            k2 = (str(t.dtype), t.numel())
+            c, s = self.staging_copy_by_key.get(k2, (0, 0.0))
+            self.staging_copy_by_key[k2] = (c + 1, s + _dt)
             self._staging_inuse.append((key, pinned))
             t = pinned
         return t.to(self.device, non_blocking=True)

@@ -651,7 +651,10 @@ class GpuBatchReader(object):
         d = {'cpu_assist_columns': sorted(self._decoder.cpu_assist_columns),
              'stage_times': dict(self.stage_times),
              'staging_allocs': self._decoder.staging_allocs,
-             'staging_copy_s': round(self._decoder.staging_copy_s, 4)}
+             'staging_copy_s': round(self._decoder.staging_copy_s, 4),
+             'staging_copy_by_key': sorted(
+                 self._decoder.staging_copy_by_key.items(),
+                 key=lambda kv: -kv[1][1])[:5]}
         if self._cache is not None:
             d.update(hbm_cache_hits=self._cache.hits,
                      hbm_cache_misses=self._cache.misses,
