@@ -126,6 +126,12 @@ def bench_imagenet(args, rank, world, device, dist):
         return b['image'].shape[0]
 
     result = _run_timed(args, step, device, dist, world)
+    if dist is not None and hasattr(reader, 'epoch_stats'):
+        # RCCL epoch all-gather over xGMI (BASELINE config 4): every rank
+        # reports its consumed rows at this synchronized point
+        stats = reader.epoch_stats()
+        if rank == 0:
+            print('epoch_stats (rows/rank):', stats, file=sys.stderr)
     reader.stop()
     reader.join()
     if rank == 0 and reader.diagnostics.get('cpu_assist_columns'):
