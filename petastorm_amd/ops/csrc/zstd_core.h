@@ -123,9 +123,6 @@ struct BackBits {
     container = load8(ptr);
     return status;
   }
-  PSA_HD int at_end() {  // all payload bits consumed exactly
-    return (smallStream || ptr == start) && bitsConsumed == 64;
-  }
 };
 
 // ---------------------------------------------------------------------------

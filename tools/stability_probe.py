@@ -53,9 +53,10 @@ def run_config(name, seconds):
         win_t0 = time.time()
         win_rows = 0
         while time.time() < t_end:
-            rows += step_fn()
-            win_rows += step_fn()
-            steps += 2
+            r = step_fn()
+            rows += r
+            win_rows += r
+            steps += 1
             if time.time() - win_t0 > 5.0:
                 windows.append((win_rows / (time.time() - win_t0),
                                 _rss_mb(), _hbm_mb()))
