@@ -83,11 +83,12 @@ struct BitReader {
   int64_t pos, end;
   uint64_t buf;   // left-aligned: next bit is bit 63
   int cnt;
-  uint32_t stash;       // prefetched NEXT 4 bytes (at stash_pos)
-  int64_t stash_pos;    // -1: invalid
+  uint32_t sw0, sw1;    // prefetched words (little-endian composed)
+  int64_t spos;         // byte position of sw0
+  int scount;           // prefetched words available (0..2)
 
   __device__ void init(const uint8_t* data, int64_t lo, int64_t hi) {
-    p = data; pos = lo; end = hi; buf = 0; cnt = 0; stash_pos = -1;
+    p = data; pos = lo; end = hi; buf = 0; cnt = 0; scount = 0; spos = 0;
   }
   // JPEG entropy data marks every 0xFF with a stuffed 0x00.  The common
   // case (no 0xFF in the next 4 bytes, ~98%) refills 32 bits from a word
@@ -99,17 +100,30 @@ struct BitReader {
     while (cnt <= 32) {
       if (pos + 4 <= end) {
         uint32_t w;
-        if (stash_pos == pos) {
-          w = stash;
-          stash_pos = -1;
+        if (scount > 0 && spos == pos) {
+          w = sw0;
+          sw0 = sw1;
+          spos += 4;
+          --scount;
         } else {
+          scount = 0;
           w = (uint32_t)p[pos] | ((uint32_t)p[pos + 1] << 8) |
               ((uint32_t)p[pos + 2] << 16) | ((uint32_t)p[pos + 3] << 24);
-          if (pos + 8 <= end) {  // issues alongside w's loads (ILP)
-            stash = (uint32_t)p[pos + 4] | ((uint32_t)p[pos + 5] << 8) |
-                    ((uint32_t)p[pos + 6] << 16) |
-                    ((uint32_t)p[pos + 7] << 24);
-            stash_pos = pos + 4;
+          if (pos + 12 <= end) {  // all 12 byte loads issue together (ILP)
+            sw0 = (uint32_t)p[pos + 4] | ((uint32_t)p[pos + 5] << 8) |
+                  ((uint32_t)p[pos + 6] << 16) |
+                  ((uint32_t)p[pos + 7] << 24);
+            sw1 = (uint32_t)p[pos + 8] | ((uint32_t)p[pos + 9] << 8) |
+                  ((uint32_t)p[pos + 10] << 16) |
+                  ((uint32_t)p[pos + 11] << 24);
+            spos = pos + 4;
+            scount = 2;
+          } else if (pos + 8 <= end) {
+            sw0 = (uint32_t)p[pos + 4] | ((uint32_t)p[pos + 5] << 8) |
+                  ((uint32_t)p[pos + 6] << 16) |
+                  ((uint32_t)p[pos + 7] << 24);
+            spos = pos + 4;
+            scount = 1;
           }
         }
         // detect any 0xFF byte: a byte of ~w is zero iff the byte is 0xFF
