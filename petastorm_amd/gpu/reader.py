@@ -323,10 +323,10 @@ class GpuBatchReader(object):
             self._rows_epoch = 0
             # software pipeline: decode of the next row-groups is LAUNCHED
             # before earlier batches are yielded, keeping `pipeline_depth`
-            # row-groups of H2D + kernels in flight on the stream (fills the
-            # chip when a single row-group's decode launches few waves);
-            # each row-group's status sync happens just before ITS batch is
-            # yielded
+            # row-groups of H2D + kernels in flight across the stream pool
+            # (fills the chip when a single row-group's decode launches few
+            # waves); each row-group's pinned status verdict was copied at
+            # dispatch, so emit only waits that row-group's event
             from collections import deque
             pending = deque()
             dispatched = self._piece_pos  # pieces fully processed so far
@@ -372,7 +372,14 @@ class GpuBatchReader(object):
             ended = 0
             while ended < n_io:
                 t0 = time.perf_counter()
-                kind, piece, host, meta = qs[rr].get()
+                while True:  # stop-aware get (mirrors _q_put)
+                    if self._stopped:
+                        return
+                    try:
+                        kind, piece, host, meta = qs[rr].get(timeout=0.2)
+                        break
+                    except queue.Empty:
+                        continue
                 self.stage_times['io_wait'] += time.perf_counter() - t0
                 if kind == 'end':
                     ended += 1
