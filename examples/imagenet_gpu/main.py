@@ -29,6 +29,13 @@ def main(url=None, rows=512, batch_size=64, steps=20):
     if url is None:
         url = 'file://' + tempfile.mkdtemp(prefix='imnet_')
         create_imagenet_dataset(url, num_rows=rows, rowgroup_size_mb=16)
+    elif not url.startswith('file://'):
+        import os
+        if not os.path.isdir(url) or not os.listdir(url):
+            os.makedirs(url, exist_ok=True)
+            create_imagenet_dataset('file://' + url, num_rows=rows,
+                                    rowgroup_size_mb=16)
+        url = 'file://' + url
 
     ext = ops.ext()
     mean = torch.tensor([0.485, 0.456, 0.406], device='cuda')
