@@ -33,10 +33,20 @@ class _StubExt(object):
         self._real = ops.ext()
         self.calls = []
 
+    # output-tensor positions per kernel whose results feed HOST-side
+    # indexing later (a no-op stub would leave uninitialized memory there,
+    # making the dry run flaky on heap reuse)
+    _OUTPUTS = {'rle_hybrid_decode_batch': (6,),
+                'plain_fixed_decode_batch': (11,)}
+
     def __getattr__(self, name):
         if name in self._KERNELS:
+            outs = self._OUTPUTS.get(name, ())
             def stub(*args, **kwargs):
                 self.calls.append(name)
+                for i in outs:
+                    if i < len(args) and isinstance(args[i], torch.Tensor):
+                        args[i].zero_()
             return stub
         return getattr(self._real, name)
 
