@@ -238,7 +238,7 @@ class GpuBatchReader(object):
         Ordering stays deterministic because the consumer reads the
         per-thread queues in the same round-robin order.
         """
-        columns = list(self._view_schema.fields.keys())
+        all_columns = list(self._view_schema.fields.keys())
         try:
             for piece in pieces:
                 if self._stopped:
@@ -247,11 +247,21 @@ class GpuBatchReader(object):
                         self._cache_key(piece) in self._cache._store:
                     self._q_put(out_q, ('cached', piece, None, None))
                     continue
+                # hive-partition fields live in the PATH, not the file:
+                # keep them out of the physical read and materialize them
+                # in _decode_piece (reference partitions= in piece.read,
+                # arrow_reader_worker.py:358)
+                parts = piece.partitions or {}
+                columns = [c for c in all_columns if c not in parts]
                 md, pschema = self._metadata(piece.path)
                 t0 = time.perf_counter()
-                host, meta = self._decoder.read_rowgroup_bytes(
-                    piece.path, md, pschema, piece.row_group, columns,
-                    self._pin_pool)
+                if columns:
+                    host, meta = self._decoder.read_rowgroup_bytes(
+                        piece.path, md, pschema, piece.row_group, columns,
+                        self._pin_pool)
+                else:  # every requested field is a partition key
+                    host = torch.empty(0, dtype=torch.uint8)
+                    meta = {'num_rows': piece.num_rows, 'chunks': []}
                 t1 = time.perf_counter()
                 # host-only parse work (page walk, offset scans, image
                 # headers) runs HERE so it overlaps GPU decode of other
@@ -475,6 +485,22 @@ class GpuBatchReader(object):
                 columns[name] = col
         if assist:
             columns.update(self._cpu_assist(piece, assist))
+        # materialize requested hive-partition fields as constant columns
+        # (CPU path: workers/batch_worker.py partition handling)
+        parts = piece.partitions or {}
+        n_rows = meta['num_rows'] if isinstance(meta, dict) else None
+        for name in self._view_schema.fields:
+            if name in columns or name not in parts:
+                continue
+            field = self._storage_schema.fields.get(name)
+            raw = parts[name]
+            try:
+                val = np.dtype(field.numpy_dtype).type(raw) \
+                    if field is not None else raw
+                columns[name] = self._decoder._up(
+                    np.full(n_rows, val))
+            except (TypeError, ValueError):
+                columns[name] = np.full(n_rows, str(raw), dtype=object)
         # NB: flush_status is called by the pipeline loop just before this
         # row-group's batch is yielded, so the sync overlaps decode of the
         # next row-group

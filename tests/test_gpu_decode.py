@@ -962,3 +962,35 @@ def test_corrupt_lz4_page_raises(ext, tmp_path):
                                shuffle_row_groups=False) as r:
             for _ in r:
                 pass
+
+
+def test_gpu_reader_hive_partitions(ext, tmp_path):
+    """Hive-partitioned dataset on the GPU path: partition columns are
+    excluded from the physical read and materialized as device constants
+    (string keys stay host-side object arrays)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    root = tmp_path / 'pds'
+    idx = 0
+    for color, code in (('red', 0), ('green', 1), ('blue', 2)):
+        d = root / 'color={}'.format(color) / 'code={}'.format(code)
+        d.mkdir(parents=True)
+        pq.write_table(
+            pa.table({'id': np.arange(idx, idx + 50, dtype=np.int64)}),
+            str(d / 'p.parquet'), use_dictionary=False, row_group_size=25)
+        idx += 50
+    url = 'file://' + str(root)
+    seen = {}
+    with make_batch_reader(url, device='cuda',
+                           shuffle_row_groups=False) as r:
+        assert 'color' in r.schema.fields and 'code' in r.schema.fields
+        for b in r:
+            assert b.id.is_cuda
+            assert b.code.is_cuda           # int partition -> device tensor
+            codes = set(b.code.cpu().tolist())
+            assert len(codes) == 1
+            colors = set(np.asarray(b.color).tolist())
+            assert len(colors) == 1
+            seen[colors.pop()] = codes.pop()
+    assert seen == {'red': 0, 'green': 1, 'blue': 2}
