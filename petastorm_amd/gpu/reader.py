@@ -107,6 +107,7 @@ class _PinnedPool(object):
 class GpuBatchReader(object):
     def __init__(self, fs, path_or_paths, schema_fields=None,
                  shuffle_row_groups=True, shuffle_rows=False, predicate=None,
+                 rowgroup_selector=None,
                  num_epochs=1, cur_shard=None, shard_count=None, seed=None,
                  transform_spec=None, filters=None, device='cuda',
                  cache_type=None, cache_size_limit=None, pipeline_depth=3,
@@ -149,6 +150,19 @@ class GpuBatchReader(object):
         if filters:
             self._pieces = dsm.select_pieces_by_filters(fs, self._pieces,
                                                         filters)
+        if rowgroup_selector is not None:
+            # rowgroup-index selectors (reference reader.py:599-618)
+            from petastorm_amd.etl.rowgroup_indexing import \
+                load_rowgroup_indexes
+            index_dict = load_rowgroup_indexes(fs, path_or_paths)
+            missing = [n for n in rowgroup_selector.select_index_names()
+                       if n not in index_dict]
+            if missing:
+                raise ValueError(
+                    'Indexes {} are not available in the dataset '
+                    '(available: {})'.format(missing, sorted(index_dict)))
+            chosen = rowgroup_selector.select_row_groups(index_dict)
+            self._pieces = [p for p in self._pieces if p.index in chosen]
         if not self._pieces:
             raise NoDataAvailableError('Dataset has no row groups')
         cur_shard, shard_count = epoch_sync.shard_for_rank(cur_shard,

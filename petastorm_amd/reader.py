@@ -157,16 +157,32 @@ def make_batch_reader(dataset_url_or_urls,
         dataset_url_or_urls, storage_options)
     if device is not None and str(device).startswith('cuda'):
         from petastorm_amd.gpu.reader import GpuBatchReader
+        # loud rejection of options the GPU pipeline does not implement —
+        # never silently change semantics the caller asked for
+        if shuffle_row_drop_partitions != 1:
+            raise NotImplementedError(
+                'shuffle_row_drop_partitions is a CPU-worker memory trick '
+                '(reference py_dict_reader_worker.py:264-286); the GPU '
+                'pipeline decodes whole row-groups in HBM')
+        gpu_kwargs = dict(gpu_options or {})
+        if cache_type not in (None, 'null'):
+            if cache_type != 'hbm':
+                raise NotImplementedError(
+                    "cache_type={!r} on the GPU path; use 'hbm' (decoded "
+                    'row-groups cached in HBM3E)'.format(cache_type))
+            gpu_kwargs.setdefault('cache_type', 'hbm')
+            gpu_kwargs.setdefault('cache_size_limit', cache_size_limit)
         return GpuBatchReader(fs, path_or_paths,
                               schema_fields=schema_fields,
                               shuffle_row_groups=shuffle_row_groups,
                               shuffle_rows=shuffle_rows,
                               predicate=predicate,
+                              rowgroup_selector=rowgroup_selector,
                               num_epochs=num_epochs,
                               cur_shard=cur_shard, shard_count=shard_count,
                               seed=seed, transform_spec=transform_spec,
                               filters=filters,
-                              device=device, **(gpu_options or {}))
+                              device=device, **gpu_kwargs)
     cache = _make_cache(cache_type, cache_location, cache_size_limit,
                         cache_row_size_estimate, cache_extra_settings)
     pool = _make_pool(reader_pool_type, workers_count, results_queue_size)

@@ -69,3 +69,25 @@ def test_missing_index_raises(indexed_dataset):
     with pytest.raises(ValueError):
         make_reader(indexed_dataset['url'], reader_pool_type='dummy',
                     rowgroup_selector=sel)
+
+
+def test_gpu_reader_accepts_rowgroup_selector(indexed_dataset):
+    """GpuBatchReader applies index selectors at planning time (CPU-only
+    check: the constructor's piece filtering needs no GPU)."""
+    from petastorm_amd.gpu.reader import GpuBatchReader
+    fs, path = get_filesystem_and_path_or_paths(indexed_dataset['url'])
+    all_r = GpuBatchReader(fs, path, shuffle_row_groups=False)
+    sel_r = GpuBatchReader(fs, path, shuffle_row_groups=False,
+                           rowgroup_selector=SingleIndexSelector(
+                               'id2_index', [1]))
+    assert 0 < len(sel_r._pieces) < len(all_r._pieces)
+
+
+def test_gpu_route_rejects_unsupported_options(indexed_dataset):
+    from petastorm_amd import make_batch_reader
+    with pytest.raises(NotImplementedError, match='shuffle_row_drop'):
+        make_batch_reader(indexed_dataset['url'], device='cuda',
+                          shuffle_row_drop_partitions=2)
+    with pytest.raises(NotImplementedError, match='cache_type'):
+        make_batch_reader(indexed_dataset['url'], device='cuda',
+                          cache_type='local-disk')
