@@ -164,6 +164,10 @@ def make_batch_reader(dataset_url_or_urls,
                 'shuffle_row_drop_partitions is a CPU-worker memory trick '
                 '(reference py_dict_reader_worker.py:264-286); the GPU '
                 'pipeline decodes whole row-groups in HBM')
+        if not decode_codecs:
+            raise NotImplementedError(
+                'decode_codecs=False (raw encoded values) is a CPU-path '
+                'option; the GPU pipeline always decodes')
         gpu_kwargs = dict(gpu_options or {})
         if cache_type not in (None, 'null'):
             if cache_type != 'hbm':
