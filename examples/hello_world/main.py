@@ -63,8 +63,12 @@ def pytorch_hello_world(url):
 
 
 if __name__ == '__main__':
-    url = sys.argv[1] if len(sys.argv) > 1 \
-        else 'file://' + tempfile.mkdtemp(prefix='hello_world_')
+    import argparse
+    parser = argparse.ArgumentParser(description=__doc__)
+    parser.add_argument('url', nargs='?',
+                        default='file://' + tempfile.mkdtemp(prefix='hello_world_'),
+                        help='dataset output URL (default: fresh temp dir)')
+    url = parser.parse_args().url
     generate_dataset(url)
     python_hello_world(url)
     pytorch_hello_world(url)

@@ -79,4 +79,8 @@ def main(url=None, rows=512, batch_size=64, steps=20):
 
 
 if __name__ == '__main__':
-    main(sys.argv[1] if len(sys.argv) > 1 else None)
+    import argparse
+    parser = argparse.ArgumentParser(description=__doc__)
+    parser.add_argument('url', nargs='?', default=None,
+                        help='dataset URL (default: generate a fresh one)')
+    main(parser.parse_args().url)

@@ -239,7 +239,8 @@ def load_row_groups(fs, path_or_paths):
             if fpath.startswith(r.rstrip('/') + '/'):
                 partitions = parse_partition_values(r.rstrip('/'), fpath)
                 break
-        md = pq.ParquetFile(fs.open(fpath, 'rb')).metadata
+        with fs.open(fpath, 'rb') as f:
+            md = pq.ParquetFile(f).metadata
         for rg in range(md.num_row_groups):
             pieces.append(RowGroupPiece(index, fpath, rg,
                                         md.row_group(rg).num_rows,
@@ -264,7 +265,8 @@ def get_schema(fs, path_or_paths):
     import pyarrow.parquet as pq
     files = list_parquet_files(fs, path_or_paths)
     if files:
-        md = pq.ParquetFile(fs.open(files[0], 'rb')).schema_arrow.metadata
+        with fs.open(files[0], 'rb') as f:
+            md = pq.ParquetFile(f).schema_arrow.metadata
         if md and UNISCHEMA_KEY.encode() in md:
             return Unischema.from_json(md[UNISCHEMA_KEY.encode()].decode())
     raise ValueError('Dataset at {} has no petastorm_amd schema metadata. Use '
@@ -311,7 +313,8 @@ def infer_or_load_unischema(fs, path_or_paths):
         files = list_parquet_files(fs, path_or_paths)
         if not files:
             raise ValueError('No parquet files found at {}'.format(path_or_paths))
-        arrow_schema = pq.ParquetFile(fs.open(files[0], 'rb')).schema_arrow
+        with fs.open(files[0], 'rb') as f:
+            arrow_schema = pq.ParquetFile(f).schema_arrow
         return _with_partition_fields(
             Unischema.from_arrow_schema(arrow_schema)), False
 
@@ -343,7 +346,8 @@ def select_pieces_by_filters(fs, pieces, filters):
             return v, v
         key = piece.path
         if key not in stats_cache:
-            stats_cache[key] = pq.ParquetFile(fs.open(key, 'rb')).metadata
+            with fs.open(key, 'rb') as f:
+                stats_cache[key] = pq.ParquetFile(f).metadata
         md = stats_cache[key].row_group(piece.row_group)
         for ci in range(md.num_columns):
             col = md.column(ci)

@@ -150,8 +150,9 @@ def build_rowgroup_index(dataset_url, indexers):
     import pyarrow.parquet as pq
     needed = sorted({c for ix in indexers for c in ix.column_names})
     for piece in pieces:
-        pf = pq.ParquetFile(fs.open(piece.path, 'rb'))
-        table = pf.read_row_group(piece.row_group, columns=needed)
+        with fs.open(piece.path, 'rb') as f:
+            table = pq.ParquetFile(f).read_row_group(piece.row_group,
+                                                     columns=needed)
         pydict = table.to_pydict()
         rows = [dict(zip(pydict.keys(), vals))
                 for vals in zip(*pydict.values())]

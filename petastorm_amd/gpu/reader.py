@@ -218,11 +218,20 @@ class GpuBatchReader(object):
         self._gen = self._generate()
 
     # ------------------------------------------------------------------
+    _FILE_MD_CACHE_CAP = 4096
+
     def _metadata(self, path):
+        # Parsed footer metadata only — the file handle is closed here, and
+        # the cache is bounded so huge multi-file datasets cannot grow it
+        # without limit (the FileMetaData objects carry no fd).
         if path not in self._file_md:
             import pyarrow.parquet as pq
-            pf = pq.ParquetFile(self._fs.open(path, 'rb'))
-            self._file_md[path] = (pf.metadata, pf.schema)
+            with self._fs.open(path, 'rb') as f:
+                pf = pq.ParquetFile(f)
+                entry = (pf.metadata, pf.schema)
+            if len(self._file_md) >= self._FILE_MD_CACHE_CAP:
+                self._file_md.pop(next(iter(self._file_md)))
+            self._file_md[path] = entry
         return self._file_md[path]
 
     def _epoch_pieces(self, epoch):
@@ -525,10 +534,10 @@ class GpuBatchReader(object):
         """CPU decode + upload for columns outside the GPU fast path."""
         from petastorm_amd.workers.batch_worker import \
             arrow_table_to_numpy_dict
-        md, _ = self._metadata(piece.path)
         import pyarrow.parquet as pq
-        pf = pq.ParquetFile(self._fs.open(piece.path, 'rb'))
-        table = pf.read_row_group(piece.row_group, columns=names)
+        with self._fs.open(piece.path, 'rb') as f:
+            table = pq.ParquetFile(f).read_row_group(
+                piece.row_group, columns=names)
         np_dict = arrow_table_to_numpy_dict(table, self._storage_schema, True)
         out = {}
         for k, v in np_dict.items():

@@ -42,8 +42,10 @@ struct Cursor {
     return (int64_t)(u >> 1) ^ -(int64_t)(u & 1);
   }
   void skip_bytes(int64_t n) {
+    // A corrupt uvarint length >= 2^63 casts to negative; a backwards move
+    // would never trip the pos > end check and could re-parse forever.
+    if (n < 0 || n > (int64_t)(end - pos)) { ok = false; return; }
     pos += n;
-    if (pos > end) ok = false;
   }
 };
 
@@ -206,6 +208,9 @@ py::dict parquet_walk_pages(torch::Tensor buf, torch::Tensor chunk_off,
       PageHeader h;
       bool good = parse_page_header(c, h);
       TORCH_CHECK(good, "malformed page header in chunk ", ci, " at ", c.pos);
+      TORCH_CHECK(h.type >= 0,
+                  "page header without a page type in chunk ", ci, " at ",
+                  c.pos);
       TORCH_CHECK(h.compressed >= 0 && c.pos + h.compressed <= c.end,
                   "page data overruns chunk ", ci);
       p_chunk.push_back(ci);

@@ -14,6 +14,7 @@ worker can evaluate a whole row-group and stream-compact matching rows
 """
 
 import hashlib
+import sys
 
 import numpy as np
 
@@ -147,9 +148,16 @@ class in_pseudorandom_split(PredicateBase):
     ``fraction_list`` partitions [0,1); a row belongs to partition
     ``predicate_index`` when the md5 hash of its id-field value falls into
     that fraction of the hash space.
+
+    SPLIT CONTRACT: bucketing is the reference's exact rule —
+    ``int(md5(str(value)).hexdigest(), 16) % sys.maxsize`` tested against
+    ``fraction * (sys.maxsize - 1)`` boundaries
+    (reference predicates.py:39-41,172-182) — so split membership of any
+    given id is identical to upstream petastorm, and existing train/val/
+    test assignments survive a migration to this framework.
     """
 
-    _MAX_HASH = float(0xFFFFFFFF)
+    _MODULUS = sys.maxsize  # 2**63 - 1, matching the reference contract
 
     def __init__(self, fraction_list, predicate_index, predicate_field):
         if predicate_index < 0 or predicate_index >= len(fraction_list):
@@ -159,18 +167,23 @@ class in_pseudorandom_split(PredicateBase):
         self._predicate_field = predicate_field
         lo = sum(self._fraction_list[:predicate_index])
         hi = lo + self._fraction_list[predicate_index]
-        self._lo, self._hi = lo, hi
+        # Boundaries scaled into bucket space exactly as the reference does.
+        self._bucket_lo = lo * (self._MODULUS - 1)
+        self._bucket_hi = hi * (self._MODULUS - 1)
 
     def get_fields(self):
         return {self._predicate_field}
 
     def _bucket(self, value):
-        h = hashlib.md5(str(value).encode('utf-8')).hexdigest()[:8]
-        return int(h, 16) / self._MAX_HASH
+        h = hashlib.md5(str(value).encode('utf-8')).hexdigest()
+        return int(h, 16) % self._MODULUS
 
     def do_include(self, values):
+        if self._predicate_field not in values:
+            raise ValueError('Tested values do not have split key: %s'
+                             % self._predicate_field)
         b = self._bucket(values[self._predicate_field])
-        return self._lo <= b < self._hi
+        return self._bucket_lo <= b < self._bucket_hi
 
     def do_include_vectorized(self, columns):
         """Batch evaluation.  The md5-of-str bucketing is the SPLIT
@@ -178,12 +191,11 @@ class in_pseudorandom_split(PredicateBase):
         hash itself cannot be replaced by a vectorizable one — this runs
         the same digest per row with the Python overhead hoisted out of
         the loop."""
-        import numpy as np
         col = columns[self._predicate_field]
         md5 = hashlib.md5
-        lo, hi, mx = self._lo, self._hi, self._MAX_HASH
+        lo, hi, mod = self._bucket_lo, self._bucket_hi, self._MODULUS
         out = np.empty(len(col), dtype=bool)
         for i, v in enumerate(col):
-            h = int(md5(str(v).encode('utf-8')).hexdigest()[:8], 16) / mx
-            out[i] = lo <= h < hi
+            b = int(md5(str(v).encode('utf-8')).hexdigest(), 16) % mod
+            out[i] = lo <= b < hi
         return out

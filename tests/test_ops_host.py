@@ -94,3 +94,17 @@ def test_jpeg_parse_rejects_progressive():
     t = torch.frombuffer(bytearray(d), dtype=torch.uint8)
     with pytest.raises(RuntimeError, match='non-baseline'):
         e.jpeg_parse_batch(t, torch.tensor([0]), torch.tensor([len(d)]))
+
+
+def test_walk_pages_corrupt_binary_length_fails_loudly():
+    """A corrupt T_BINARY length >= 2^63 casts to a negative skip; the
+    cursor must fail (raise) instead of moving backwards and re-parsing
+    forever (thrift_pages.cpp skip_bytes)."""
+    e = _ext()
+    # Thrift compact struct: field delta=4 (crc, skipped via skip_value)
+    # declared as T_BINARY(8), then a 10-byte uvarint >= 2^63.
+    blob = bytes([0x48]) + bytes([0xFF] * 9) + bytes([0x01]) + bytes(16)
+    t = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+    with pytest.raises(RuntimeError):
+        e.parquet_walk_pages(t, torch.tensor([0]),
+                             torch.tensor([len(blob)]))

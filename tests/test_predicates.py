@@ -1,5 +1,6 @@
 """Predicates (parity: reference tests/test_predicates.py)."""
 import numpy as np
+import pytest
 
 from petastorm_amd.predicates import (in_intersection, in_lambda, in_negate,
                                       in_pseudorandom_split, in_reduce,
@@ -81,3 +82,28 @@ def test_pseudorandom_split_vectorized_matches_scalar():
     scalar = np.array([p.do_include({'id': i}) for i in ids])
     np.testing.assert_array_equal(vec, scalar)
     assert 200 < vec.sum() < 400  # roughly the 60% fraction
+
+
+def test_pseudorandom_split_upstream_bucket_contract():
+    """Split membership must equal upstream petastorm's rule:
+    int(md5(str(v)).hexdigest(), 16) % sys.maxsize vs fraction*(maxsize-1)
+    boundaries (reference predicates.py:39-41,172-182)."""
+    import hashlib
+    import sys
+    splits = [0.5, 0.2, 0.3]
+    borders = [sum(splits[:i + 1]) for i in range(len(splits))]
+    ids = ['vol-%d' % i for i in range(500)] + [str(i) for i in range(500)]
+    for idx in range(3):
+        p = in_pseudorandom_split(splits, idx, 'id')
+        lo = (borders[idx - 1] if idx else 0) * (sys.maxsize - 1)
+        hi = borders[idx] * (sys.maxsize - 1)
+        for v in ids:
+            bucket = int(hashlib.md5(str(v).encode('utf-8')).hexdigest(),
+                         16) % sys.maxsize
+            assert p.do_include({'id': v}) == (lo <= bucket < hi)
+
+
+def test_pseudorandom_split_missing_field_raises():
+    p = in_pseudorandom_split([0.5, 0.5], 0, 'id')
+    with pytest.raises(ValueError):
+        p.do_include({'other': 1})
