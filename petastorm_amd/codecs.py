@@ -208,6 +208,13 @@ class CompressedNdarrayCodec(DataframeColumnCodec):
         return zlib.compress(memfile.getvalue(), self.level)
 
     def decode(self, unischema_field, value):
+        if bytes(value[:2]) == b'PK':
+            # np.savez_compressed container: what UPSTREAM petastorm's
+            # CompressedNdarrayCodec writes (reference codecs.py:193-198).
+            # Accepted for read interop with reference-written datasets.
+            with np.load(io.BytesIO(bytes(value)),
+                         allow_pickle=False) as npz:
+                return npz[npz.files[0]]
         raw = zlib.decompress(value)
         return np.load(io.BytesIO(raw), allow_pickle=False)
 

@@ -269,6 +269,12 @@ def get_schema(fs, path_or_paths):
             md = pq.ParquetFile(f).schema_arrow.metadata
         if md and UNISCHEMA_KEY.encode() in md:
             return Unischema.from_json(md[UNISCHEMA_KEY.encode()].decode())
+    # 3) upstream-petastorm store: pickled Unischema in _common_metadata
+    # (read-only interop; reference etl/dataset_metadata.py:356-385)
+    from petastorm_amd.etl import interop
+    schema = interop.load_reference_unischema(fs, paths)
+    if schema is not None:
+        return schema
     raise ValueError('Dataset at {} has no petastorm_amd schema metadata. Use '
                      'make_batch_reader for plain Parquet stores.'
                      .format(path_or_paths))

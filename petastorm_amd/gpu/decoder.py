@@ -963,9 +963,60 @@ class GpuRowGroupDecoder(object):
             t = t.to(torch.int64) & 0xFFFFFFFF
         return t
 
+    def _is_npz_column(self, col, field):
+        """True when the column's payloads are np.savez(_compressed) zip
+        containers — what UPSTREAM petastorm's CompressedNdarrayCodec
+        writes (reference codecs.py:193-198) — rather than this
+        framework's bare zlib(npy) framing.  Cached per field name: the
+        peek costs a device sync, so it runs once per reader."""
+        cache = getattr(self, '_npz_field_cache', None)
+        if cache is None:
+            cache = self._npz_field_cache = {}
+        if field.name in cache:
+            return cache[field.name]
+        if col.n == 0:
+            return False  # don't cache an empty-row-group answer
+        if col.host_buf is not None and col.host_val_off is not None:
+            off = int(col.host_val_off[0])
+            head = bytes(col.host_buf[off:off + 2].numpy().tobytes())
+        else:
+            off = int(col.val_off[0].item())
+            head = bytes(col.device_buf[off:off + 2].cpu().numpy().tobytes())
+        cache[field.name] = (head == b'PK')
+        return cache[field.name]
+
+    def _npz_deflate_segments(self, col):
+        """Per-value (seg_off, seg_len) of the raw-DEFLATE stream inside
+        each one-entry zip container.  Local file header: 30 fixed bytes +
+        name_len(@26) + extra_len(@28); compression method(@8) must be 8
+        (deflate).  Returns None if any entry is not deflate-compressed."""
+        ext = self._ext
+        dev = self.device
+        n = col.n
+        cap = 64  # header + 'arr_0.npy' + zip extras fit comfortably
+        hdr = torch.empty(n * cap + _SLACK, dtype=torch.uint8, device=dev)
+        hdr_off = torch.arange(n, dtype=torch.int64, device=dev) * cap
+        hdr_take = torch.minimum(
+            col.val_len.to(torch.int64),
+            torch.full((n,), cap, dtype=torch.int64, device=dev))
+        ext.varlen_gather(col.device_buf, col.val_off, hdr_take, hdr,
+                          hdr_off)
+        h = hdr[:n * cap].view(n, cap).cpu().numpy().astype(np.int64)
+        method = h[:, 8] | (h[:, 9] << 8)
+        if not (method == 8).all():
+            return None  # stored (uncompressed) npz: CPU assist
+        name_len = h[:, 26] | (h[:, 27] << 8)
+        extra_len = h[:, 28] | (h[:, 29] << 8)
+        data_off = 30 + name_len + extra_len
+        seg_off = col.val_off + self._up(data_off)
+        seg_len = col.val_len.to(torch.int64) - self._up(data_off)
+        return seg_off, seg_len
+
     def decode_compressed_ndarray_column(self, col, field):
         """CompressedNdarrayCodec: zlib(npy) -> inflate kernel -> dense
-        [n, *shape] tensor (reference petastorm/codecs.py:174-212)."""
+        [n, *shape] tensor (reference petastorm/codecs.py:174-212).
+        Upstream-written npz containers inflate through the same kernel in
+        raw-DEFLATE mode after host-side zip header parsing."""
         ext = self._ext
         dev = self.device
         np_dtype = np.dtype(field.numpy_dtype)
@@ -976,7 +1027,14 @@ class GpuRowGroupDecoder(object):
         row_bytes = elem * np_dtype.itemsize
         cap = row_bytes + 256  # npy header upper bound
         n = col.n
-        # each value is a single zlib stream: segment table == value table
+        # each value is a single compressed stream: segment == value
+        seg_off, seg_len, mode = col.val_off, col.val_len.to(torch.int64), 0
+        if self._is_npz_column(col, field):
+            segs = self._npz_deflate_segments(col)
+            if segs is None:
+                return None
+            seg_off, seg_len = segs
+            mode = 1  # raw deflate (zip entries carry no zlib header)
         seg_first = torch.arange(n, dtype=torch.int32, device=dev)
         seg_count = torch.ones(n, dtype=torch.int32, device=dev)
         raw = torch.empty(n * cap + _SLACK, dtype=torch.uint8, device=dev)
@@ -984,9 +1042,9 @@ class GpuRowGroupDecoder(object):
         raw_cap = torch.full((n,), cap, dtype=torch.int64, device=dev)
         produced = torch.zeros(n, dtype=torch.int64, device=dev)
         status = torch.zeros(n, dtype=torch.int32, device=dev)
-        ext.inflate_batch(col.device_buf, col.val_off,
-                          col.val_len.to(torch.int64), seg_first, seg_count,
-                          raw, raw_off, raw_cap, produced, 0, status)
+        ext.inflate_batch(col.device_buf, seg_off, seg_len, seg_first,
+                          seg_count, raw, raw_off, raw_cap, produced, mode,
+                          status)
         self._check(status, 'inflate:' + field.name)
         pay_off = torch.empty(n, dtype=torch.int64, device=dev)
         pay_len = torch.empty(n, dtype=torch.int64, device=dev)
