@@ -43,11 +43,35 @@ def _init_dist(world):
     import torch.distributed as dist
     if world <= 1 or dist.is_initialized():
         return None
-    backend = os.environ.get(
-        'PSA_DIST_BACKEND',
-        'nccl' if torch.cuda.is_available() else 'gloo')
+    backend = os.environ.get('PSA_DIST_BACKEND')
+    if backend is None:
+        # nccl (=RCCL on ROCm) whenever every rank can own a GPU; gloo when
+        # ranks must share one GPU (single-GPU validation boxes: two RCCL
+        # ranks on one device deadlock)
+        if torch.cuda.is_available() and \
+                torch.cuda.device_count() >= world:
+            backend = 'nccl'
+        else:
+            backend = 'gloo'
     dist.init_process_group(backend=backend)
     return dist
+
+
+def _self_spawn(args):
+    """The driver may invoke ``bench.py --gpus N`` directly (no torchrun);
+    spawn one rank per GPU ourselves via torch.distributed.run."""
+    import socket
+    import subprocess
+    with socket.socket() as s:
+        s.bind(('127.0.0.1', 0))
+        port = s.getsockname()[1]
+    cmd = [sys.executable, '-m', 'torch.distributed.run',
+           '--nnodes=1', '--nproc-per-node', str(args.gpus),
+           '--master-addr', '127.0.0.1', '--master-port', str(port),
+           os.path.abspath(__file__)] + sys.argv[1:]
+    env = dict(os.environ)
+    env.setdefault('HSA_ENABLE_IPC_MODE_LEGACY', '0')
+    return subprocess.call(cmd, env=env)
 
 
 def _sync(device):
@@ -290,11 +314,40 @@ def _run_timed(args, step_fn, device, dist, world):
     for _ in range(args.warmup):
         warm_samples += step_fn()
     _sync(device)
+    # Calibrate how many loader batches form one timed step so the timed
+    # region spans >= --min-region seconds regardless of the requested step
+    # count (a 20-step imagenet run is ~22 ms of GPU time otherwise —
+    # invisible to SMI sampling and too short to trust).
+    batches_per_step = 1
+    if args.min_region > 0:
+        t0 = time.perf_counter()
+        cal = 0
+        while cal < 3 or time.perf_counter() - t0 < 0.25:
+            step_fn()
+            cal += 1
+            if cal >= 10000:
+                break
+        _sync(device)
+        per_batch = (time.perf_counter() - t0) / cal
+        # 1.25x headroom: steady-state batches run faster than the
+        # calibration window (caches warm, pipeline full)
+        batches_per_step = max(
+            1, int(np.ceil(1.25 * args.min_region /
+                           (args.steps * per_batch))))
+        if dist is not None:
+            # all ranks must agree or lock-step collectives skew
+            t = torch.tensor([batches_per_step], dtype=torch.int64)
+            if torch.cuda.is_available():
+                t = t.to(device)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            batches_per_step = int(t.item())
     _barrier(dist)
+    _sync(device)
     t0 = time.perf_counter()
     samples = 0
     for _ in range(args.steps):
-        samples += step_fn()
+        for _ in range(batches_per_step):
+            samples += step_fn()
     _sync(device)
     _barrier(dist)
     elapsed = time.perf_counter() - t0
@@ -311,7 +364,8 @@ def _run_timed(args, step_fn, device, dist, world):
         dist.all_reduce(s, op=dist.ReduceOp.SUM)
         samples = int(s.item())
     return {'elapsed_s': elapsed, 'samples': samples,
-            'ms_per_step': elapsed * 1000.0 / args.steps}
+            'ms_per_step': elapsed * 1000.0 / args.steps,
+            'batches_per_step': batches_per_step}
 
 
 CONFIGS = {
@@ -330,7 +384,15 @@ def main():
     ap.add_argument('--batch-size', type=int, default=256)
     ap.add_argument('--config', choices=sorted(CONFIGS), default='imagenet')
     ap.add_argument('--rows', type=int, default=None)
+    ap.add_argument('--min-region', type=float,
+                    default=float(os.environ.get('PSA_MIN_REGION', '5.0')),
+                    help='minimum timed-region seconds: each of the K steps '
+                         'consumes as many loader batches as needed to span '
+                         'this (0 disables; one step == one batch then)')
     args = ap.parse_args()
+
+    if args.gpus > 1 and 'WORLD_SIZE' not in os.environ:
+        sys.exit(_self_spawn(args))
 
     si = os.environ.get('PSA_SWITCH_INTERVAL')
     if si:
@@ -355,6 +417,10 @@ def main():
 
     if rank == 0:
         value = result['samples'] / result['elapsed_s']
+        config = dict(config)
+        bps = result.get('batches_per_step', 1)
+        config['batches_per_step'] = bps
+        config['global_batch'] = config.get('global_batch', 0) * bps
         out = {
             'metric': 'samples/sec/node (make_batch_reader->PyTorch '
                       'DataLoader), {} config'.format(args.config),
@@ -364,6 +430,7 @@ def main():
             'steps': args.steps,
             'warmup': args.warmup,
             'ms_per_step': round(result['ms_per_step'], 3),
+            'timed_region_s': round(result['elapsed_s'], 3),
             'higher_is_better': True,
             'scaling': 'weak',
             'vs_baseline': round(value / BASELINE_SAMPLES_PER_SEC, 3),

@@ -101,3 +101,31 @@ def test_reader_throughput_spawned_process(test_dataset):
                                measure_cycles_count=5, loaders_count=2,
                                spawn_new_process=True)
     assert result.samples_per_second > 0
+
+
+def test_bench_py_contract(tmp_path):
+    """bench.py (driver contract): helloworld config on CPU prints one JSON
+    line with the required keys, and the timed region respects
+    --min-region."""
+    import json
+    import os
+    import subprocess
+    import sys
+    env = dict(os.environ, PSA_BENCH_DATA=str(tmp_path))
+    out = subprocess.run(
+        [sys.executable, 'bench.py', '--config', 'helloworld',
+         '--steps', '3', '--warmup', '1', '--batch-size', '10',
+         '--min-region', '1.0'],
+        capture_output=True, text=True, timeout=300,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        env=env)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith('{')][-1]
+    j = json.loads(line)
+    for key in ('metric', 'value', 'unit', 'n_gpus', 'steps', 'warmup',
+                'ms_per_step', 'higher_is_better', 'scaling', 'vs_baseline',
+                'dtype', 'data', 'config'):
+        assert key in j, key
+    assert j['steps'] == 3 and j['n_gpus'] == 1
+    assert j['timed_region_s'] >= 0.45
+    assert j['config']['batches_per_step'] >= 1
