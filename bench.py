@@ -41,7 +41,12 @@ def _dist_env():
 
 def _init_dist(world):
     import torch.distributed as dist
-    if world <= 1 or dist.is_initialized():
+    if dist.is_initialized():
+        return None
+    # Under torchrun, init the process group even at world=1 so the real
+    # backend (RCCL on a GPU box) initializes and the epoch collectives
+    # run over it — keeps the 8-GPU path exercised on 1-GPU leases.
+    if world <= 1 and 'WORLD_SIZE' not in os.environ:
         return None
     backend = os.environ.get('PSA_DIST_BACKEND')
     if backend is None:
