@@ -48,6 +48,7 @@ def _init_dist(world):
     # run over it — keeps the 8-GPU path exercised on 1-GPU leases.
     if world <= 1 and 'WORLD_SIZE' not in os.environ:
         return None
+    kwargs = {}
     backend = os.environ.get('PSA_DIST_BACKEND')
     if backend is None:
         # nccl (=RCCL on ROCm) whenever every rank can own a GPU; gloo when
@@ -58,7 +59,11 @@ def _init_dist(world):
             backend = 'nccl'
         else:
             backend = 'gloo'
-    dist.init_process_group(backend=backend)
+    if backend == 'nccl':
+        local = int(os.environ.get('LOCAL_RANK', '0'))
+        local = min(local, torch.cuda.device_count() - 1)
+        kwargs['device_id'] = torch.device('cuda', local)
+    dist.init_process_group(backend=backend, **kwargs)
     return dist
 
 
