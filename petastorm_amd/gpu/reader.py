@@ -471,6 +471,20 @@ class GpuBatchReader(object):
         return (piece, cols, pmeta, self._decoder.take_pending())
 
     # ------------------------------------------------------------------
+    @staticmethod
+    def _is_string_field(field):
+        """Scalar str/bytes field (incl. ScalarCodec'd and schema-inferred
+        string columns): eligible for the device string path."""
+        if field is None or field.shape not in ((), None):
+            return False
+        dt = field.numpy_dtype
+        if dt in (np.str_, np.bytes_):
+            return True
+        try:
+            return np.dtype(dt).kind in 'SU'
+        except TypeError:
+            return False
+
     def _decode_piece(self, piece, host, meta, plan=None):
         t0 = time.perf_counter()
         with _TraceRange('psa.decode_rowgroup'):
@@ -500,6 +514,10 @@ class GpuBatchReader(object):
                         codec is None and field is not None and
                         field.shape not in ((), None)):
                     decoded = self._decoder.decode_ndarray_column(col, field)
+                elif self._is_string_field(field):
+                    # strings/raw binary: device page decode + boundary
+                    # materialization (no pyarrow re-read)
+                    decoded = self._decoder.decode_string_column(col, field)
                 if decoded is None:
                     assist.append(name)
                 else:

@@ -333,3 +333,27 @@ def test_dryrun_npz_interop_column(stub_decoder, tmp_path):
                                              sch.fields['matrix_z'])
     assert z is None or z.shape == (4, 4, 5)
     assert 'inflate_batch' in stub.calls or 'varlen_gather' in stub.calls
+
+
+def test_dryrun_string_column_host_visible_exact(stub_decoder, tmp_path):
+    """Uncompressed REQUIRED string column: the device string path's
+    host-visible branch materializes EXACT values with no GPU (and no
+    pyarrow re-read) — value correctness of decode_string_column."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    dec, stub = stub_decoder
+    d = tmp_path / 'strs'
+    d.mkdir()
+    vals = ['s-%d' % i for i in range(40)] + ['', 'unicode-é中']
+    table = pa.Table.from_arrays(
+        [pa.array(np.arange(len(vals), dtype=np.int64)),
+         pa.array(vals, type=pa.string())],
+        schema=pa.schema([pa.field('id', pa.int64(), nullable=False),
+                          pa.field('name', pa.string(), nullable=False)]))
+    pq.write_table(table, str(d / 'p.parquet'), compression='none',
+                   use_dictionary=False)
+    out, sch = _decode_all(dec, 'file://' + str(d), ['id', 'name'])
+    col = out['name']
+    assert isinstance(col, ByteArrayColumn) and col.host_buf is not None
+    got = dec.decode_string_column(col, sch.fields['name'])
+    assert got.tolist() == vals
