@@ -243,27 +243,114 @@ class _EagerValue(object):
         return self._value
 
 
+def _apply_element_fn(fn, element):
+    """tf.data calls map/flat_map fns with plain tuples unpacked but
+    namedtuples passed whole (nest semantics)."""
+    if isinstance(element, tuple) and not hasattr(element, '_fields'):
+        return fn(*element)
+    return fn(element)
+
+
+def _as_eager(v):
+    if isinstance(v, _EagerValue):
+        return v
+    arr = np.asarray(v)
+    dt = string if arr.dtype.kind in 'SUO' else as_dtype(arr.dtype)
+    return _EagerValue(arr if dt is not string else v, dt)
+
+
+AUTOTUNE = -1
+
+
 class Dataset(object):
-    def __init__(self, gen, dtypes, map_fn=None):
-        self._gen = gen
-        self._dtypes = dtypes
-        self._map_fn = map_fn
+    """Element-iterator dataset supporting the ops the adapters use:
+    from_generator, from_tensor_slices, map, flat_map, batch, prefetch."""
+
+    def __init__(self, factory):
+        self._factory = factory  # callable -> iterator of elements
 
     @staticmethod
     def from_generator(generator, output_types):
-        return Dataset(generator, tuple(output_types))
+        dtypes = tuple(output_types)
+
+        def factory():
+            for vals in generator():
+                if not isinstance(vals, tuple):
+                    vals = (vals,)
+                yield tuple(
+                    _EagerValue(_check_value(v, dt, None, 'dataset'), dt)
+                    for v, dt in zip(vals, dtypes))
+        return Dataset(factory)
+
+    @staticmethod
+    def from_tensor_slices(element):
+        def factory():
+            if hasattr(element, '_fields'):
+                fields = [_as_eager(v) for v in element]
+                n = len(np.asarray(fields[0].numpy()))
+                for i in range(n):
+                    yield element.__class__(*[
+                        _as_eager(np.asarray(f.numpy())[i])
+                        for f in fields])
+            else:
+                parts = element if isinstance(element, tuple) else (element,)
+                fields = [_as_eager(v) for v in parts]
+                n = len(np.asarray(fields[0].numpy()))
+                for i in range(n):
+                    yield tuple(_as_eager(np.asarray(f.numpy())[i])
+                                for f in fields)
+        return Dataset(factory)
 
     def map(self, fn):
-        return Dataset(self._gen, self._dtypes, map_fn=fn)
+        def factory():
+            for e in self._factory():
+                yield _apply_element_fn(fn, e)
+        return Dataset(factory)
+
+    def flat_map(self, fn):
+        def factory():
+            for e in self._factory():
+                sub = _apply_element_fn(fn, e)
+                for x in sub:
+                    yield x
+        return Dataset(factory)
+
+    def batch(self, batch_size, drop_remainder=False):
+        def factory():
+            buf = []
+            for e in self._factory():
+                buf.append(e)
+                if len(buf) == batch_size:
+                    yield self._stack(buf)
+                    buf = []
+            if buf and not drop_remainder:
+                yield self._stack(buf)
+        return Dataset(factory)
+
+    @staticmethod
+    def _stack(elements):
+        first = elements[0]
+
+        def stack_field(i_or_name):
+            vals = [np.asarray((e[i_or_name] if isinstance(i_or_name, int)
+                                else getattr(e, i_or_name)).numpy())
+                    for e in elements]
+            try:
+                return _as_eager(np.stack(vals))
+            except ValueError:
+                arr = np.empty(len(vals), dtype=object)
+                arr[:] = vals
+                return _as_eager(arr)
+        if hasattr(first, '_fields'):
+            return first.__class__(*[stack_field(i)
+                                     for i in range(len(first))])
+        return tuple(stack_field(i) for i in range(len(first)))
+
+    def prefetch(self, n):
+        return self
 
     def __iter__(self):
-        for vals in self._gen():
-            if not isinstance(vals, tuple):
-                vals = (vals,)
-            eager = tuple(_EagerValue(_check_value(v, dt, None, 'dataset'),
-                                      dt)
-                          for v, dt in zip(vals, self._dtypes))
-            yield self._map_fn(*eager) if self._map_fn else eager
+        return iter(self._factory())
 
 
 def build_module():
@@ -290,4 +377,5 @@ def build_module():
 
     tf.data = types.ModuleType('tensorflow.data')
     tf.data.Dataset = Dataset
+    tf.data.AUTOTUNE = AUTOTUNE
     return tf

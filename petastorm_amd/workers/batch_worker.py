@@ -93,8 +93,15 @@ def arrow_table_to_numpy_dict(table, schema, decode_codecs=True):
         elif pt.is_list(t) or pt.is_large_list(t) or pt.is_fixed_size_list(t):
             pylist = col.to_pylist()
             try:
-                mat = np.vstack([np.asarray(v) for v in pylist]) if pylist \
-                    else np.empty((0, 0))
+                elem_np = np.dtype(t.value_type.to_pandas_dtype())
+            except (NotImplementedError, TypeError):
+                elem_np = None
+            try:
+                # preserve the arrow element dtype (to_pylist widens
+                # float32 -> python float -> float64 otherwise)
+                mat = np.vstack([np.asarray(v, dtype=elem_np)
+                                 for v in pylist]) if pylist \
+                    else np.empty((0, 0), dtype=elem_np)
                 if field is not None and field.shape and \
                         all(d is not None for d in field.shape) and \
                         len(field.shape) > 1:
@@ -102,7 +109,9 @@ def arrow_table_to_numpy_dict(table, schema, decode_codecs=True):
                 out[name] = mat
             except ValueError:
                 arr = np.empty(len(pylist), dtype=object)
-                arr[:] = pylist
+                arr[:] = [v if elem_np is None or v is None
+                          else np.asarray(v, dtype=elem_np)
+                          for v in pylist]
                 out[name] = arr
         elif pt.is_decimal(t):
             arr = np.empty(len(col), dtype=object)
