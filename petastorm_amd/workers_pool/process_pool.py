@@ -142,6 +142,20 @@ class ProcessPool(object):
                 try:
                     kind, msg = self._results_q.get(timeout=0.05)
                 except Exception:
+                    # a crashed worker (segfault, OOM-kill, os._exit) never
+                    # posts its 'processed' control message — detect the
+                    # death instead of spinning forever (reference keeps
+                    # workers under a zmq socket whose closure surfaces
+                    # similarly, process_pool.py:289-294)
+                    dead = [p for p in self._procs
+                            if p.exitcode is not None and p.exitcode != 0]
+                    if dead and self._results_q.empty():
+                        self.stop()
+                        self.join()
+                        raise RuntimeError(
+                            'Worker process(es) died with exit code(s) {} '
+                            'before finishing their work items'
+                            .format([p.exitcode for p in dead]))
                     continue
             if kind == 'payload':
                 return self._serializer.deserialize(msg)

@@ -13,3 +13,75 @@ def test_process_pool_roundtrip(test_dataset):
     by_id = {int(x.id): x for x in rows}
     src = test_dataset['rows'][0]
     np.testing.assert_array_equal(by_id[int(src['id'])].matrix, src['matrix'])
+
+
+# ---------------------------------------------------------------------------
+# crash / exception / shutdown behavior (reference
+# workers_pool/tests/test_workers_pool.py stress suite)
+# ---------------------------------------------------------------------------
+import pytest
+
+from petastorm_amd.test_util.stub_workers import (CrashingWorker, EchoWorker,
+                                                  FailingWorker, SlowWorker)
+from petastorm_amd.workers_pool import EmptyResultError
+from petastorm_amd.workers_pool.process_pool import ProcessPool
+
+
+def test_process_pool_many_items_all_arrive():
+    pool = ProcessPool(3)
+    pool.start(EchoWorker)
+    try:
+        n = 60
+        for i in range(n):
+            pool.ventilate(i)
+        got = set()
+        for _ in range(n):
+            tag, wid, item = pool.get_results()
+            assert tag == 'echo'
+            got.add(item)
+        assert got == set(range(n))
+        with pytest.raises(EmptyResultError):
+            pool.get_results()
+        assert pool.diagnostics['items_processed'] == n
+    finally:
+        pool.stop()
+        pool.join()
+
+
+def test_process_pool_worker_exception_reraised():
+    pool = ProcessPool(2)
+    pool.start(FailingWorker, worker_args=3)
+    for i in [1, 2, 3]:
+        pool.ventilate(i)
+    with pytest.raises(ValueError, match='boom'):
+        for _ in range(3):
+            pool.get_results()
+
+
+def test_process_pool_worker_crash_detected():
+    """A hard worker death (os._exit — no exception message possible) must
+    surface as a loud error, not an infinite get_results() spin."""
+    pool = ProcessPool(1)
+    pool.start(CrashingWorker)
+    pool.ventilate('die')
+    pool.ventilate('never-processed')
+    with pytest.raises(RuntimeError, match='died with exit code'):
+        for _ in range(2):
+            pool.get_results()
+
+
+def test_process_pool_stop_with_unconsumed_results():
+    """Slow-joiner shutdown: stop()+join() must return promptly even with
+    results still queued and items in flight (reference :272-301)."""
+    import time
+    pool = ProcessPool(2)
+    pool.start(SlowWorker)
+    for i in range(10):
+        pool.ventilate(i)
+    pool.get_results()  # consume one, leave the rest queued/in-flight
+    t0 = time.time()
+    pool.stop()
+    pool.join()
+    assert time.time() - t0 < 20
+    with pytest.raises(EmptyResultError):
+        pool.get_results()
