@@ -270,8 +270,14 @@ class CompressedImageCodec(DataframeColumnCodec):
             # dominates at 1).  Override with PSA_JPEG_RST_BLOCKS.
             import os
             rst = int(os.environ.get('PSA_JPEG_RST_BLOCKS', '2'))
-            img.save(buf, format='JPEG', quality=self.quality,
-                     restart_marker_blocks=rst)
+            if rst > 0:
+                img.save(buf, format='JPEG', quality=self.quality,
+                         restart_marker_blocks=rst)
+            else:
+                # rst=0: plain baseline stream with no restart markers —
+                # what cv2/PIL-default writers (and thus most foreign
+                # datasets) produce; decodes as one segment per image
+                img.save(buf, format='JPEG', quality=self.quality)
         else:
             img.save(buf, format='PNG')
         return buf.getvalue()

@@ -129,3 +129,45 @@ def test_bench_py_contract(tmp_path):
     assert j['steps'] == 3 and j['n_gpus'] == 1
     assert j['timed_region_s'] >= 0.45
     assert j['config']['batches_per_step'] >= 1
+
+
+def test_reencode_dataset_adds_restart_markers(tmp_path):
+    """Foreign-style (no-RST) jpeg dataset re-encoded for parallel GPU
+    Huffman decode: output carries DRI/RSTn markers and images stay
+    visually identical (one jpeg generation loss)."""
+    import os
+    import numpy as np
+    from petastorm_amd import make_reader
+    from petastorm_amd.test_util.dataset_gen import create_imagenet_dataset
+    from petastorm_amd.tools.reencode_dataset import reencode_dataset
+
+    src = 'file://' + str(tmp_path / 'src')
+    dst = 'file://' + str(tmp_path / 'dst')
+    os.environ['PSA_JPEG_RST_BLOCKS'] = '0'  # foreign-style: no markers
+    try:
+        create_imagenet_dataset(src, num_rows=6, rowgroup_size_mb=8)
+    finally:
+        del os.environ['PSA_JPEG_RST_BLOCKS']
+
+    rows, cols = reencode_dataset(src, dst, rst_blocks=2, quality=95)
+    assert rows == 6 and cols == ['image']
+
+    import pyarrow.parquet as pq
+    import glob
+    src_file = glob.glob(str(tmp_path / 'src' / '*.parquet'))[0]
+    dst_file = glob.glob(str(tmp_path / 'dst' / '*.parquet'))[0]
+    src_jpg = pq.read_table(src_file, columns=['image'])['image'][0].as_py()
+    dst_jpg = pq.read_table(dst_file, columns=['image'])['image'][0].as_py()
+    assert b'\xff\xdd' not in src_jpg      # no DRI in foreign source
+    assert b'\xff\xdd' in dst_jpg          # DRI present after re-encode
+    assert any(bytes([0xFF, 0xD0 + i]) in dst_jpg for i in range(8))
+
+    with make_reader(src, reader_pool_type='dummy',
+                     shuffle_row_groups=False) as r1, \
+            make_reader(dst, reader_pool_type='dummy',
+                        shuffle_row_groups=False) as r2:
+        for a, b in zip(r1, r2):
+            assert a.label == b.label
+            diff = np.abs(a.image.astype(np.int16) -
+                          b.image.astype(np.int16))
+            assert diff.mean() < 6  # one extra jpeg generation at q95

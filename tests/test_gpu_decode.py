@@ -1058,3 +1058,35 @@ def test_nullable_string_and_binary_columns_gpu(ext, tmp_path):
     for i, rid in enumerate(ids):
         assert svals[i] == names[int(rid)]
         assert bvals[i] == blobs[int(rid)]
+
+
+def test_foreign_jpeg_no_rst_decodes_on_gpu(ext, tmp_path):
+    """Baseline JPEGs WITHOUT restart markers (cv2/PIL defaults — foreign
+    datasets): single-segment-per-image GPU decode, values matching the
+    CPU (PIL) oracle."""
+    import os
+    from petastorm_amd import make_batch_reader, make_reader
+    from petastorm_amd.test_util.dataset_gen import create_imagenet_dataset
+    url = 'file://' + str(tmp_path / 'foreign')
+    os.environ['PSA_JPEG_RST_BLOCKS'] = '0'
+    try:
+        create_imagenet_dataset(url, num_rows=16, rowgroup_size_mb=8)
+    finally:
+        del os.environ['PSA_JPEG_RST_BLOCKS']
+    with make_batch_reader(url, device='cuda',
+                           shuffle_row_groups=False) as r:
+        batches = list(r)
+        assert r.diagnostics['cpu_assist_columns'] == []
+    with make_reader(url, reader_pool_type='dummy',
+                     shuffle_row_groups=False) as cr:
+        cpu = {int(row.label): row.image for row in cr}
+    total = 0
+    for b in batches:
+        labels = b.label.cpu().numpy()
+        imgs = b.image.cpu().numpy()
+        for lab, img in zip(labels, imgs):
+            diff = np.abs(img.astype(np.int16) -
+                          cpu[int(lab)].astype(np.int16))
+            assert diff.max() <= 2, diff.max()
+            total += 1
+    assert total == 16
