@@ -120,7 +120,7 @@ def bench_imagenet(args, rank, world, device, dist):
     # (sharding is per row-group, reference reader.py:573-597)
     # uniform 256-row row groups, 24 per 6144 rows: shards stay equal-sized
     # so the RCCL epoch collectives stay in lock-step across ranks
-    rpg = int(os.environ.get('PSA_IMAGENET_RPG', '256'))
+    rpg = int(os.environ.get('PSA_IMAGENET_RPG', '512'))
     n_rows = args.rows or 24 * rpg  # 24 uniform row-groups (equal 8-GPU shards)
     rst = os.environ.get('PSA_JPEG_RST_BLOCKS', '2')
     url = _dataset_dir('imagenet_{}_r{}_g{}'.format(n_rows, rst, rpg), rank,
@@ -150,7 +150,7 @@ def bench_imagenet(args, rank, world, device, dist):
         shard_count=world if world > 1 else None,
         gpu_options=dict(
             pipeline_depth=int(os.environ.get('PSA_PIPELINE_DEPTH', '6')),
-            decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '4'))))
+            decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '6'))))
     loader = BatchedDataLoader(reader, batch_size=args.batch_size)
 
     it = iter(loader)

@@ -93,6 +93,8 @@ class GpuRowGroupDecoder(object):
         # pinned scalar verdicts for take_pending()'s dispatch-time status
         # reduction (recycled)
         self._host_scalar_free = []
+        # recycled (still-zero) int32 status tensors, keyed by length
+        self._status_free = {}
         # content-addressed device cache for jpeg metadata tensors: the
         # geometry-derived arrays (kmap alone is ~1.2 MB/row-group) are
         # identical across row-groups of same-shaped images, so upload once
@@ -455,7 +457,7 @@ class GpuRowGroupDecoder(object):
                                device=dev)
             u_off = np.zeros(n_pages + 1, dtype=np.int64)
             u_off[1:] = np.cumsum(uncomp_size)
-            status = torch.zeros(n_pages, dtype=torch.int32, device=dev)
+            status = self._status(n_pages)
             if ch['compression'] == 'SNAPPY':
                 ext.snappy_decompress_batch(
                     dbuf, self._up(data_off.astype(np.int64)),
@@ -469,8 +471,7 @@ class GpuRowGroupDecoder(object):
                     blocks = self._lz4_parse_framing(host_buf, pages)
                 blk_src = blocks['src']
                 blk_dst = u_off[blocks['page']] + blocks['dst_rel']
-                bstatus = torch.zeros(len(blk_src), dtype=torch.int32,
-                                      device=dev)
+                bstatus = self._status(len(blk_src))
                 ext.lz4_decompress_batch(
                     dbuf, self._up(blk_src),
                     self._up(blk_src + blocks['src_len']),
@@ -568,8 +569,7 @@ class GpuRowGroupDecoder(object):
             lv_off[1:] = np.cumsum(page_nval)
             levels = torch.empty(int(lv_off[-1]), dtype=torch.int32,
                                  device=dev)
-            status = torch.zeros(len(data_idx), dtype=torch.int32,
-                                 device=dev)
+            status = self._status(len(data_idx))
             ext.rle_hybrid_decode_batch(
                 page_buf, self._up(def_start), self._up(def_end),
                 torch.ones(len(data_idx), dtype=torch.int32, device=dev),
@@ -618,7 +618,7 @@ class GpuRowGroupDecoder(object):
         total = int(page_nval.sum())
         out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
                           device=dev)
-        status = torch.zeros(len(p_start), dtype=torch.int32, device=dev)
+        status = self._status(len(p_start))
         empty8 = torch.empty(0, dtype=torch.uint8, device=dev)
         empty64 = torch.empty(0, dtype=torch.int64, device=dev)
         ext.plain_fixed_decode_batch(
@@ -669,7 +669,7 @@ class GpuRowGroupDecoder(object):
             vbuf = torch.empty(int(v_un.sum()) + _SLACK, dtype=torch.uint8,
                                device=dev)
             ci = np.nonzero(page_compressed)[0]
-            status = torch.zeros(len(ci), dtype=torch.int32, device=dev)
+            status = self._status(len(ci))
             if ch['compression'] == 'SNAPPY':
                 ext.snappy_decompress_batch(
                     dbuf, self._up(val_comp_start[ci]),
@@ -681,8 +681,7 @@ class GpuRowGroupDecoder(object):
                 blocks = self._lz4_spans(
                     host_buf.numpy(), val_comp_start[ci],
                     (v_comp_end - val_comp_start)[ci], v_un[ci])
-                status = torch.zeros(len(blocks['src']), dtype=torch.int32,
-                                     device=dev)
+                status = self._status(len(blocks['src']))
                 blk_dst = u_off[ci][blocks['page']] + blocks['dst_rel']
                 ext.lz4_decompress_batch(
                     dbuf, self._up(blocks['src']),
@@ -720,7 +719,7 @@ class GpuRowGroupDecoder(object):
         total = int(page_nval.sum())
         out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
                           device=dev)
-        status2 = torch.zeros(n, dtype=torch.int32, device=dev)
+        status2 = self._status(n)
         has_def = 2 if ch['max_def'] > 0 else 0
         empty8 = torch.empty(0, dtype=torch.uint8, device=dev)
         ext.plain_fixed_decode_batch(
@@ -750,7 +749,7 @@ class GpuRowGroupDecoder(object):
             o_off[1:] = np.cumsum(counts)[:-1]
             val_off = torch.empty(total, dtype=torch.int64, device=dev)
             val_len = torch.empty(total, dtype=torch.int32, device=dev)
-            status = torch.zeros(len(counts), dtype=torch.int32, device=dev)
+            status = self._status(len(counts))
             ext.byte_array_offsets_batch(
                 page_buf, self._up(val_start), self._up(val_end),
                 self._up(counts.astype(np.int32)),
@@ -800,7 +799,7 @@ class GpuRowGroupDecoder(object):
         dend = dstart + int(size_arr[di])
         d_off = torch.empty(dict_n, dtype=torch.int64, device=dev)
         d_len = torch.empty(dict_n, dtype=torch.int32, device=dev)
-        status = torch.zeros(1, dtype=torch.int32, device=dev)
+        status = self._status(1)
         ext.byte_array_offsets_batch(
             page_buf, self._up(np.array([dstart], dtype=np.int64)),
             self._up(np.array([dend], dtype=np.int64)),
@@ -814,7 +813,7 @@ class GpuRowGroupDecoder(object):
         i_off = np.zeros(len(counts) + 1, dtype=np.int64)
         i_off[1:] = np.cumsum(counts)
         indices = torch.empty(int(i_off[-1]), dtype=torch.int32, device=dev)
-        st2 = torch.zeros(len(counts), dtype=torch.int32, device=dev)
+        st2 = self._status(len(counts))
         ext.rle_hybrid_decode_batch(
             page_buf, self._up(val_start + 1), self._up(val_end),
             self._up(bw), self._up(counts.astype(np.int32)),
@@ -853,7 +852,7 @@ class GpuRowGroupDecoder(object):
         i_off = np.zeros(len(counts) + 1, dtype=np.int64)
         i_off[1:] = np.cumsum(counts)
         indices = torch.empty(int(i_off[-1]), dtype=torch.int32, device=dev)
-        status = torch.zeros(len(counts), dtype=torch.int32, device=dev)
+        status = self._status(len(counts))
         ext.rle_hybrid_decode_batch(
             page_buf, self._up(val_start + 1), self._up(val_end),
             self._up(bw), self._up(counts.astype(np.int32)),
@@ -869,6 +868,16 @@ class GpuRowGroupDecoder(object):
         return out
 
     # ------------------------------------------------------------------
+    def _status(self, n):
+        """Zeroed int32 device status tensor from a recycle pool: statuses
+        are still zero whenever the row-group verdict passes, so pooled
+        tensors need no re-fill (was 3 small FillFunctor launches per
+        row-group)."""
+        free = self._status_free.get(n)
+        if free:
+            return free.pop()
+        return torch.zeros(n, dtype=torch.int32, device=self.device)
+
     def _check(self, status, what):
         """Queue a status tensor for the end-of-row-group flush."""
         self._pending_status.append((what, status))
@@ -892,8 +901,13 @@ class GpuRowGroupDecoder(object):
         ev = host_total = None
         if self.device.type == 'cuda' and torch.cuda.is_available():
             if pending:
-                total = torch.stack(
-                    [s.abs().sum() for _, s in pending]).sum()
+                # ONE cat + abs + sum instead of abs+sum per status tensor
+                # (was ~8 eager launches/row-group, ~5% of GPU time)
+                if len(pending) == 1:
+                    total = pending[0][1].abs().sum()
+                else:
+                    total = torch.cat(
+                        [s.view(-1) for _, s in pending]).abs_().sum()
                 host_total = (self._host_scalar_free.pop()
                               if self._host_scalar_free else
                               torch.empty((), dtype=torch.int64,
@@ -921,6 +935,13 @@ class GpuRowGroupDecoder(object):
         bad = int(host_total.item()) != 0
         if ev is not None:
             self._host_scalar_free.append(host_total)
+        if not bad:
+            # statuses proven all-zero: recycle them un-refilled
+            for _, s in pending:
+                if s.is_cuda:
+                    free = self._status_free.setdefault(s.numel(), [])
+                    if len(free) < 32:
+                        free.append(s)
         if bad:
             for what, s in pending:
                 vals = s.cpu()
@@ -952,7 +973,7 @@ class GpuRowGroupDecoder(object):
         row_bytes = elem * np_dtype.itemsize
         pay_off = torch.empty(col.n, dtype=torch.int64, device=dev)
         pay_len = torch.empty(col.n, dtype=torch.int64, device=dev)
-        status = torch.zeros(1, dtype=torch.int32, device=dev)
+        status = self._status(1)
         ext.npy_payload_offsets(col.device_buf, col.val_off, col.val_len,
                                 pay_off, pay_len, status)
         self._check(status, 'npy:' + field.name)
@@ -1052,14 +1073,14 @@ class GpuRowGroupDecoder(object):
         raw_off = torch.arange(n, dtype=torch.int64, device=dev) * cap
         raw_cap = torch.full((n,), cap, dtype=torch.int64, device=dev)
         produced = torch.zeros(n, dtype=torch.int64, device=dev)
-        status = torch.zeros(n, dtype=torch.int32, device=dev)
+        status = self._status(n)
         ext.inflate_batch(col.device_buf, seg_off, seg_len, seg_first,
                           seg_count, raw, raw_off, raw_cap, produced, mode,
                           status)
         self._check(status, 'inflate:' + field.name)
         pay_off = torch.empty(n, dtype=torch.int64, device=dev)
         pay_len = torch.empty(n, dtype=torch.int64, device=dev)
-        st2 = torch.zeros(1, dtype=torch.int32, device=dev)
+        st2 = self._status(1)
         ext.npy_payload_offsets(raw, raw_off, produced.to(torch.int32),
                                 pay_off, pay_len, st2)
         self._check(st2, 'npy:' + field.name)
@@ -1156,7 +1177,7 @@ class GpuRowGroupDecoder(object):
         raw = torch.empty(int(raw_size.sum()) + _SLACK, dtype=torch.uint8,
                           device=dev)
         produced = torch.zeros(n, dtype=torch.int64, device=dev)
-        status = torch.zeros(n, dtype=torch.int32, device=dev)
+        status = self._status(n)
         ext.inflate_batch(col.device_buf, self._up(meta['seg_off']),
                           self._up(meta['seg_len']),
                           self._up(meta['seg_first']),
@@ -1170,7 +1191,7 @@ class GpuRowGroupDecoder(object):
         out_off[1:] = np.cumsum(out_bytes)[:-1]
         out = torch.empty(int(out_bytes.sum()) + _SLACK, dtype=torch.uint8,
                           device=dev)
-        st2 = torch.zeros(n, dtype=torch.int32, device=dev)
+        st2 = self._status(n)
         ext.png_unfilter_batch(raw, self._up(raw_off), out,
                                self._up(out_off),
                                self._up(meta['height']),
@@ -1234,7 +1255,7 @@ class GpuRowGroupDecoder(object):
         out = torch.empty(int(out_bytes.sum()), dtype=torch.uint8,
                           device=dev)
         n_segs = int(meta['seg_img'].numel())
-        status = torch.zeros(max(n_segs, 1), dtype=torch.int32, device=dev)
+        status = self._status(max(n_segs, 1))
         ext.jpeg_decode_batch(col.device_buf, meta_dev, coef, samples, out,
                               self._up_cached(torch.from_numpy(out_off)),
                               status)

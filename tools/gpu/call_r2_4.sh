@@ -1,0 +1,8 @@
+set -x
+cd /root/repo
+mkdir -p gpurun_out
+python -m pytest tests -m gpu -x -q 2>&1 | tail -3
+for cfg in imagenet scalar ngram; do
+  timeout 420 python bench.py --config $cfg --steps 10 --warmup 5 --min-region 4 > gpurun_out/b2_$cfg.json 2> gpurun_out/b2_$cfg.err
+  echo "$cfg rc=$?:"; tail -1 gpurun_out/b2_$cfg.json | python -c "import json,sys; j=json.load(sys.stdin); print(j['value'], 'region', j['timed_region_s'])" 2>/dev/null || tail -3 gpurun_out/b2_$cfg.err
+done
