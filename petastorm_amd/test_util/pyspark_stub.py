@@ -206,10 +206,38 @@ class _Conf(object):
         return self._d.get(k, default)
 
 
+class RDD(object):
+    """Eager local-list RDD: enough surface for dataset_as_rdd."""
+
+    def __init__(self, items):
+        self._items = list(items)
+
+    def flatMap(self, fn):
+        return RDD(x for it in self._items for x in fn(it))
+
+    def map(self, fn):
+        return RDD(fn(it) for it in self._items)
+
+    def collect(self):
+        return list(self._items)
+
+    def count(self):
+        return len(self._items)
+
+    def take(self, n):
+        return self._items[:n]
+
+
+class SparkContext(object):
+    def parallelize(self, seq, numSlices=None):
+        return RDD(seq)
+
+
 class SparkSession(object):
     def __init__(self):
         self.conf = _Conf()
         self._next_id = 0
+        self.sparkContext = SparkContext()
 
     def createDataFrame(self, pdf, source_id=None):
         if source_id is None:
