@@ -89,7 +89,15 @@ def get_filesystem_and_path_or_paths(url_or_urls, storage_options=None,
         raise ValueError('All dataset URLs must share scheme and netloc; got {}'
                          .format(sorted(schemes)))
     scheme = parsed[0].scheme or 'file'
-    fs = fsspec.filesystem(scheme, **(storage_options or {}))
+    opts = dict(storage_options or {})
+    if scheme in ('hdfs', 'webhdfs') and 'host' not in opts:
+        # resolve HA nameservices from the Hadoop site configuration
+        # (reference hdfs/namenode.py:31-128; see hdfs_config.py)
+        from petastorm_amd.hdfs_config import hdfs_storage_options
+        resolved = hdfs_storage_options(urls[0])
+        resolved.pop('fallback_namenodes', None)
+        opts.update(resolved)
+    fs = fsspec.filesystem(scheme, **opts)
     if retry_attempts > 1 and scheme != 'file':
         fs = RetryingFilesystem(fs, attempts=retry_attempts)
 
