@@ -108,3 +108,26 @@ def test_walk_pages_corrupt_binary_length_fails_loudly():
     with pytest.raises(RuntimeError):
         e.parquet_walk_pages(t, torch.tensor([0]),
                              torch.tensor([len(blob)]))
+
+
+def test_zstd_decoder_under_asan_ubsan():
+    """Build and run the standalone ASAN/UBSan fuzz harness over the
+    from-scratch zstd decoder (sanitizer coverage for host
+    code).  Malformed input must produce error codes, never an
+    out-of-bounds access."""
+    import os
+    import subprocess
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    src = os.path.join(root, 'tools', 'sanitize', 'zstd_fuzz_main.cpp')
+    csrc = os.path.join(root, 'petastorm_amd', 'ops', 'csrc')
+    exe = os.path.join(root, 'tools', 'sanitize', '.zstd_fuzz_bin')
+    build = subprocess.run(
+        ['g++', '-std=c++17', '-O1', '-g',
+         '-fsanitize=address,undefined', '-fno-sanitize-recover=all',
+         '-I', csrc, src, '-o', exe],
+        capture_output=True, text=True, timeout=300)
+    assert build.returncode == 0, build.stderr[-2000:]
+    run = subprocess.run([exe], capture_output=True, text=True,
+                         timeout=300)
+    assert run.returncode == 0, (run.stdout + run.stderr)[-2000:]
+    assert 'zstd fuzz OK' in run.stdout
