@@ -82,17 +82,18 @@ def test_metadata_util_cli(test_dataset, capsys):
     assert 'row groups' in out and 'image_png' in out
 
 
-def test_shuffling_analysis(test_dataset):
-    from petastorm_amd.test_util.shuffling_analysis import \
-        compute_correlation_distribution
-    corr_shuffled = compute_correlation_distribution(
-        test_dataset['url'], 'id',
-        {'shuffle_row_groups': True, 'shuffle_rows': True},
-        num_corr_samples=2)
-    corr_ordered = compute_correlation_distribution(
-        test_dataset['url'], 'id', {'shuffle_row_groups': False},
-        num_corr_samples=2)
-    assert corr_shuffled.mean() < corr_ordered.mean()
+def test_shuffling_analysis(tmp_path):
+    from petastorm_amd.test_util.shuffling_analysis import (
+        compute_correlation_distribution, generate_shuffle_analysis_dataset)
+    url = 'file://' + str(tmp_path / 'shuffle_ds')
+    generate_shuffle_analysis_dataset(url, num_rows=400, row_group_size=40)
+    mean1, std1 = compute_correlation_distribution(
+        url, 'id', shuffle_row_drop_partitions=1, num_corr_samples=3)
+    mean2, std2 = compute_correlation_distribution(
+        url, 'id', shuffle_row_drop_partitions=2, num_corr_samples=3)
+    # shuffled orders decorrelate strongly from the natural order
+    assert mean1 < 0.5 and mean2 < 0.5
+    assert std1 >= 0.0 and std2 >= 0.0
 
 
 def test_reader_throughput_spawned_process(test_dataset):

@@ -35,9 +35,14 @@ def run_config(name, seconds):
     device = torch.device('cuda')
 
     # build the loader exactly as bench.py does, then loop it ourselves
-    if name == 'imagenet':
+    base, _, variant = name.partition('-')
+    if variant:  # e.g. scalar-zstd / scalar-lz4: page-codec variants
+        os.environ['PSA_SCALAR_COMPRESSION'] = variant
+    else:
+        os.environ.pop('PSA_SCALAR_COMPRESSION', None)
+    if base == 'imagenet':
         fn = bench.bench_imagenet
-    elif name == 'scalar':
+    elif base == 'scalar':
         fn = bench.bench_scalar
     else:
         fn = bench.bench_ngram
@@ -77,10 +82,15 @@ def run_config(name, seconds):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument('--minutes', type=float, default=3.0)
+    p.add_argument('--configs', default='imagenet,scalar,scalar-zstd,'
+                   'scalar-lz4,ngram',
+                   help='comma list; scalar-<codec> runs the scalar '
+                        'pipeline with that Parquet page compression')
     args = p.parse_args()
-    per_cfg = args.minutes * 60 / 3
+    configs = [c for c in args.configs.split(',') if c]
+    per_cfg = args.minutes * 60 / len(configs)
     bad = 0
-    for cfg in ('imagenet', 'scalar', 'ngram'):
+    for cfg in configs:
         w = run_config(cfg, per_cfg)
         if len(w) < 2:
             print(f'{cfg}: too few windows ({len(w)})')
