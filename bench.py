@@ -150,7 +150,8 @@ def bench_imagenet(args, rank, world, device, dist):
         shard_count=world if world > 1 else None,
         gpu_options=dict(
             pipeline_depth=int(os.environ.get('PSA_PIPELINE_DEPTH', '6')),
-            decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '6'))))
+            decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '6')),
+            io_threads=int(os.environ.get('PSA_IO_THREADS', '2'))))
     loader = BatchedDataLoader(reader, batch_size=args.batch_size)
 
     it = iter(loader)
