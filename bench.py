@@ -151,7 +151,7 @@ def bench_imagenet(args, rank, world, device, dist):
         gpu_options=dict(
             pipeline_depth=int(os.environ.get('PSA_PIPELINE_DEPTH', '6')),
             decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '6')),
-            io_threads=int(os.environ.get('PSA_IO_THREADS', '2'))))
+            io_threads=int(os.environ.get('PSA_IO_THREADS', '3'))))
     loader = BatchedDataLoader(reader, batch_size=args.batch_size)
 
     it = iter(loader)
