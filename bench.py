@@ -117,10 +117,11 @@ def bench_imagenet(args, rank, world, device, dist):
     from petastorm_amd.unischema import UnischemaField
 
     # >= 8 row-groups are required so every rank of an 8-GPU run gets data
-    # (sharding is per row-group, reference reader.py:573-597)
-    # uniform 256-row row groups, 24 per 6144 rows: shards stay equal-sized
-    # so the RCCL epoch collectives stay in lock-step across ranks
-    rpg = int(os.environ.get('PSA_IMAGENET_RPG', '512'))
+    # (sharding is per row-group, reference reader.py:573-597).
+    # 24 uniform row-groups -> every one of 8 shards gets exactly 3.
+    # rpg x io_threads sweep on MI355X (profiles/RESULTS.md r2.4):
+    # 256=228k, 512=404k, 1024=551k, 2048=385k samples/s
+    rpg = int(os.environ.get('PSA_IMAGENET_RPG', '1024'))
     n_rows = args.rows or 24 * rpg  # 24 uniform row-groups (equal 8-GPU shards)
     rst = os.environ.get('PSA_JPEG_RST_BLOCKS', '2')
     url = _dataset_dir('imagenet_{}_r{}_g{}'.format(n_rows, rst, rpg), rank,
@@ -151,7 +152,7 @@ def bench_imagenet(args, rank, world, device, dist):
         gpu_options=dict(
             pipeline_depth=int(os.environ.get('PSA_PIPELINE_DEPTH', '6')),
             decode_streams=int(os.environ.get('PSA_DECODE_STREAMS', '6')),
-            io_threads=int(os.environ.get('PSA_IO_THREADS', '3'))))
+            io_threads=int(os.environ.get('PSA_IO_THREADS', '4'))))
     loader = BatchedDataLoader(reader, batch_size=args.batch_size)
 
     it = iter(loader)
