@@ -594,15 +594,13 @@ class GpuRowGroupDecoder(object):
                                     else comp_size, val_start, val_end,
                                     page_nval, nonnull_per_page, valid,
                                     n_rows, phys)
-        all_valid = valid is None or (
-            nonnull_per_page is not None and
-            bool((nonnull_per_page == page_nval).all()))
         if data_enc in (_ENC_PLAIN_DICT, _ENC_RLE_DICT) and \
-                phys == 'BYTE_ARRAY' and dict_idx and all_valid:
+                phys == 'BYTE_ARRAY' and dict_idx:
             return self._dict_byte_array(ext, dev, page_buf, page_start,
                                          dict_idx, num_values,
                                          uncomp_size if snappy else comp_size,
-                                         val_start, val_end, page_nval, ch)
+                                         val_start, val_end, page_nval,
+                                         nonnull_per_page, valid, ch)
         return self._cpu_assist_marker(ch['name'])
 
     # ------------------------------------------------------------------
@@ -784,7 +782,7 @@ class GpuRowGroupDecoder(object):
 
     def _dict_byte_array(self, ext, dev, page_buf, page_start, dict_idx,
                          num_values, size_arr, val_start, val_end, page_nval,
-                         ch):
+                         nonnull_per_page, valid, ch):
         """Dictionary-encoded binary column: decode the dictionary page's
         value offsets once, decode the RLE indices, and gather per-row
         (offset, length) — blob bytes are never copied.
@@ -807,7 +805,10 @@ class GpuRowGroupDecoder(object):
             self._up(np.array([0], dtype=np.int64)), d_off, d_len, status)
         self._check(status, 'dictba-dict:' + ch['name'])
 
-        counts = page_nval
+        # RLE indices exist only for NON-NULL rows (def levels already
+        # decoded by the caller when the column is OPTIONAL)
+        counts = nonnull_per_page if nonnull_per_page is not None \
+            else page_nval
         bw_t = torch.stack([page_buf[int(s)] for s in val_start]).cpu()
         bw = bw_t.numpy().astype(np.int32)
         i_off = np.zeros(len(counts) + 1, dtype=np.int64)
@@ -820,9 +821,13 @@ class GpuRowGroupDecoder(object):
             self._up(i_off[:-1]), indices, st2)
         self._check(st2, 'dictba-idx:' + ch['name'])
         idx = indices.long()
-        return ByteArrayColumn(page_buf, d_off.index_select(0, idx),
-                               d_len.index_select(0, idx), None, None,
-                               int(i_off[-1]))
+        col = ByteArrayColumn(page_buf, d_off.index_select(0, idx),
+                              d_len.index_select(0, idx), None, None,
+                              int(i_off[-1]))
+        if valid is not None and nonnull_per_page is not None and \
+                not bool((nonnull_per_page == page_nval).all()):
+            col.valid = valid
+        return col
 
     def _dict_fixed(self, ext, dev, page_buf, page_start, dict_idx,
                     num_values, size_arr, val_start, val_end, page_nval,

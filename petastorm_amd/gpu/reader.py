@@ -480,6 +480,12 @@ class GpuBatchReader(object):
         dt = field.numpy_dtype
         if dt in (np.str_, np.bytes_):
             return True
+        from decimal import Decimal
+        if dt is Decimal:
+            # stored as a string column (codecs._arrow_scalar_type); the
+            # batch route surfaces decimals as strings on CPU too, so the
+            # device string path gives route parity
+            return True
         try:
             return np.dtype(dt).kind in 'SU'
         except TypeError:

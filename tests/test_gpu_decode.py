@@ -1093,3 +1093,67 @@ def test_foreign_jpeg_no_rst_decodes_on_gpu(ext, tmp_path):
                 (diff.mean(), diff.max())
             total += 1
     assert total == 16
+
+
+def test_dict_encoded_nullable_strings_gpu(ext, tmp_path):
+    """Dictionary-encoded string column WITH nulls: def levels + RLE
+    indices + dictionary gather on GPU, nulls as None (previously a CPU
+    assist)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'dictnull'
+    d.mkdir()
+    n = 600
+    cats = ['alpha', 'beta', 'gamma', 'delta']
+    vals = [None if i % 9 == 0 else cats[i % 4] for i in range(n)]
+    table = pa.table({
+        'id': pa.array(np.arange(n, dtype=np.int64)),
+        'cat': pa.array(vals, type=pa.string()),
+    })
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=150,
+                   compression='snappy', use_dictionary=['cat'])
+    with make_batch_reader('file://' + str(d), device='cuda',
+                           shuffle_row_groups=False) as r:
+        got_ids, got_cats = [], []
+        for b in r:
+            got_ids.append(b.id.cpu().numpy())
+            got_cats.append(np.asarray(b.cat, dtype=object))
+        assert r.diagnostics['cpu_assist_columns'] == []
+    ids = np.concatenate(got_ids)
+    catv = np.concatenate(got_cats)
+    for i, rid in enumerate(ids):
+        assert catv[i] == vals[int(rid)], (rid, catv[i], vals[int(rid)])
+
+
+def test_decimal_column_gpu_matches_cpu_batch_route(ext, tmp_path):
+    """Decimal scalar fields ride the device string path; values match
+    the CPU batch route (both surface decimal-as-string on batch
+    readers)."""
+    from decimal import Decimal
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.codecs import ScalarCodec
+    from petastorm_amd.etl.dataset_metadata import materialize_dataset
+    from petastorm_amd.unischema import Unischema, UnischemaField
+    schema = Unischema('D', [
+        UnischemaField('id', np.int64, (), ScalarCodec(), False),
+        UnischemaField('price', Decimal, (), ScalarCodec(), False)])
+    url = 'file://' + str(tmp_path / 'dec')
+    with materialize_dataset(url, schema, 1) as w:
+        w.write_rows([{'id': np.int64(i),
+                       'price': Decimal(i) / Decimal(7)}
+                      for i in range(50)])
+    with make_batch_reader(url, device='cuda',
+                           shuffle_row_groups=False) as r:
+        b = next(iter(r))
+        assert r.diagnostics['cpu_assist_columns'] == []
+        gpu_prices = {int(i): p for i, p in
+                      zip(b.id.cpu().numpy(), np.asarray(b.price))}
+    with make_batch_reader(url, shuffle_row_groups=False) as cr:
+        cb = next(iter(cr))
+        cpu_prices = {int(i): p for i, p in
+                      zip(np.asarray(cb.id), np.asarray(cb.price))}
+    assert set(gpu_prices) == set(cpu_prices)
+    for k in gpu_prices:
+        assert str(gpu_prices[k]) == str(cpu_prices[k])
+        assert Decimal(str(gpu_prices[k])) == Decimal(k) / Decimal(7)
