@@ -110,11 +110,10 @@ def _dataset_dir(tag, rank, dist, gen_fn):
 def bench_imagenet(args, rank, world, device, dist):
     """BASELINE configs 3/4: jpeg CompressedImageCodec + on-GPU decode +
     NHWC->NCHW normalize TransformSpec."""
-    from petastorm_amd import TransformSpec, make_batch_reader
+    from petastorm_amd import make_batch_reader
     from petastorm_amd import ops
     from petastorm_amd.pytorch import BatchedDataLoader
     from petastorm_amd.test_util.dataset_gen import create_imagenet_dataset
-    from petastorm_amd.unischema import UnischemaField
 
     # >= 8 row-groups are required so every rank of an 8-GPU run gets data
     # (sharding is per row-group, reference reader.py:573-597).
@@ -129,21 +128,13 @@ def bench_imagenet(args, rank, world, device, dist):
                        lambda u: create_imagenet_dataset(
                            u, num_rows=n_rows, rows_per_rowgroup=rpg))
 
-    ext = ops.ext()
-    mean = torch.tensor([0.485, 0.456, 0.406], device=device)
-    inv_std = 1.0 / torch.tensor([0.229, 0.224, 0.225], device=device)
-
-    def transform(cols):
-        img = cols['image']
-        out = torch.empty(img.shape[0], 3, img.shape[1], img.shape[2],
-                          dtype=torch.float32, device=img.device)
-        ext.nhwc_to_nchw_normalize(img, out, mean, inv_std, 1.0 / 255.0)
-        return {'image': out, 'label': cols['label']}
-
-    ts = TransformSpec(
-        transform,
-        edit_fields=[UnischemaField('image', np.float32, (3, 224, 224),
-                                    None, False)])
+    ops.ext()  # loud failure if the native extension is missing
+    # fused into the jpeg color kernel epilogue on the GPU route
+    # (transform.FusedImageNormalize): no NHWC uint8 intermediate, no
+    # separate normalize kernel
+    from petastorm_amd.transform import fused_image_normalize
+    ts = fused_image_normalize('image', mean=[0.485, 0.456, 0.406],
+                               std=[0.229, 0.224, 0.225])
     reader = make_batch_reader(
         url, device=str(device), num_epochs=None, shuffle_row_groups=True,
         seed=1234, transform_spec=ts,

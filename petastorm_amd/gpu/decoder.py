@@ -1218,8 +1218,13 @@ class GpuRowGroupDecoder(object):
             t = out.view(n, h, w, c)
         return t.squeeze(-1) if c == 1 else t
 
-    def decode_jpeg_column(self, col, field):
-        """CompressedImageCodec(jpeg): restart-parallel GPU decode."""
+    def decode_jpeg_column(self, col, field, fused_norm=None):
+        """CompressedImageCodec(jpeg): restart-parallel GPU decode.
+
+        ``fused_norm`` (a transform.FusedImageNormalize) switches the color
+        stage to the fused YCbCr->RGB + normalize + NCHW fp32 epilogue —
+        output is [n, 3, H, W] float32 and the separate normalize kernel
+        (and its NHWC uint8 intermediate) never runs."""
         if col.host_buf is None:
             return None  # compressed storage: host can't parse headers
         ext = self._ext
@@ -1253,21 +1258,48 @@ class GpuRowGroupDecoder(object):
         samp_total = int(meta['samp_total'])
         coef = torch.zeros(block_total * 64, dtype=torch.float32, device=dev)
         samples = torch.empty(samp_total, dtype=torch.uint8, device=dev)
+        n_segs = int(meta['seg_img'].numel())
+        uniform = (len(set(widths.tolist())) == 1 and
+                   len(set(heights.tolist())) == 1 and
+                   len(set(ncomp.tolist())) == 1)
+        if fused_norm is not None and uniform and \
+                self.device.type == 'cuda':
+            H, W = int(heights[0]), int(widths[0])
+            out_px = (widths.astype(np.int64) * heights * 3)
+            out_off = np.zeros(n, dtype=np.int64)
+            out_off[1:] = np.cumsum(out_px)[:-1]
+            out = torch.empty(n, 3, H, W, dtype=torch.float32, device=dev)
+            cache = getattr(self, '_fused_norm_cache', None)
+            if cache is None:
+                cache = self._fused_norm_cache = {}
+            key = (tuple(fused_norm.mean), tuple(fused_norm.std))
+            if key not in cache:
+                mean_t = torch.tensor(fused_norm.mean, dtype=torch.float32,
+                                      device=dev)
+                inv_t = 1.0 / torch.tensor(fused_norm.std,
+                                           dtype=torch.float32, device=dev)
+                cache[key] = (mean_t, inv_t)
+            mean_t, inv_t = cache[key]
+            status = self._status(max(n_segs, 1))
+            ext.jpeg_decode_fused_batch(
+                col.device_buf, meta_dev, coef, samples, out.view(-1),
+                self._up_cached(torch.from_numpy(out_off)), mean_t, inv_t,
+                1.0 / fused_norm.scale_div, status)
+            self._check(status, 'jpeg-fused:' + field.name)
+            return out
         out_bytes = (widths.astype(np.int64) * heights *
                      np.where(ncomp == 3, 3, 1))
         out_off = np.zeros(n, dtype=np.int64)
         out_off[1:] = np.cumsum(out_bytes)[:-1]
         out = torch.empty(int(out_bytes.sum()), dtype=torch.uint8,
                           device=dev)
-        n_segs = int(meta['seg_img'].numel())
         status = self._status(max(n_segs, 1))
         ext.jpeg_decode_batch(col.device_buf, meta_dev, coef, samples, out,
                               self._up_cached(torch.from_numpy(out_off)),
                               status)
         self._check(status, 'jpeg:' + field.name)
         # uniform-shape batch -> dense [n, H, W, C]
-        if len(set(widths.tolist())) == 1 and len(set(heights.tolist())) == 1 \
-                and len(set(ncomp.tolist())) == 1:
+        if uniform:
             c = 3 if ncomp[0] == 3 else 1
             t = out.view(n, int(heights[0]), int(widths[0]), c)
             return t if c == 3 else t.squeeze(-1)

@@ -142,6 +142,13 @@ class GpuBatchReader(object):
         self.transform_spec = transform_spec
         self.schema = transform_schema(self._view_schema, transform_spec) \
             if transform_spec else self._view_schema
+        # FusedImageNormalize transforms fuse into the jpeg color kernel
+        # epilogue (transform.py); the callable detects the fused output
+        # at emit time and passes it through
+        from petastorm_amd.transform import FusedImageNormalize
+        f = getattr(transform_spec, 'func', None)
+        self._fused_image_norm = f if isinstance(f, FusedImageNormalize) \
+            else None
         self.batched_output = True
         self.ngram = requested_ngram
         self.last_row_consumed = False
@@ -510,7 +517,11 @@ class GpuBatchReader(object):
                 codec = field.codec if field is not None else None
                 if isinstance(codec, CompressedImageCodec) and \
                         codec.image_codec == 'jpeg':
-                    decoded = self._decoder.decode_jpeg_column(col, field)
+                    fused = (self._fused_image_norm
+                             if self._fused_image_norm is not None and
+                             self._fused_image_norm.field == name else None)
+                    decoded = self._decoder.decode_jpeg_column(
+                        col, field, fused_norm=fused)
                 elif isinstance(codec, CompressedImageCodec):
                     decoded = self._decoder.decode_png_column(col, field)
                 elif isinstance(codec, CompressedNdarrayCodec):

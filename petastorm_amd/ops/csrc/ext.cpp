@@ -88,6 +88,11 @@ py::dict byte_array_host_offsets(torch::Tensor buf, torch::Tensor val_start,
 void jpeg_decode_batch(torch::Tensor data, py::dict meta, torch::Tensor coef,
                        torch::Tensor samples, torch::Tensor out,
                        torch::Tensor out_off, torch::Tensor status);
+void jpeg_decode_fused_batch(torch::Tensor data, py::dict meta,
+                             torch::Tensor coef, torch::Tensor samples,
+                             torch::Tensor out, torch::Tensor out_off,
+                             torch::Tensor mean, torch::Tensor inv_std,
+                             double scale, torch::Tensor status);
 
 }  // namespace psa
 
@@ -131,4 +136,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Host-side JPEG header/segment parse");
   m.def("jpeg_decode_batch", &psa::jpeg_decode_batch,
         "GPU baseline JPEG decode (huffman/idct/color kernels)");
+  m.def("jpeg_decode_fused_batch", &psa::jpeg_decode_fused_batch,
+        "JPEG decode with fused YCbCr->RGB + normalize + NCHW fp32 "
+        "epilogue (skips the NHWC uint8 intermediate)");
 }

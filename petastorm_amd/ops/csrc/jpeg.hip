@@ -448,6 +448,84 @@ __global__ void jpeg_color_kernel(const uint8_t* __restrict__ samples,
   }
 }
 
+// Fused variant: YCbCr->RGB + normalize + NCHW fp32 in one pass.  Replaces
+// color kernel + separate nhwc_to_nchw_normalize for the dominant
+// training-input transform: skips writing/re-reading the NHWC uint8
+// intermediate (154 MB per 1024-image row-group — profiled at ~15% of
+// imagenet GPU time as a separate kernel).  Bit-exact with the two-step
+// path: the u8 clamp/round happens first, then the normalize.
+__global__ void jpeg_color_norm_kernel(
+    const uint8_t* __restrict__ samples, JpegGeom g,
+    float* __restrict__ out, const int64_t* __restrict__ out_off,
+    const float* __restrict__ mean, const float* __restrict__ inv_std,
+    float scale, int n_imgs) {
+  const int img = blockIdx.y;
+  const int W = g.width[img], H = g.height[img];
+  const int64_t plane = (int64_t)W * H;
+  const int nc = g.ncomp[img];
+  const uint8_t* yplane = samples + g.samp_off[img * 3 + 0];
+  const int ystride = g.samp_stride[img * 3 + 0];
+  float* dst = out + out_off[img];          // [3][H][W] floats
+
+  const int hmax = g.comp_h[img * 3 + 0];
+  const int vmax = g.comp_v[img * 3 + 0];
+  const int Wq = (W + 3) >> 2;
+  const int64_t nquads = (int64_t)Wq * H;
+  const uint8_t* cbp = samples + g.samp_off[img * 3 + 1];
+  const uint8_t* crp = samples + g.samp_off[img * 3 + 2];
+  const int cstride = g.samp_stride[img * 3 + 1];
+  const int cw = (W + hmax - 1) / hmax;
+  const int chh = (H + vmax - 1) / vmax;
+  const float m0 = mean[0], m1 = mean[1], m2 = mean[2];
+  const float s0 = inv_std[0], s1 = inv_std[1], s2 = inv_std[2];
+
+  for (int64_t q = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       q < nquads; q += (int64_t)gridDim.x * blockDim.x) {
+    const int y = (int)(q / Wq);
+    const int x0 = (int)(q % Wq) << 2;
+    const int cnt = min(4, W - x0);
+    const int64_t pix0 = (int64_t)y * W + x0;
+    const uint8_t* yrow = yplane + (int64_t)y * ystride + x0;
+
+    float rpix[4], gpix[4], bpix[4];
+    uint32_t y4 = (cnt == 4) ? load_u32_unaligned(yrow) : 0;
+    for (int k = 0; k < cnt; ++k) {
+      const int x = x0 + k;
+      const float fy = (cnt == 4) ? (float)((y4 >> (8 * k)) & 0xff)
+                                  : (float)yrow[k];
+      int rv, gv, bv;
+      if (nc == 1) {
+        rv = gv = bv = (int)fy;
+      } else {
+        float cb = fancy_sample(cbp, cstride, cw, chh, x, y, hmax, vmax)
+                       * (1.f / 16.f) - 128.f;
+        float cr = fancy_sample(crp, cstride, cw, chh, x, y, hmax, vmax)
+                       * (1.f / 16.f) - 128.f;
+        rv = __float2int_rn(fy + 1.40200f * cr);
+        gv = __float2int_rn(fy - 0.34414f * cb - 0.71414f * cr);
+        bv = __float2int_rn(fy + 1.77200f * cb);
+      }
+      rpix[k] = ((float)min(255, max(0, rv)) * scale - m0) * s0;
+      gpix[k] = ((float)min(255, max(0, gv)) * scale - m1) * s1;
+      bpix[k] = ((float)min(255, max(0, bv)) * scale - m2) * s2;
+    }
+    float* dr = dst + pix0;
+    float* dg = dst + plane + pix0;
+    float* db = dst + 2 * plane + pix0;
+    if (cnt == 4 && ((pix0 & 3) == 0)) {   // 16B-aligned float4 stores
+      *(float4*)dr = make_float4(rpix[0], rpix[1], rpix[2], rpix[3]);
+      *(float4*)dg = make_float4(gpix[0], gpix[1], gpix[2], gpix[3]);
+      *(float4*)db = make_float4(bpix[0], bpix[1], bpix[2], bpix[3]);
+    } else {
+      for (int k = 0; k < cnt; ++k) {
+        dr[k] = rpix[k];
+        dg[k] = gpix[k];
+        db[k] = bpix[k];
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // host launchers
 // ---------------------------------------------------------------------------
@@ -525,6 +603,54 @@ void jpeg_decode_batch(torch::Tensor data, py::dict meta, torch::Tensor coef,
                      stream, samples.data_ptr<uint8_t>(), g,
                      out.data_ptr<uint8_t>(), out_off.data_ptr<int64_t>(),
                      n_imgs);
+}
+
+// Same pipeline with the fused color+normalize epilogue: `out` is fp32
+// [n, 3, H, W]; out_off is in FLOAT elements.
+void jpeg_decode_fused_batch(torch::Tensor data, py::dict meta,
+                             torch::Tensor coef, torch::Tensor samples,
+                             torch::Tensor out, torch::Tensor out_off,
+                             torch::Tensor mean, torch::Tensor inv_std,
+                             double scale, torch::Tensor status) {
+  TORCH_CHECK(data.is_cuda() && coef.is_cuda() && samples.is_cuda() &&
+              out.is_cuda() && out.scalar_type() == torch::kFloat32);
+  TORCH_CHECK(mean.is_cuda() && inv_std.is_cuda() &&
+              mean.numel() == 3 && inv_std.numel() == 3);
+  JpegTables tabs = make_tables(meta);
+  JpegGeom g = make_geom(meta);
+  auto seg_img = meta["seg_img"].cast<torch::Tensor>();
+  auto seg_pos = meta["seg_pos"].cast<torch::Tensor>();
+  auto seg_end = meta["seg_end"].cast<torch::Tensor>();
+  auto seg_mcu0 = meta["seg_mcu0"].cast<torch::Tensor>();
+  auto seg_nmcu = meta["seg_nmcu"].cast<torch::Tensor>();
+  int n_segs = (int)seg_img.numel();
+  int n_imgs = (int)meta["width"].cast<torch::Tensor>().numel();
+  int64_t block_total = coef.numel() / 64;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+
+  hipLaunchKernelGGL(jpeg_huffman_kernel,
+                     dim3((n_segs + 63) / 64), dim3(64), 0, stream,
+                     data.data_ptr<uint8_t>(), tabs, g,
+                     seg_img.data_ptr<int32_t>(),
+                     seg_pos.data_ptr<int64_t>(),
+                     seg_end.data_ptr<int64_t>(),
+                     seg_mcu0.data_ptr<int32_t>(),
+                     seg_nmcu.data_ptr<int32_t>(), coef.data_ptr<float>(),
+                     status.data_ptr<int32_t>(), n_segs);
+
+  hipLaunchKernelGGL(jpeg_idct_kernel,
+                     dim3((unsigned)((block_total + 255) / 256)), dim3(256),
+                     0, stream, coef.data_ptr<float>(), g,
+                     samples.data_ptr<uint8_t>(), n_imgs, block_total);
+
+  int64_t avg_px = n_imgs > 0 ? out.numel() / (3 * n_imgs) : 0;
+  int64_t est_blocks = (avg_px / 4 + 255) / 256;
+  int grid_x = (int)std::min<int64_t>(std::max<int64_t>(est_blocks, 1), 2048);
+  hipLaunchKernelGGL(jpeg_color_norm_kernel, dim3(grid_x, n_imgs), dim3(256),
+                     0, stream, samples.data_ptr<uint8_t>(), g,
+                     out.data_ptr<float>(), out_off.data_ptr<int64_t>(),
+                     mean.data_ptr<float>(), inv_std.data_ptr<float>(),
+                     (float)scale, n_imgs);
 }
 
 }  // namespace psa

@@ -42,6 +42,76 @@ class TransformSpec(object):
         return isinstance(other, TransformSpec) and self.__dict__ == other.__dict__
 
 
+class FusedImageNormalize(object):
+    """The dominant training-input transform, declared so the GPU route can
+    FUSE it into the JPEG color kernel epilogue: uint8 HWC image ->
+    ``((u8/`` `scale_div` ``) - mean) / std`` as NCHW float32.
+
+    Use through :func:`fused_image_normalize`.  Works on every route:
+
+    * GPU route + jpeg column: the decoder writes normalized NCHW fp32
+      directly (no NHWC uint8 intermediate, no separate normalize kernel).
+    * any other route: ``__call__`` performs the same math on the
+      column-dict batch (torch on device, numpy on CPU) — the callable
+      detects an already-fused column (float32 NCHW) and passes it through.
+    """
+
+    def __init__(self, field, mean, std, scale_div=255.0):
+        self.field = field
+        self.mean = [float(m) for m in mean]
+        self.std = [float(s) for s in std]
+        self.scale_div = float(scale_div)
+
+    def __call__(self, columns):
+        import numpy as _np
+        img = columns.get(self.field)
+        if img is None:
+            return columns
+        out = dict(columns)
+        try:
+            import torch as _torch
+            is_torch = isinstance(img, _torch.Tensor)
+        except ImportError:
+            is_torch = False
+        if is_torch:
+            import torch as _torch
+            if img.dtype == _torch.float32 and img.dim() == 4 and \
+                    img.shape[1] == 3:
+                return columns  # already fused by the decoder
+            mean = _torch.tensor(self.mean, device=img.device)
+            std = _torch.tensor(self.std, device=img.device)
+            x = img.to(_torch.float32) / self.scale_div
+            x = (x - mean) / std
+            out[self.field] = x.permute(0, 3, 1, 2).contiguous()
+        else:
+            arr = _np.asarray(img)
+            if arr.dtype == _np.float32 and arr.ndim == 4 and \
+                    arr.shape[1] == 3:
+                return columns
+            single = arr.ndim == 3  # row path: one HWC image
+            if single:
+                arr = arr[None]
+            x = arr.astype(_np.float32) / self.scale_div
+            x = (x - _np.asarray(self.mean, dtype=_np.float32)) / \
+                _np.asarray(self.std, dtype=_np.float32)
+            x = _np.ascontiguousarray(x.transpose(0, 3, 1, 2))
+            out[self.field] = x[0] if single else x
+        return out
+
+
+def fused_image_normalize(field, mean, std, scale_div=255.0,
+                          extra_edit_fields=None):
+    """A TransformSpec normalizing ``field`` (uint8 HWC image) into NCHW
+    float32 — fused into the GPU JPEG decode when possible."""
+    import numpy as _np
+    func = FusedImageNormalize(field, mean, std, scale_div)
+    edits = [UnischemaField(field, _np.float32, (3, None, None), None,
+                            False)]
+    if extra_edit_fields:
+        edits.extend(extra_edit_fields)
+    return TransformSpec(func, edit_fields=edits)
+
+
 def _as_unischema_field(entry):
     if isinstance(entry, UnischemaField):
         return entry

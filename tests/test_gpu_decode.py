@@ -1157,3 +1157,45 @@ def test_decimal_column_gpu_matches_cpu_batch_route(ext, tmp_path):
     for k in gpu_prices:
         assert str(gpu_prices[k]) == str(cpu_prices[k])
         assert Decimal(str(gpu_prices[k])) == Decimal(k) / Decimal(7)
+
+
+def test_fused_jpeg_normalize_bit_exact(ext, tmp_path):
+    """fused_image_normalize through the GPU route must be BIT-EXACT with
+    the two-step path (decode to uint8 NHWC, then the standalone
+    normalize kernel): the fused epilogue rounds/clamps to u8 first."""
+    from petastorm_amd import TransformSpec, make_batch_reader
+    from petastorm_amd.transform import fused_image_normalize
+    from petastorm_amd.test_util.dataset_gen import create_imagenet_dataset
+    from petastorm_amd.unischema import UnischemaField
+    url = 'file://' + str(tmp_path / 'fuse')
+    create_imagenet_dataset(url, num_rows=24, rowgroup_size_mb=8)
+    mean, std = [0.485, 0.456, 0.406], [0.229, 0.224, 0.225]
+
+    ts_fused = fused_image_normalize('image', mean, std)
+    with make_batch_reader(url, device='cuda', shuffle_row_groups=False,
+                           transform_spec=ts_fused) as r:
+        fused = [(b.label.cpu().numpy(), b.image) for b in r]
+        assert r.diagnostics['cpu_assist_columns'] == []
+
+    e = ext
+    mean_t = torch.tensor(mean, device='cuda')
+    inv_t = 1.0 / torch.tensor(std, device='cuda')
+
+    def two_step(cols):
+        img = cols['image']
+        out = torch.empty(img.shape[0], 3, img.shape[1], img.shape[2],
+                          dtype=torch.float32, device=img.device)
+        e.nhwc_to_nchw_normalize(img, out, mean_t, inv_t, 1.0 / 255.0)
+        return {'image': out, 'label': cols['label']}
+
+    ts_two = TransformSpec(two_step, edit_fields=[
+        UnischemaField('image', np.float32, (3, 224, 224), None, False)])
+    with make_batch_reader(url, device='cuda', shuffle_row_groups=False,
+                           transform_spec=ts_two) as r:
+        two = [(b.label.cpu().numpy(), b.image) for b in r]
+
+    assert len(fused) == len(two)
+    for (lf, imf), (lt, imt) in zip(fused, two):
+        np.testing.assert_array_equal(lf, lt)
+        assert imf.shape == imt.shape and imf.dtype == imt.dtype
+        torch.testing.assert_close(imf, imt, rtol=0, atol=0)
