@@ -332,10 +332,11 @@ def _run_timed(args, step_fn, device, dist, world):
                 break
         _sync(device)
         per_batch = (time.perf_counter() - t0) / cal
-        # 1.25x headroom: steady-state batches run faster than the
-        # calibration window (caches warm, pipeline full)
+        # 1.5x headroom: steady-state batches run faster than the
+        # calibration window (caches warm, pipeline full; measured up to
+        # ~1.35x on the fused imagenet path)
         batches_per_step = max(
-            1, int(np.ceil(1.25 * args.min_region /
+            1, int(np.ceil(1.5 * args.min_region /
                            (args.steps * per_batch))))
         if dist is not None:
             # all ranks must agree or lock-step collectives skew
