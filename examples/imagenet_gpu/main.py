@@ -18,10 +18,9 @@ import tempfile
 import numpy as np
 import torch
 
-from petastorm_amd import TransformSpec, make_batch_reader, ops
+from petastorm_amd import make_batch_reader, ops
 from petastorm_amd.pytorch import BatchedDataLoader
 from petastorm_amd.test_util.dataset_gen import create_imagenet_dataset
-from petastorm_amd.unischema import UnischemaField
 
 
 def main(url=None, rows=512, batch_size=64, steps=20):
@@ -37,19 +36,12 @@ def main(url=None, rows=512, batch_size=64, steps=20):
                                     rowgroup_size_mb=16)
         url = 'file://' + url
 
-    ext = ops.ext()
-    mean = torch.tensor([0.485, 0.456, 0.406], device='cuda')
-    inv_std = 1.0 / torch.tensor([0.229, 0.224, 0.225], device='cuda')
-
-    def transform(cols):
-        img = cols['image']
-        out = torch.empty(img.shape[0], 3, 224, 224, dtype=torch.float32,
-                          device=img.device)
-        ext.nhwc_to_nchw_normalize(img, out, mean, inv_std, 1.0 / 255.0)
-        return {'image': out, 'label': cols['label']}
-
-    ts = TransformSpec(transform, edit_fields=[
-        UnischemaField('image', np.float32, (3, 224, 224), None, False)])
+    ops.ext()  # loud failure if the native extension is missing
+    # fused into the jpeg color kernel epilogue: decode emits normalized
+    # NCHW float32 directly (no NHWC uint8 intermediate)
+    from petastorm_amd.transform import fused_image_normalize
+    ts = fused_image_normalize('image', mean=[0.485, 0.456, 0.406],
+                               std=[0.229, 0.224, 0.225])
 
     # a tiny conv net standing in for the real model
     model = torch.nn.Sequential(
