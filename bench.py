@@ -277,7 +277,9 @@ def bench_ngram(args, rank, world, device, dist):
         seed=5, predicate=pred,
         cur_shard=rank if world > 1 else None,
         shard_count=world if world > 1 else None,
-        gpu_options=dict(cache_type='hbm', cache_size_limit=64 << 30))
+        gpu_options=dict(cache_type='hbm', cache_size_limit=64 << 30,
+                         decode_streams=int(os.environ.get(
+                             'PSA_DECODE_STREAMS', '8'))))
     loader = BatchedDataLoader(reader, batch_size=args.batch_size * 4,
                                shuffling_queue_capacity=args.batch_size * 16,
                                seed=3)
