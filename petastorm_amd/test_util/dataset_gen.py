@@ -200,3 +200,62 @@ def create_sequence_dataset(url, num_rows=200, rowgroup_size_mb=4, seed=0,
                 'tokens': rng.randint(0, 50000, 1024).astype(np.int32),
                 'source': np.int32(i % 4),
             })
+
+
+def create_rich_scalar_dataset(url, num_rows=200, rowgroup_size=50):
+    """Non-petastorm store with the reference's full scalar-type surface
+    (reference tests/test_common.py:161-245): date, timestamp, strings,
+    float64, fixed-size int list, nested struct."""
+    import datetime
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd.fs_utils import get_filesystem_and_path_or_paths
+    rows = []
+    for i in range(num_rows):
+        rows.append({
+            'id': np.int32(i),
+            'id_div_700': np.int32(i // 700),
+            'datetime': datetime.date(2019, 1, 2),
+            'timestamp': datetime.datetime(2005, 2, 25, 3, 30),
+            'string': 'hello_%d' % i,
+            'string2': 'world_%d' % i,
+            'float64': np.float64(i) * 0.66,
+            'int_fixed_size_list': list(range(1 + i, 10 + i)),
+            'nested_struct': {'nested_int': i},
+        })
+    table = pa.table({
+        'id': pa.array([r['id'] for r in rows], pa.int32()),
+        'id_div_700': pa.array([r['id_div_700'] for r in rows], pa.int32()),
+        'datetime': pa.array([r['datetime'] for r in rows], pa.date32()),
+        'timestamp': pa.array([r['timestamp'] for r in rows],
+                              pa.timestamp('us')),
+        'string': pa.array([r['string'] for r in rows], pa.string()),
+        'string2': pa.array([r['string2'] for r in rows], pa.string()),
+        'float64': pa.array([r['float64'] for r in rows], pa.float64()),
+        'int_fixed_size_list': pa.array(
+            [r['int_fixed_size_list'] for r in rows],
+            pa.list_(pa.int32(), 9)),
+        'nested_struct': pa.array(
+            [r['nested_struct'] for r in rows],
+            pa.struct([('nested_int', pa.int32())])),
+    })
+    fs, path = get_filesystem_and_path_or_paths(url)
+    fs.makedirs(path, exist_ok=True)
+    pq.write_table(table, path + '/data.parquet',
+                   row_group_size=rowgroup_size, compression='none')
+    return rows
+
+
+def create_many_columns_dataset(url, num_rows=25, num_columns=1000):
+    """1000-int32-column plain store (reference tests/test_common.py:
+    248-294)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd.fs_utils import get_filesystem_and_path_or_paths
+    cols = {'col_{}'.format(c): np.arange(num_rows, dtype=np.int32) + c
+            for c in range(num_columns)}
+    table = pa.table(cols)
+    fs, path = get_filesystem_and_path_or_paths(url)
+    fs.makedirs(path, exist_ok=True)
+    pq.write_table(table, path + '/data.parquet', row_group_size=10)
+    return cols

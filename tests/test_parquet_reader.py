@@ -128,3 +128,60 @@ def test_filters_dnf_or(scalar_dataset):
         ids = _collect(list(r))['id']
         assert r.diagnostics['items_ventilated'] == 2
     assert set(ids) == set(range(0, 100)) | set(range(400, 500))
+
+
+# ---------------------------------------------------------------------------
+# full scalar-type surface + many-columns stores (reference
+# tests/test_common.py:161-294, tests/test_parquet_reader.py)
+# ---------------------------------------------------------------------------
+
+def test_rich_scalar_store_types(tmp_path):
+    """date/timestamp/strings/float64/fixed-size list/nested struct all
+    come back with the reference's observable type mapping."""
+    import datetime
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_rich_scalar_dataset
+    url = 'file://' + str(tmp_path / 'rich')
+    rows = create_rich_scalar_dataset(url, num_rows=60, rowgroup_size=20)
+    got = {}
+    with make_batch_reader(url, shuffle_row_groups=False) as r:
+        for b in r:
+            ids = np.asarray(b.id)
+            for i, rid in enumerate(ids):
+                got[int(rid)] = {f: np.asarray(getattr(b, f))[i]
+                                 for f in b._fields}
+    assert len(got) == 60
+    for src in rows:
+        row = got[int(src['id'])]
+        assert row['string'] == src['string']
+        assert row['float64'] == src['float64']
+        np.testing.assert_array_equal(row['int_fixed_size_list'],
+                                      src['int_fixed_size_list'])
+        # datetime.date -> datetime64[D]-compatible value
+        assert np.datetime64(row['datetime'], 'D') == \
+            np.datetime64(src['datetime'], 'D')
+        assert np.datetime64(row['timestamp'], 'us') == \
+            np.datetime64(src['timestamp'], 'us')
+        # nested structs are omitted from the inferred schema with a
+        # warning — the reference does the same (unischema.py:303
+        # omit_unsupported_fields=True by default)
+        assert 'nested_struct' not in row
+
+
+def test_many_columns_store(tmp_path):
+    """1000 int32 columns (reference many_columns_non_petastorm_dataset):
+    full read and a 3-column subset."""
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_many_columns_dataset
+    url = 'file://' + str(tmp_path / 'wide')
+    cols = create_many_columns_dataset(url, num_rows=25, num_columns=1000)
+    with make_batch_reader(url, shuffle_row_groups=False) as r:
+        b = next(iter(r))
+        assert len(b._fields) == 1000
+        np.testing.assert_array_equal(np.sort(np.asarray(b.col_0)),
+                                      cols['col_0'][:len(b.col_0)])
+    with make_batch_reader(url, shuffle_row_groups=False,
+                           schema_fields=['col_1', 'col_42',
+                                          'col_999']) as r:
+        b = next(iter(r))
+        assert sorted(b._fields) == ['col_1', 'col_42', 'col_999']
