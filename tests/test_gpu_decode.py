@@ -1199,3 +1199,48 @@ def test_fused_jpeg_normalize_bit_exact(ext, tmp_path):
         np.testing.assert_array_equal(lf, lt)
         assert imf.shape == imt.shape and imf.dtype == imt.dtype
         torch.testing.assert_close(imf, imt, rtol=0, atol=0)
+
+
+def test_rich_scalar_store_gpu_route(ext, tmp_path):
+    """Full scalar-type surface through the GPU route: ints/floats/strings
+    native; date/timestamp values must MATCH the CPU route's mapping
+    regardless of which side decodes them."""
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_rich_scalar_dataset
+    url = 'file://' + str(tmp_path / 'richg')
+    rows = create_rich_scalar_dataset(url, num_rows=60, rowgroup_size=20)
+
+    def read_all(device):
+        got = {}
+        kwargs = {'device': device} if device else {}
+        with make_batch_reader(url, shuffle_row_groups=False,
+                               **kwargs) as r:
+            for b in r:
+                ids = np.asarray(b.id.cpu() if hasattr(b.id, 'cpu')
+                                 else b.id)
+                for i, rid in enumerate(ids):
+                    row = {}
+                    for f in b._fields:
+                        v = getattr(b, f)
+                        if hasattr(v, 'cpu'):
+                            v = v.cpu().numpy()
+                        row[f] = np.asarray(v)[i]
+                    got[int(rid)] = row
+            assist = list(r.diagnostics.get('cpu_assist_columns', []))
+        return got, assist
+
+    gpu, assist = read_all('cuda')
+    cpu, _ = read_all(None)
+    # scalar ints/floats/strings must be native on the GPU route
+    for native_col in ('id', 'id_div_700', 'float64', 'string', 'string2'):
+        assert native_col not in assist, (native_col, assist)
+    assert len(gpu) == len(rows)
+    for rid, crow in cpu.items():
+        grow = gpu[rid]
+        for f, cv in crow.items():
+            gv = grow[f]
+            if isinstance(cv, np.ndarray) or isinstance(gv, np.ndarray):
+                np.testing.assert_array_equal(np.asarray(gv),
+                                              np.asarray(cv), err_msg=f)
+            else:
+                assert gv == cv, (f, gv, cv)
