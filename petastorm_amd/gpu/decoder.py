@@ -179,7 +179,12 @@ class GpuRowGroupDecoder(object):
         md = file_metadata.row_group(rg)
         name_to_idx = {}
         for ci in range(md.num_columns):
-            name_to_idx[md.column(ci).path_in_schema] = ci
+            col_path = md.column(ci).path_in_schema
+            name_to_idx[col_path] = ci
+            # list columns carry leaf paths ('col.list.element'); map the
+            # ROOT name to its (first) leaf chunk so the request resolves —
+            # repeated columns then take the CPU-assist route in decode()
+            name_to_idx.setdefault(col_path.split('.')[0], ci)
         chunks = []
         lo, hi = None, 0
         for name in columns:
@@ -194,7 +199,8 @@ class GpuRowGroupDecoder(object):
             chunks.append((name, ci, start, col.total_compressed_size,
                            col.physical_type, col.compression,
                            col.num_values,
-                           parquet_schema.column(ci).max_definition_level))
+                           parquet_schema.column(ci).max_definition_level,
+                           parquet_schema.column(ci).max_repetition_level))
         # merge the requested chunk byte ranges into extents (gaps below 256
         # KiB are read through rather than seeking) so unrequested columns
         # between them are not read or uploaded
@@ -232,8 +238,8 @@ class GpuRowGroupDecoder(object):
             'chunks': [
                 dict(name=n, col_index=ci, offset=_rebase(s), length=ln,
                      physical=pt, compression=comp, num_values=nv,
-                     max_def=mdl)
-                for (n, ci, s, ln, pt, comp, nv, mdl) in chunks],
+                     max_def=mdl, max_rep=mrl)
+                for (n, ci, s, ln, pt, comp, nv, mdl, mrl) in chunks],
         }
         return host, chunk_meta
 
@@ -406,6 +412,12 @@ class GpuRowGroupDecoder(object):
             comp = ch['compression']
             if comp not in ('UNCOMPRESSED', 'SNAPPY', 'GZIP',
                             'LZ4', 'ZSTD'):
+                out[name] = self._cpu_assist_marker(name)
+                continue
+            if ch.get('max_rep', 0) > 0:
+                # repeated (list) columns need repetition-level assembly —
+                # CPU assist (reference delegates these to Arrow's nested
+                # reassembly too)
                 out[name] = self._cpu_assist_marker(name)
                 continue
             col = self._decode_chunk(ext, dev, dbuf, host_buf, ch,

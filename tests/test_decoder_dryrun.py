@@ -357,3 +357,18 @@ def test_dryrun_string_column_host_visible_exact(stub_decoder, tmp_path):
     assert isinstance(col, ByteArrayColumn) and col.host_buf is not None
     got = dec.decode_string_column(col, sch.fields['name'])
     assert got.tolist() == vals
+
+
+def test_dryrun_list_column_takes_assist_not_crash(stub_decoder, tmp_path):
+    """Repeated (list) columns must route to CPU assist gracefully — the
+    leaf path ('col.list.element') used to KeyError in
+    read_rowgroup_bytes."""
+    from petastorm_amd.test_util.dataset_gen import create_rich_scalar_dataset
+    dec, stub = stub_decoder
+    d = str(tmp_path / 'rich')
+    create_rich_scalar_dataset('file://' + d, num_rows=20, rowgroup_size=10)
+    out, sch = _decode_all(dec, 'file://' + d,
+                           ['id', 'int_fixed_size_list', 'string'])
+    assert out['id'] is not None
+    assert out['int_fixed_size_list'] is None  # assist marker
+    assert 'int_fixed_size_list' in dec.cpu_assist_columns

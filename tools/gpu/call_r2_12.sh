@@ -1,0 +1,4 @@
+set -x
+cd /root/repo
+mkdir -p gpurun_out
+python -m pytest tests -m gpu -q 2>&1 | tail -4
