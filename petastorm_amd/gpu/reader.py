@@ -478,6 +478,26 @@ class GpuBatchReader(object):
         return (piece, cols, pmeta, self._decoder.take_pending())
 
     # ------------------------------------------------------------------
+    def _datetime_unit(self, path, name):
+        """numpy datetime64 unit of a DATE/TIMESTAMP column ('D'/'ms'/'us'/
+        'ns'), from the Parquet logical type; None for non-datetime."""
+        _, pschema = self._metadata(path)
+        for i in range(len(pschema)):
+            col = pschema.column(i)
+            if col.path.split('.')[0] != name:
+                continue
+            lt = str(col.logical_type).lower()
+            if lt.startswith('date'):
+                return 'D'
+            if 'timestamp' in lt:
+                for unit in ('nanoseconds', 'microseconds', 'milliseconds'):
+                    if unit in lt:
+                        return {'nanoseconds': 'ns', 'microseconds': 'us',
+                                'milliseconds': 'ms'}[unit]
+                return 'us'
+            return None
+        return None
+
     @staticmethod
     def _is_string_field(field):
         """Scalar str/bytes field (incl. ScalarCodec'd and schema-inferred
@@ -540,6 +560,15 @@ class GpuBatchReader(object):
                 else:
                     columns[name] = decoded
             else:
+                if field is not None and field.numpy_dtype is np.datetime64 \
+                        and isinstance(col, torch.Tensor):
+                    # DATE/TIMESTAMP: physical ints decoded on device;
+                    # materialize numpy datetime64 at the boundary (torch
+                    # has no datetime dtype — CPU-route type parity,
+                    # reference unischema.py:467-502 mapping)
+                    unit = self._datetime_unit(piece.path, name) or 'us'
+                    col = col.cpu().numpy().astype(
+                        'datetime64[{}]'.format(unit))
                 columns[name] = col
         if assist:
             columns.update(self._cpu_assist(piece, assist))
