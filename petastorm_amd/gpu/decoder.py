@@ -38,6 +38,8 @@ from petastorm_amd.codecs import (CompressedImageCodec, NdarrayCodec)
 _ENC_PLAIN = 0
 _ENC_PLAIN_DICT = 2
 _ENC_RLE = 3
+_ENC_DELTA_BINARY = 5
+_ENC_DELTA_LENGTH_BA = 6
 _ENC_RLE_DICT = 8
 _PAGE_DATA_V1 = 0
 _PAGE_DICT = 2
@@ -613,7 +615,60 @@ class GpuRowGroupDecoder(object):
                                          uncomp_size if snappy else comp_size,
                                          val_start, val_end, page_nval,
                                          nonnull_per_page, valid, ch)
+        all_valid = nonnull_per_page is None or \
+            bool((nonnull_per_page == page_nval).all())
+        if data_enc == _ENC_DELTA_BINARY and phys in ('INT32', 'INT64') \
+                and all_valid:
+            return self._delta_fixed(ext, dev, page_buf, val_start, val_end,
+                                     page_nval, phys, ch)
+        if data_enc == _ENC_DELTA_LENGTH_BA and phys == 'BYTE_ARRAY':
+            return self._delta_length_byte_array(
+                ext, dev, page_buf, val_start, val_end, page_nval,
+                nonnull_per_page, valid, ch)
         return self._cpu_assist_marker(ch['name'])
+
+    def _delta_fixed(self, ext, dev, page_buf, val_start, val_end,
+                     page_nval, phys, ch):
+        """DELTA_BINARY_PACKED int columns (wave-per-page shfl prefix
+        scan kernel); what pyarrow/Spark v2 writers emit for ints."""
+        dtype, esize = _PHYS_TO_TORCH[phys]
+        counts = page_nval.astype(np.int32)
+        total = int(counts.sum())
+        out_off = np.zeros(len(counts), dtype=np.int64)
+        out_off[1:] = np.cumsum(counts)[:-1]
+        out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        status = self._status(len(counts))
+        ext.delta_binary_packed_batch(
+            page_buf, self._up(val_start), self._up(val_end),
+            self._up(counts), self._up(out_off), out, esize, status)
+        self._check(status, 'delta:' + ch['name'])
+        return out[:total * esize].view(dtype)
+
+    def _delta_length_byte_array(self, ext, dev, page_buf, val_start,
+                                 val_end, page_nval, nonnull_per_page,
+                                 valid, ch):
+        """DELTA_LENGTH_BYTE_ARRAY -> per-value (offset, length) tables
+        into the page buffer; flows into the standard ByteArrayColumn
+        consumers (strings/ndarray codecs)."""
+        counts = (nonnull_per_page if nonnull_per_page is not None
+                  else page_nval).astype(np.int64)
+        total = int(counts.sum())
+        out_idx = np.zeros(len(counts), dtype=np.int64)
+        out_idx[1:] = np.cumsum(counts)[:-1]
+        val_off = torch.empty(total, dtype=torch.int64, device=dev)
+        val_len = torch.empty(total, dtype=torch.int32, device=dev)
+        status = self._status(len(counts))
+        ext.delta_length_byte_array_batch(
+            page_buf, self._up(val_start), self._up(val_end),
+            self._up(counts.astype(np.int32)), self._up(out_idx), val_off,
+            val_len, status)
+        self._check(status, 'deltaba:' + ch['name'])
+        col = ByteArrayColumn(page_buf, val_off, val_len, None, None, total)
+        if valid is not None and nonnull_per_page is not None and \
+                not bool((nonnull_per_page == page_nval).all()):
+            col.valid = valid
+        return col
 
     # ------------------------------------------------------------------
     _FILL_PATTERNS = {

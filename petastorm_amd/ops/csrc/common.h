@@ -32,6 +32,12 @@ __device__ __forceinline__ T wave_bcast(T v) {
   return __shfl(v, 0, PSA_WAVE);
 }
 
+// Broadcast from an arbitrary source lane.
+template <typename T>
+__device__ __forceinline__ T wave_bcast_from(T v, int src_lane) {
+  return __shfl(v, src_lane, PSA_WAVE);
+}
+
 // Unaligned little-endian loads from a byte stream (global memory).
 __device__ __forceinline__ uint32_t load_u32_unaligned(const uint8_t* p) {
   return (uint32_t)p[0] | ((uint32_t)p[1] << 8) | ((uint32_t)p[2] << 16) |

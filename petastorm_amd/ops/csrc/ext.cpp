@@ -88,6 +88,17 @@ py::dict byte_array_host_offsets(torch::Tensor buf, torch::Tensor val_start,
 void jpeg_decode_batch(torch::Tensor data, py::dict meta, torch::Tensor coef,
                        torch::Tensor samples, torch::Tensor out,
                        torch::Tensor out_off, torch::Tensor status);
+void delta_binary_packed_batch(torch::Tensor page_buf, torch::Tensor start,
+                               torch::Tensor end, torch::Tensor n_values,
+                               torch::Tensor out_off, torch::Tensor out,
+                               int64_t esize, torch::Tensor status);
+void delta_length_byte_array_batch(torch::Tensor page_buf,
+                                   torch::Tensor start, torch::Tensor end,
+                                   torch::Tensor n_values,
+                                   torch::Tensor out_idx,
+                                   torch::Tensor val_off,
+                                   torch::Tensor val_len,
+                                   torch::Tensor status);
 void jpeg_decode_fused_batch(torch::Tensor data, py::dict meta,
                              torch::Tensor coef, torch::Tensor samples,
                              torch::Tensor out, torch::Tensor out_off,
@@ -136,6 +147,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Host-side JPEG header/segment parse");
   m.def("jpeg_decode_batch", &psa::jpeg_decode_batch,
         "GPU baseline JPEG decode (huffman/idct/color kernels)");
+  m.def("delta_binary_packed_batch", &psa::delta_binary_packed_batch,
+        "DELTA_BINARY_PACKED page decode (wave-per-page, shfl prefix scan)");
+  m.def("delta_length_byte_array_batch",
+        &psa::delta_length_byte_array_batch,
+        "DELTA_LENGTH_BYTE_ARRAY page -> per-value (offset, length)");
   m.def("jpeg_decode_fused_batch", &psa::jpeg_decode_fused_batch,
         "JPEG decode with fused YCbCr->RGB + normalize + NCHW fp32 "
         "epilogue (skips the NHWC uint8 intermediate)");

@@ -521,4 +521,274 @@ void plain_fixed_decode_batch(torch::Tensor page_buf,
                      status.data_ptr<int32_t>(), n);
 }
 
+
+// ---------------------------------------------------------------------------
+// DELTA_BINARY_PACKED (Parquet spec encodings.md): header = <block_size>
+// <miniblocks_per_block> <total_count> <first_value zigzag>, then blocks of
+// <min_delta zigzag> <1 bitwidth byte per miniblock> <bit-packed deltas>.
+// value[i+1] = value[i] + min_delta + delta[i].
+//
+// One wave per page: lane 0 walks the varints; the wave cooperatively
+// extracts 64 deltas at a time and turns them into values with a shfl
+// inclusive scan + running carry — the CDNA4-idiomatic form of the
+// sequential prefix dependency.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint64_t read_varint_u64(const uint8_t* p,
+                                                    int64_t& pos,
+                                                    int64_t end) {
+  uint64_t result = 0;
+  int shift = 0;
+  while (pos < end && shift < 70) {
+    uint8_t b = p[pos++];
+    result |= (uint64_t)(b & 0x7f) << shift;
+    if (!(b & 0x80)) break;
+    shift += 7;
+  }
+  return result;
+}
+
+__device__ __forceinline__ int64_t zigzag64(uint64_t u) {
+  return (int64_t)(u >> 1) ^ -(int64_t)(u & 1);
+}
+
+__device__ __forceinline__ uint64_t extract_bits64(const uint8_t* base,
+                                                   int64_t bit_off, int bw) {
+  // up to 64 bits spanning <= 9 bytes
+  int64_t byte = bit_off >> 3;
+  int shift = (int)(bit_off & 7);
+  uint64_t lo = 0;
+  for (int i = 0; i < 8; ++i) lo |= (uint64_t)base[byte + i] << (8 * i);
+  uint64_t v = lo >> shift;
+  if (shift) v |= (uint64_t)base[byte + 8] << (64 - shift);
+  if (bw < 64) v &= (~0ull) >> (64 - bw);
+  return v;
+}
+
+// wave-wide inclusive prefix sum (int64), Hillis-Steele over 64 lanes
+__device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t v) {
+  const int lane = lane_id();
+  for (int d = 1; d < PSA_WAVE; d <<= 1) {
+    int64_t up = __shfl_up((long long)v, d, PSA_WAVE);
+    if (lane >= d) v += up;
+  }
+  return v;
+}
+
+__global__ void delta_binary_packed_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
+    const int64_t* __restrict__ end, const int32_t* __restrict__ n_values,
+    const int64_t* __restrict__ out_off, uint8_t* __restrict__ out,
+    int32_t esize, int32_t* __restrict__ status, int n_pages) {
+  const int waves_per_block = blockDim.x / PSA_WAVE;
+  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  if (page >= n_pages) return;
+  const int lane = lane_id();
+  const uint8_t* base = data;
+  int64_t pos = start[page];
+  const int64_t pend = end[page];
+  const int32_t want = n_values[page];
+
+  // header (lane 0 parses, broadcast)
+  uint64_t block_size = 0, mbs_per_block = 0, total = 0;
+  int64_t first = 0;
+  if (lane == 0) {
+    block_size = read_varint_u64(base, pos, pend);
+    mbs_per_block = read_varint_u64(base, pos, pend);
+    total = read_varint_u64(base, pos, pend);
+    first = zigzag64(read_varint_u64(base, pos, pend));
+  }
+  block_size = wave_bcast((unsigned long long)block_size);
+  mbs_per_block = wave_bcast((unsigned long long)mbs_per_block);
+  total = wave_bcast((unsigned long long)total);
+  first = wave_bcast((long long)first);
+  pos = wave_bcast((long long)pos);
+  if (block_size == 0 || mbs_per_block == 0 ||
+      block_size % (mbs_per_block * 8) != 0) {
+    if (lane == 0) status[page] = 40;
+    return;
+  }
+  const int vpm = (int)(block_size / mbs_per_block);  // values/miniblock
+
+  uint8_t* dst = out + out_off[page] * esize;
+  auto emit = [&](int64_t idx, int64_t v) {
+    if (idx >= want) return;
+    if (esize == 8) {
+      uint8_t* d = dst + idx * 8;
+      uint64_t u = (uint64_t)v;
+      for (int i = 0; i < 8; ++i) d[i] = (uint8_t)(u >> (8 * i));
+    } else {
+      uint8_t* d = dst + idx * 4;
+      uint32_t u = (uint32_t)(int32_t)v;
+      for (int i = 0; i < 4; ++i) d[i] = (uint8_t)(u >> (8 * i));
+    }
+  };
+  if (lane == 0) emit(0, first);
+  int64_t produced = 1;          // values written (includes first)
+  int64_t prev = first;          // last emitted value
+
+  while (produced < want && produced < (int64_t)total) {
+    // block header
+    int64_t min_delta = 0;
+    if (lane == 0) min_delta = zigzag64(read_varint_u64(base, pos, pend));
+    min_delta = wave_bcast((long long)min_delta);
+    pos = wave_bcast((long long)pos);
+    int64_t bw_pos = pos;               // one bitwidth byte per miniblock
+    pos += mbs_per_block;
+    if (pos > pend) { if (lane == 0) status[page] = 41; return; }
+    for (uint64_t mb = 0; mb < mbs_per_block; ++mb) {
+      int bw = base[bw_pos + mb];
+      if (bw > 64) { if (lane == 0) status[page] = 42; return; }
+      const int64_t mb_bits = (int64_t)vpm * bw;
+      if (produced >= want || produced >= (int64_t)total) {
+        pos += (mb_bits + 7) >> 3;      // skip remaining miniblocks
+        continue;
+      }
+      // cooperative unpack + scan, 64 deltas per pass
+      for (int off = 0; off < vpm; off += PSA_WAVE) {
+        int64_t my_delta = 0;
+        int j = off + lane;
+        if (j < vpm && bw > 0)
+          my_delta = (int64_t)extract_bits64(base + pos, (int64_t)j * bw,
+                                             bw);
+        int64_t step = my_delta + min_delta;
+        int64_t incl = wave_incl_scan_i64(step);
+        int64_t v = prev + incl;
+        if (j < vpm) emit(produced + j, v);
+        // carry: value of the last lane of this pass
+        prev = wave_bcast_from(v, PSA_WAVE - 1);
+      }
+      produced += vpm;
+      pos += (mb_bits + 7) >> 3;
+      if (pos > pend + 8) { if (lane == 0) status[page] = 43; return; }
+    }
+  }
+}
+
+void delta_binary_packed_batch(torch::Tensor page_buf, torch::Tensor start,
+                               torch::Tensor end, torch::Tensor n_values,
+                               torch::Tensor out_off, torch::Tensor out,
+                               int64_t esize, torch::Tensor status) {
+  int n = (int)start.numel();
+  if (!n) return;
+  const int WPB = 4;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(delta_binary_packed_kernel, dim3((n + WPB - 1) / WPB),
+                     dim3(WPB * PSA_WAVE), 0, stream,
+                     page_buf.data_ptr<uint8_t>(), start.data_ptr<int64_t>(),
+                     end.data_ptr<int64_t>(), n_values.data_ptr<int32_t>(),
+                     out_off.data_ptr<int64_t>(), out.data_ptr<uint8_t>(),
+                     (int32_t)esize, status.data_ptr<int32_t>(), n);
+}
+
+// DELTA_LENGTH_BYTE_ARRAY: a DELTA_BINARY_PACKED int32 length block, then
+// the concatenated value bytes.  Emits per-value (offset, length) tables
+// pointing into the page (same contract as byte_array_offsets_batch).
+__global__ void delta_length_byte_array_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
+    const int64_t* __restrict__ end, const int32_t* __restrict__ n_values,
+    const int64_t* __restrict__ out_idx,
+    int64_t* __restrict__ val_off, int32_t* __restrict__ val_len,
+    int32_t* __restrict__ status, int n_pages) {
+  const int waves_per_block = blockDim.x / PSA_WAVE;
+  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  if (page >= n_pages) return;
+  const int lane = lane_id();
+  const uint8_t* base = data;
+  int64_t pos = start[page];
+  const int64_t pend = end[page];
+  const int32_t want = n_values[page];
+
+  uint64_t block_size = 0, mbs_per_block = 0, total = 0;
+  int64_t first = 0;
+  if (lane == 0) {
+    block_size = read_varint_u64(base, pos, pend);
+    mbs_per_block = read_varint_u64(base, pos, pend);
+    total = read_varint_u64(base, pos, pend);
+    first = zigzag64(read_varint_u64(base, pos, pend));
+  }
+  block_size = wave_bcast((unsigned long long)block_size);
+  mbs_per_block = wave_bcast((unsigned long long)mbs_per_block);
+  total = wave_bcast((unsigned long long)total);
+  first = wave_bcast((long long)first);
+  pos = wave_bcast((long long)pos);
+  if (block_size == 0 || mbs_per_block == 0 ||
+      block_size % (mbs_per_block * 8) != 0) {
+    if (lane == 0) status[page] = 45;
+    return;
+  }
+  const int vpm = (int)(block_size / mbs_per_block);
+
+  int64_t* po = val_off + out_idx[page];
+  int32_t* pl = val_len + out_idx[page];
+  // first pass: lengths -> val_len, and a running byte total
+  if (lane == 0 && want > 0) pl[0] = (int32_t)first;
+  int64_t produced = 1;
+  int64_t prev = first;
+  while (produced < want && produced < (int64_t)total) {
+    int64_t min_delta = 0;
+    if (lane == 0) min_delta = zigzag64(read_varint_u64(base, pos, pend));
+    min_delta = wave_bcast((long long)min_delta);
+    pos = wave_bcast((long long)pos);
+    int64_t bw_pos = pos;
+    pos += mbs_per_block;
+    if (pos > pend) { if (lane == 0) status[page] = 46; return; }
+    for (uint64_t mb = 0; mb < mbs_per_block; ++mb) {
+      int bw = base[bw_pos + mb];
+      if (bw > 64) { if (lane == 0) status[page] = 47; return; }
+      const int64_t mb_bits = (int64_t)vpm * bw;
+      if (produced >= want || produced >= (int64_t)total) {
+        pos += (mb_bits + 7) >> 3;
+        continue;
+      }
+      for (int off = 0; off < vpm; off += PSA_WAVE) {
+        int64_t my_delta = 0;
+        int j = off + lane;
+        if (j < vpm && bw > 0)
+          my_delta = (int64_t)extract_bits64(base + pos, (int64_t)j * bw,
+                                             bw);
+        int64_t incl = wave_incl_scan_i64(my_delta + min_delta);
+        int64_t v = prev + incl;
+        if (j < vpm && produced + j < want) pl[produced + j] = (int32_t)v;
+        prev = wave_bcast_from(v, PSA_WAVE - 1);
+      }
+      produced += vpm;
+      pos += (mb_bits + 7) >> 3;
+      if (pos > pend + 8) { if (lane == 0) status[page] = 48; return; }
+    }
+  }
+  // second pass: offsets = pos (bytes base) + exclusive scan of lengths.
+  // lane 0 walks sequentially (tiny: one add per value).
+  if (lane == 0) {
+    int64_t byte_pos = pos;
+    for (int32_t i = 0; i < want; ++i) {
+      po[i] = byte_pos;
+      byte_pos += pl[i];
+    }
+    if (byte_pos > pend) status[page] = 49;
+  }
+}
+
+void delta_length_byte_array_batch(torch::Tensor page_buf,
+                                   torch::Tensor start, torch::Tensor end,
+                                   torch::Tensor n_values,
+                                   torch::Tensor out_idx,
+                                   torch::Tensor val_off,
+                                   torch::Tensor val_len,
+                                   torch::Tensor status) {
+  int n = (int)start.numel();
+  if (!n) return;
+  const int WPB = 4;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(delta_length_byte_array_kernel,
+                     dim3((n + WPB - 1) / WPB), dim3(WPB * PSA_WAVE), 0,
+                     stream, page_buf.data_ptr<uint8_t>(),
+                     start.data_ptr<int64_t>(), end.data_ptr<int64_t>(),
+                     n_values.data_ptr<int32_t>(),
+                     out_idx.data_ptr<int64_t>(),
+                     val_off.data_ptr<int64_t>(),
+                     val_len.data_ptr<int32_t>(),
+                     status.data_ptr<int32_t>(), n);
+}
+
 }  // namespace psa

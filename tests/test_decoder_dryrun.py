@@ -27,7 +27,9 @@ class _StubExt(object):
                 'npy_payload_offsets', 'plain_fixed_decode_batch',
                 'nhwc_to_nchw_normalize', 'jpeg_decode_batch',
                 'lz4_decompress_batch',
-                'inflate_batch', 'png_unfilter_batch', 'bswap16'}
+                'inflate_batch', 'png_unfilter_batch', 'bswap16',
+                'delta_binary_packed_batch',
+                'delta_length_byte_array_batch'}
 
     def __init__(self):
         self._real = ops.ext()
@@ -372,3 +374,29 @@ def test_dryrun_list_column_takes_assist_not_crash(stub_decoder, tmp_path):
     assert out['id'] is not None
     assert out['int_fixed_size_list'] is None  # assist marker
     assert 'int_fixed_size_list' in dec.cpu_assist_columns
+
+
+def test_dryrun_delta_encodings(stub_decoder, tmp_path):
+    """DELTA_BINARY_PACKED / DELTA_LENGTH_BYTE_ARRAY orchestration
+    (kernels stubbed): shapes and kernel selection."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    dec, stub = stub_decoder
+    d = tmp_path / 'delta'
+    d.mkdir()
+    n = 400
+    table = pa.Table.from_arrays(
+        [pa.array(np.arange(n, dtype=np.int64) * 3 - 100),
+         pa.array(['v%d' % i for i in range(n)], pa.string())],
+        schema=pa.schema([pa.field('i64', pa.int64(), nullable=False),
+                          pa.field('s', pa.string(), nullable=False)]))
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=200,
+                   use_dictionary=False, compression='none',
+                   column_encoding={'i64': 'DELTA_BINARY_PACKED',
+                                    's': 'DELTA_LENGTH_BYTE_ARRAY'})
+    out, sch = _decode_all(dec, 'file://' + str(d), ['i64', 's'])
+    assert out['i64'] is not None and out['i64'].shape == (200,)
+    assert out['s'] is not None and out['s'].n == 200
+    assert 'delta_binary_packed_batch' in stub.calls
+    assert 'delta_length_byte_array_batch' in stub.calls
+    assert not dec.cpu_assist_columns
