@@ -651,12 +651,14 @@ __global__ void delta_binary_packed_kernel(
         if (j < vpm && bw > 0)
           my_delta = (int64_t)extract_bits64(base + pos, (int64_t)j * bw,
                                              bw);
-        int64_t step = my_delta + min_delta;
+        // lanes beyond the miniblock contribute NOTHING to the scan
+        int64_t step = (j < vpm) ? (my_delta + min_delta) : 0;
         int64_t incl = wave_incl_scan_i64(step);
         int64_t v = prev + incl;
         if (j < vpm) emit(produced + j, v);
-        // carry: value of the last lane of this pass
-        prev = wave_bcast_from(v, PSA_WAVE - 1);
+        // carry: value at the LAST VALID lane of this pass
+        int last = min(PSA_WAVE, vpm - off) - 1;
+        prev = wave_bcast_from(v, last);
       }
       produced += vpm;
       pos += (mb_bits + 7) >> 3;
@@ -747,10 +749,12 @@ __global__ void delta_length_byte_array_kernel(
         if (j < vpm && bw > 0)
           my_delta = (int64_t)extract_bits64(base + pos, (int64_t)j * bw,
                                              bw);
-        int64_t incl = wave_incl_scan_i64(my_delta + min_delta);
+        int64_t step = (j < vpm) ? (my_delta + min_delta) : 0;
+        int64_t incl = wave_incl_scan_i64(step);
         int64_t v = prev + incl;
         if (j < vpm && produced + j < want) pl[produced + j] = (int32_t)v;
-        prev = wave_bcast_from(v, PSA_WAVE - 1);
+        int last = min(PSA_WAVE, vpm - off) - 1;
+        prev = wave_bcast_from(v, last);
       }
       produced += vpm;
       pos += (mb_bits + 7) >> 3;
