@@ -40,7 +40,9 @@ _ENC_PLAIN_DICT = 2
 _ENC_RLE = 3
 _ENC_DELTA_BINARY = 5
 _ENC_DELTA_LENGTH_BA = 6
+_ENC_DELTA_BA = 7
 _ENC_RLE_DICT = 8
+_ENC_BYTE_STREAM_SPLIT = 9
 _PAGE_DATA_V1 = 0
 _PAGE_DICT = 2
 _PAGE_DATA_V2 = 3
@@ -625,7 +627,90 @@ class GpuRowGroupDecoder(object):
             return self._delta_length_byte_array(
                 ext, dev, page_buf, val_start, val_end, page_nval,
                 nonnull_per_page, valid, ch)
+        if data_enc == _ENC_DELTA_BA and phys == 'BYTE_ARRAY':
+            return self._delta_byte_array(
+                ext, dev, page_buf, val_start, val_end, page_nval,
+                nonnull_per_page, valid, ch)
+        if data_enc == _ENC_BYTE_STREAM_SPLIT and phys in _PHYS_TO_TORCH \
+                and all_valid:
+            return self._byte_stream_split(ext, dev, page_buf, val_start,
+                                           page_nval, phys)
+        if data_enc == _ENC_PLAIN and phys == 'BOOLEAN' and all_valid:
+            return self._bool_plain(ext, dev, page_buf, val_start,
+                                    page_nval)
         return self._cpu_assist_marker(ch['name'])
+
+    def _bool_plain(self, ext, dev, page_buf, val_start, page_nval):
+        """PLAIN BOOLEAN: bit-packed LSB-first -> bool tensor."""
+        counts = page_nval.astype(np.int32)
+        total = int(counts.sum())
+        out_off = np.zeros(len(counts), dtype=np.int64)
+        out_off[1:] = np.cumsum(counts)[:-1]
+        out = torch.empty(total + _SLACK, dtype=torch.uint8, device=dev)
+        ext.bool_unpack_batch(page_buf, self._up(val_start),
+                              self._up(counts), self._up(out_off), out)
+        return out[:total].to(torch.bool)
+
+    def _byte_stream_split(self, ext, dev, page_buf, val_start, page_nval,
+                           phys):
+        """BYTE_STREAM_SPLIT: de-interleave K byte planes."""
+        dtype, esize = _PHYS_TO_TORCH[phys]
+        counts = page_nval.astype(np.int32)
+        total = int(counts.sum())
+        out_off = np.zeros(len(counts), dtype=np.int64)
+        out_off[1:] = np.cumsum(counts)[:-1]
+        out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        ext.byte_stream_split_batch(page_buf, self._up(val_start),
+                                    self._up(counts), self._up(out_off),
+                                    out, esize)
+        return out[:total * esize].view(dtype)
+
+    def _delta_byte_array(self, ext, dev, page_buf, val_start, val_end,
+                          page_nval, nonnull_per_page, valid, ch):
+        """DELTA_BYTE_ARRAY (front-coded strings): lengths pass -> exact
+        output sizing (ONE host sync — unavoidable: materialized size is
+        data-dependent) -> reconstruct pass.  Values land in a fresh
+        device buffer; standard ByteArrayColumn consumers follow."""
+        counts = (nonnull_per_page if nonnull_per_page is not None
+                  else page_nval).astype(np.int64)
+        total = int(counts.sum())
+        n_pages = len(counts)
+        out_idx = np.zeros(n_pages, dtype=np.int64)
+        out_idx[1:] = np.cumsum(counts)[:-1]
+        pre = torch.empty(total + 1, dtype=torch.int32, device=dev)
+        sfx = torch.empty(total + 1, dtype=torch.int32, device=dev)
+        suf_pos = torch.empty(n_pages, dtype=torch.int64, device=dev)
+        status = self._status(n_pages)
+        counts_dev = self._up(counts.astype(np.int32))
+        out_idx_dev = self._up(out_idx)
+        ext.delta_byte_array_lengths_batch(
+            page_buf, self._up(val_start), self._up(val_end), counts_dev,
+            out_idx_dev, pre, sfx, suf_pos, status)
+        self._check(status, 'deltaba-len:' + ch['name'])
+        lens = (pre[:total] + sfx[:total]).to(torch.int64)
+        val_off = torch.cumsum(lens, 0) - lens
+        total_bytes = int((val_off[-1] + lens[-1]).item()) if total else 0
+        page_bytes = int((val_end - val_start).sum())
+        if total_bytes < 0 or total_bytes > max(1 << 20,
+                                                64 * page_bytes):
+            raise RuntimeError(
+                'DELTA_BYTE_ARRAY column {!r}: implausible materialized '
+                'size {} from {} page bytes (corrupt lengths?)'
+                .format(ch['name'], total_bytes, page_bytes))
+        out = torch.empty(total_bytes + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        st2 = self._status(n_pages)
+        ext.delta_byte_array_reconstruct_batch(
+            page_buf, counts_dev, out_idx_dev, pre, sfx, suf_pos, val_off,
+            out, st2)
+        self._check(st2, 'deltaba-rec:' + ch['name'])
+        col = ByteArrayColumn(out, val_off, lens.to(torch.int32), None,
+                              None, total)
+        if valid is not None and nonnull_per_page is not None and \
+                not bool((nonnull_per_page == page_nval).all()):
+            col.valid = valid
+        return col
 
     def _delta_fixed(self, ext, dev, page_buf, val_start, val_end,
                      page_nval, phys, ch):

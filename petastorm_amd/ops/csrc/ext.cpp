@@ -99,6 +99,23 @@ void delta_length_byte_array_batch(torch::Tensor page_buf,
                                    torch::Tensor val_off,
                                    torch::Tensor val_len,
                                    torch::Tensor status);
+void delta_byte_array_lengths_batch(torch::Tensor page_buf,
+                                    torch::Tensor start, torch::Tensor end,
+                                    torch::Tensor n_values,
+                                    torch::Tensor out_idx,
+                                    torch::Tensor pre, torch::Tensor sfx,
+                                    torch::Tensor suf_data_pos,
+                                    torch::Tensor status);
+void delta_byte_array_reconstruct_batch(
+    torch::Tensor page_buf, torch::Tensor n_values, torch::Tensor out_idx,
+    torch::Tensor pre, torch::Tensor sfx, torch::Tensor suf_data_pos,
+    torch::Tensor val_off, torch::Tensor out, torch::Tensor status);
+void byte_stream_split_batch(torch::Tensor page_buf, torch::Tensor start,
+                             torch::Tensor n_values, torch::Tensor out_off,
+                             torch::Tensor out, int64_t esize);
+void bool_unpack_batch(torch::Tensor page_buf, torch::Tensor start,
+                       torch::Tensor n_values, torch::Tensor out_off,
+                       torch::Tensor out);
 void jpeg_decode_fused_batch(torch::Tensor data, py::dict meta,
                              torch::Tensor coef, torch::Tensor samples,
                              torch::Tensor out, torch::Tensor out_off,
@@ -152,6 +169,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("delta_length_byte_array_batch",
         &psa::delta_length_byte_array_batch,
         "DELTA_LENGTH_BYTE_ARRAY page -> per-value (offset, length)");
+  m.def("delta_byte_array_lengths_batch",
+        &psa::delta_byte_array_lengths_batch,
+        "DELTA_BYTE_ARRAY pass 1: prefix/suffix lengths");
+  m.def("delta_byte_array_reconstruct_batch",
+        &psa::delta_byte_array_reconstruct_batch,
+        "DELTA_BYTE_ARRAY pass 2: materialize front-coded values");
+  m.def("byte_stream_split_batch", &psa::byte_stream_split_batch,
+        "BYTE_STREAM_SPLIT de-interleave");
+  m.def("bool_unpack_batch", &psa::bool_unpack_batch,
+        "PLAIN boolean bit-unpack");
   m.def("jpeg_decode_fused_batch", &psa::jpeg_decode_fused_batch,
         "JPEG decode with fused YCbCr->RGB + normalize + NCHW fp32 "
         "epilogue (skips the NHWC uint8 intermediate)");

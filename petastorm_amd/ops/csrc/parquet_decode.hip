@@ -529,9 +529,10 @@ void plain_fixed_decode_batch(torch::Tensor page_buf,
 // value[i+1] = value[i] + min_delta + delta[i].
 //
 // One wave per page: lane 0 walks the varints; the wave cooperatively
-// extracts 64 deltas at a time and turns them into values with a shfl
-// inclusive scan + running carry — the CDNA4-idiomatic form of the
-// sequential prefix dependency.
+// extracts 64 deltas at a time and resolves the prefix dependency with a
+// shfl inclusive scan + running carry — the CDNA4-idiomatic form of a
+// sequential dependency.  Shared by the int-column kernel, the
+// delta-length and the delta-byte-array (prefix) string kernels.
 // ---------------------------------------------------------------------------
 
 __device__ __forceinline__ uint64_t read_varint_u64(const uint8_t* p,
@@ -575,21 +576,17 @@ __device__ __forceinline__ int64_t wave_incl_scan_i64(int64_t v) {
   return v;
 }
 
-__global__ void delta_binary_packed_kernel(
-    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
-    const int64_t* __restrict__ end, const int32_t* __restrict__ n_values,
-    const int64_t* __restrict__ out_off, uint8_t* __restrict__ out,
-    int32_t esize, int32_t* __restrict__ status, int n_pages) {
-  const int waves_per_block = blockDim.x / PSA_WAVE;
-  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
-  if (page >= n_pages) return;
+// Wave-cooperative decode of one DELTA_BINARY_PACKED stream into `out`
+// (first `want` values).  Returns the stream end position (where the
+// following section, e.g. delta-length byte data, begins), or -1 on
+// malformed input (err code written via *err).
+template <typename OutT>
+__device__ int64_t delta_ints_decode_wave(const uint8_t* __restrict__ base,
+                                          int64_t pos, int64_t pend,
+                                          int32_t want,
+                                          OutT* __restrict__ out,
+                                          int32_t* err) {
   const int lane = lane_id();
-  const uint8_t* base = data;
-  int64_t pos = start[page];
-  const int64_t pend = end[page];
-  const int32_t want = n_values[page];
-
-  // header (lane 0 parses, broadcast)
   uint64_t block_size = 0, mbs_per_block = 0, total = 0;
   int64_t first = 0;
   if (lane == 0) {
@@ -605,46 +602,31 @@ __global__ void delta_binary_packed_kernel(
   pos = wave_bcast((long long)pos);
   if (block_size == 0 || mbs_per_block == 0 ||
       block_size % (mbs_per_block * 8) != 0) {
-    if (lane == 0) status[page] = 40;
-    return;
+    if (lane == 0) *err = 40;
+    return -1;
   }
-  const int vpm = (int)(block_size / mbs_per_block);  // values/miniblock
+  const int vpm = (int)(block_size / mbs_per_block);
 
-  uint8_t* dst = out + out_off[page] * esize;
-  auto emit = [&](int64_t idx, int64_t v) {
-    if (idx >= want) return;
-    if (esize == 8) {
-      uint8_t* d = dst + idx * 8;
-      uint64_t u = (uint64_t)v;
-      for (int i = 0; i < 8; ++i) d[i] = (uint8_t)(u >> (8 * i));
-    } else {
-      uint8_t* d = dst + idx * 4;
-      uint32_t u = (uint32_t)(int32_t)v;
-      for (int i = 0; i < 4; ++i) d[i] = (uint8_t)(u >> (8 * i));
-    }
-  };
-  if (lane == 0) emit(0, first);
-  int64_t produced = 1;          // values written (includes first)
-  int64_t prev = first;          // last emitted value
+  if (lane == 0 && want > 0) out[0] = (OutT)first;
+  int64_t produced = 1;
+  int64_t prev = first;
 
   while (produced < want && produced < (int64_t)total) {
-    // block header
     int64_t min_delta = 0;
     if (lane == 0) min_delta = zigzag64(read_varint_u64(base, pos, pend));
     min_delta = wave_bcast((long long)min_delta);
     pos = wave_bcast((long long)pos);
     int64_t bw_pos = pos;               // one bitwidth byte per miniblock
     pos += mbs_per_block;
-    if (pos > pend) { if (lane == 0) status[page] = 41; return; }
+    if (pos > pend) { if (lane == 0) *err = 41; return -1; }
     for (uint64_t mb = 0; mb < mbs_per_block; ++mb) {
       int bw = base[bw_pos + mb];
-      if (bw > 64) { if (lane == 0) status[page] = 42; return; }
+      if (bw > 64) { if (lane == 0) *err = 42; return -1; }
       const int64_t mb_bits = (int64_t)vpm * bw;
       if (produced >= want || produced >= (int64_t)total) {
         pos += (mb_bits + 7) >> 3;      // skip remaining miniblocks
         continue;
       }
-      // cooperative unpack + scan, 64 deltas per pass
       for (int off = 0; off < vpm; off += PSA_WAVE) {
         int64_t my_delta = 0;
         int j = off + lane;
@@ -655,16 +637,36 @@ __global__ void delta_binary_packed_kernel(
         int64_t step = (j < vpm) ? (my_delta + min_delta) : 0;
         int64_t incl = wave_incl_scan_i64(step);
         int64_t v = prev + incl;
-        if (j < vpm) emit(produced + j, v);
+        if (j < vpm && produced + j < want) out[produced + j] = (OutT)v;
         // carry: value at the LAST VALID lane of this pass
         int last = min(PSA_WAVE, vpm - off) - 1;
         prev = wave_bcast_from(v, last);
       }
       produced += vpm;
       pos += (mb_bits + 7) >> 3;
-      if (pos > pend + 8) { if (lane == 0) status[page] = 43; return; }
+      if (pos > pend + 8) { if (lane == 0) *err = 43; return -1; }
     }
   }
+  return pos;
+}
+
+__global__ void delta_binary_packed_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
+    const int64_t* __restrict__ end, const int32_t* __restrict__ n_values,
+    const int64_t* __restrict__ out_off, uint8_t* __restrict__ out,
+    int32_t esize, int32_t* __restrict__ status, int n_pages) {
+  const int waves_per_block = blockDim.x / PSA_WAVE;
+  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  if (page >= n_pages) return;
+  uint8_t* dst = out + out_off[page] * esize;
+  if (esize == 8)
+    delta_ints_decode_wave<int64_t>(data, start[page], end[page],
+                                    n_values[page], (int64_t*)dst,
+                                    status + page);
+  else
+    delta_ints_decode_wave<int32_t>(data, start[page], end[page],
+                                    n_values[page], (int32_t*)dst,
+                                    status + page);
 }
 
 void delta_binary_packed_batch(torch::Tensor page_buf, torch::Tensor start,
@@ -696,80 +698,20 @@ __global__ void delta_length_byte_array_kernel(
   const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
   if (page >= n_pages) return;
   const int lane = lane_id();
-  const uint8_t* base = data;
-  int64_t pos = start[page];
-  const int64_t pend = end[page];
   const int32_t want = n_values[page];
-
-  uint64_t block_size = 0, mbs_per_block = 0, total = 0;
-  int64_t first = 0;
-  if (lane == 0) {
-    block_size = read_varint_u64(base, pos, pend);
-    mbs_per_block = read_varint_u64(base, pos, pend);
-    total = read_varint_u64(base, pos, pend);
-    first = zigzag64(read_varint_u64(base, pos, pend));
-  }
-  block_size = wave_bcast((unsigned long long)block_size);
-  mbs_per_block = wave_bcast((unsigned long long)mbs_per_block);
-  total = wave_bcast((unsigned long long)total);
-  first = wave_bcast((long long)first);
-  pos = wave_bcast((long long)pos);
-  if (block_size == 0 || mbs_per_block == 0 ||
-      block_size % (mbs_per_block * 8) != 0) {
-    if (lane == 0) status[page] = 45;
-    return;
-  }
-  const int vpm = (int)(block_size / mbs_per_block);
-
   int64_t* po = val_off + out_idx[page];
   int32_t* pl = val_len + out_idx[page];
-  // first pass: lengths -> val_len, and a running byte total
-  if (lane == 0 && want > 0) pl[0] = (int32_t)first;
-  int64_t produced = 1;
-  int64_t prev = first;
-  while (produced < want && produced < (int64_t)total) {
-    int64_t min_delta = 0;
-    if (lane == 0) min_delta = zigzag64(read_varint_u64(base, pos, pend));
-    min_delta = wave_bcast((long long)min_delta);
-    pos = wave_bcast((long long)pos);
-    int64_t bw_pos = pos;
-    pos += mbs_per_block;
-    if (pos > pend) { if (lane == 0) status[page] = 46; return; }
-    for (uint64_t mb = 0; mb < mbs_per_block; ++mb) {
-      int bw = base[bw_pos + mb];
-      if (bw > 64) { if (lane == 0) status[page] = 47; return; }
-      const int64_t mb_bits = (int64_t)vpm * bw;
-      if (produced >= want || produced >= (int64_t)total) {
-        pos += (mb_bits + 7) >> 3;
-        continue;
-      }
-      for (int off = 0; off < vpm; off += PSA_WAVE) {
-        int64_t my_delta = 0;
-        int j = off + lane;
-        if (j < vpm && bw > 0)
-          my_delta = (int64_t)extract_bits64(base + pos, (int64_t)j * bw,
-                                             bw);
-        int64_t step = (j < vpm) ? (my_delta + min_delta) : 0;
-        int64_t incl = wave_incl_scan_i64(step);
-        int64_t v = prev + incl;
-        if (j < vpm && produced + j < want) pl[produced + j] = (int32_t)v;
-        int last = min(PSA_WAVE, vpm - off) - 1;
-        prev = wave_bcast_from(v, last);
-      }
-      produced += vpm;
-      pos += (mb_bits + 7) >> 3;
-      if (pos > pend + 8) { if (lane == 0) status[page] = 48; return; }
-    }
-  }
-  // second pass: offsets = pos (bytes base) + exclusive scan of lengths.
-  // lane 0 walks sequentially (tiny: one add per value).
+  int64_t bytes0 = delta_ints_decode_wave<int32_t>(
+      data, start[page], end[page], want, pl, status + page);
+  if (bytes0 < 0) return;
+  __threadfence();  // lane 0 re-reads lengths other lanes stored
   if (lane == 0) {
-    int64_t byte_pos = pos;
+    int64_t byte_pos = bytes0;
     for (int32_t i = 0; i < want; ++i) {
       po[i] = byte_pos;
       byte_pos += pl[i];
     }
-    if (byte_pos > pend) status[page] = 49;
+    if (byte_pos > end[page]) status[page] = 49;
   }
 }
 
@@ -793,6 +735,176 @@ void delta_length_byte_array_batch(torch::Tensor page_buf,
                      val_off.data_ptr<int64_t>(),
                      val_len.data_ptr<int32_t>(),
                      status.data_ptr<int32_t>(), n);
+}
+
+// ---------------------------------------------------------------------------
+// DELTA_BYTE_ARRAY (incremental/front-coded strings — Spark v2 writers):
+// <prefix lengths: delta block> <suffix lengths+bytes: DELTA_LENGTH>.
+// value[i] = value[i-1][:prefix_len[i]] + suffix[i].  Two kernels: the
+// LENGTHS pass decodes prefix/suffix lengths (host then sizes the exact
+// output and cumsums offsets); the RECONSTRUCT pass materializes values
+// (sequential prefix dependency per page, wave-parallel byte copies).
+// ---------------------------------------------------------------------------
+__global__ void delta_byte_array_lengths_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
+    const int64_t* __restrict__ end, const int32_t* __restrict__ n_values,
+    const int64_t* __restrict__ out_idx,
+    int32_t* __restrict__ pre, int32_t* __restrict__ sfx,
+    int64_t* __restrict__ suf_data_pos, int32_t* __restrict__ status,
+    int n_pages) {
+  const int waves_per_block = blockDim.x / PSA_WAVE;
+  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  if (page >= n_pages) return;
+  const int32_t want = n_values[page];
+  int64_t suf_pos = delta_ints_decode_wave<int32_t>(
+      data, start[page], end[page], want, pre + out_idx[page],
+      status + page);
+  if (suf_pos < 0) return;
+  int64_t bytes0 = delta_ints_decode_wave<int32_t>(
+      data, suf_pos, end[page], want, sfx + out_idx[page], status + page);
+  if (bytes0 < 0) return;
+  if (lane_id() == 0) suf_data_pos[page] = bytes0;
+}
+
+__global__ void delta_byte_array_reconstruct_kernel(
+    const uint8_t* __restrict__ data, const int32_t* __restrict__ n_values,
+    const int64_t* __restrict__ out_idx,
+    const int32_t* __restrict__ pre, const int32_t* __restrict__ sfx,
+    const int64_t* __restrict__ suf_data_pos,
+    const int64_t* __restrict__ val_off,   // per value, into `out`
+    uint8_t* __restrict__ out, int64_t out_cap,
+    int32_t* __restrict__ status, int n_pages) {
+  const int waves_per_block = blockDim.x / PSA_WAVE;
+  const int page = blockIdx.x * waves_per_block + (threadIdx.x / PSA_WAVE);
+  if (page >= n_pages) return;
+  const int lane = lane_id();
+  const int32_t want = n_values[page];
+  const int64_t base_idx = out_idx[page];
+  int64_t src = suf_data_pos[page];
+  for (int32_t i = 0; i < want; ++i) {
+    int32_t p = pre[base_idx + i], sl = sfx[base_idx + i];
+    int64_t dst = val_off[base_idx + i];
+    if (p < 0 || sl < 0 || (i == 0 && p != 0) || dst + p + sl > out_cap) {
+      if (lane == 0) status[page] = 50;
+      return;
+    }
+    if (p > 0) {
+      int64_t prev = val_off[base_idx + i - 1];
+      for (int b = lane; b < p; b += PSA_WAVE) out[dst + b] = out[prev + b];
+    }
+    for (int b = lane; b < sl; b += PSA_WAVE)
+      out[dst + p + b] = data[src + b];
+    src += sl;
+    // iteration i+1's prefix copy reads THIS value's bytes: wait for the
+    // wave's own stores (no cross-wave traffic; vmcnt wait only)
+    __threadfence();
+  }
+}
+
+void delta_byte_array_lengths_batch(torch::Tensor page_buf,
+                                    torch::Tensor start, torch::Tensor end,
+                                    torch::Tensor n_values,
+                                    torch::Tensor out_idx,
+                                    torch::Tensor pre, torch::Tensor sfx,
+                                    torch::Tensor suf_data_pos,
+                                    torch::Tensor status) {
+  int n = (int)start.numel();
+  if (!n) return;
+  const int WPB = 4;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(delta_byte_array_lengths_kernel,
+                     dim3((n + WPB - 1) / WPB), dim3(WPB * PSA_WAVE), 0,
+                     stream, page_buf.data_ptr<uint8_t>(),
+                     start.data_ptr<int64_t>(), end.data_ptr<int64_t>(),
+                     n_values.data_ptr<int32_t>(),
+                     out_idx.data_ptr<int64_t>(), pre.data_ptr<int32_t>(),
+                     sfx.data_ptr<int32_t>(),
+                     suf_data_pos.data_ptr<int64_t>(),
+                     status.data_ptr<int32_t>(), n);
+}
+
+void delta_byte_array_reconstruct_batch(
+    torch::Tensor page_buf, torch::Tensor n_values, torch::Tensor out_idx,
+    torch::Tensor pre, torch::Tensor sfx, torch::Tensor suf_data_pos,
+    torch::Tensor val_off, torch::Tensor out, torch::Tensor status) {
+  int n = (int)n_values.numel();
+  if (!n) return;
+  const int WPB = 4;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(delta_byte_array_reconstruct_kernel,
+                     dim3((n + WPB - 1) / WPB), dim3(WPB * PSA_WAVE), 0,
+                     stream, page_buf.data_ptr<uint8_t>(),
+                     n_values.data_ptr<int32_t>(),
+                     out_idx.data_ptr<int64_t>(), pre.data_ptr<int32_t>(),
+                     sfx.data_ptr<int32_t>(),
+                     suf_data_pos.data_ptr<int64_t>(),
+                     val_off.data_ptr<int64_t>(), out.data_ptr<uint8_t>(),
+                     out.numel(), status.data_ptr<int32_t>(), n);
+}
+
+// ---------------------------------------------------------------------------
+// BYTE_STREAM_SPLIT: K byte planes (plane j = byte j of every value) —
+// de-interleave.  Trivially parallel grid-stride over values.
+// ---------------------------------------------------------------------------
+__global__ void byte_stream_split_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
+    const int32_t* __restrict__ n_values,
+    const int64_t* __restrict__ out_off, uint8_t* __restrict__ out,
+    int32_t esize, int n_pages) {
+  const int page = blockIdx.y;
+  if (page >= n_pages) return;
+  const int32_t n = n_values[page];
+  const uint8_t* src = data + start[page];
+  uint8_t* dst = out + out_off[page] * esize;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    for (int j = 0; j < esize; ++j)
+      dst[i * esize + j] = src[(int64_t)j * n + i];
+}
+
+void byte_stream_split_batch(torch::Tensor page_buf, torch::Tensor start,
+                             torch::Tensor n_values, torch::Tensor out_off,
+                             torch::Tensor out, int64_t esize) {
+  int n = (int)start.numel();
+  if (!n) return;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(byte_stream_split_kernel, dim3(64, n), dim3(256), 0,
+                     stream, page_buf.data_ptr<uint8_t>(),
+                     start.data_ptr<int64_t>(),
+                     n_values.data_ptr<int32_t>(),
+                     out_off.data_ptr<int64_t>(), out.data_ptr<uint8_t>(),
+                     (int32_t)esize, n);
+}
+
+// ---------------------------------------------------------------------------
+// PLAIN BOOLEAN: bit-packed LSB-first -> one uint8 per value.
+// ---------------------------------------------------------------------------
+__global__ void bool_unpack_kernel(
+    const uint8_t* __restrict__ data, const int64_t* __restrict__ start,
+    const int32_t* __restrict__ n_values,
+    const int64_t* __restrict__ out_off, uint8_t* __restrict__ out,
+    int n_pages) {
+  const int page = blockIdx.y;
+  if (page >= n_pages) return;
+  const int32_t n = n_values[page];
+  const uint8_t* src = data + start[page];
+  uint8_t* dst = out + out_off[page];
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = (src[i >> 3] >> (i & 7)) & 1;
+}
+
+void bool_unpack_batch(torch::Tensor page_buf, torch::Tensor start,
+                       torch::Tensor n_values, torch::Tensor out_off,
+                       torch::Tensor out) {
+  int n = (int)start.numel();
+  if (!n) return;
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(bool_unpack_kernel, dim3(16, n), dim3(256), 0, stream,
+                     page_buf.data_ptr<uint8_t>(), start.data_ptr<int64_t>(),
+                     n_values.data_ptr<int32_t>(),
+                     out_off.data_ptr<int64_t>(), out.data_ptr<uint8_t>(),
+                     n);
 }
 
 }  // namespace psa

@@ -29,7 +29,10 @@ class _StubExt(object):
                 'lz4_decompress_batch',
                 'inflate_batch', 'png_unfilter_batch', 'bswap16',
                 'delta_binary_packed_batch',
-                'delta_length_byte_array_batch'}
+                'delta_length_byte_array_batch',
+                'delta_byte_array_lengths_batch',
+                'delta_byte_array_reconstruct_batch',
+                'byte_stream_split_batch', 'bool_unpack_batch'}
 
     def __init__(self):
         self._real = ops.ext()
@@ -40,7 +43,8 @@ class _StubExt(object):
     # making the dry run flaky on heap reuse)
     _OUTPUTS = {'rle_hybrid_decode_batch': (6,),
                 'plain_fixed_decode_batch': (11,),
-                'varlen_gather': (3,)}
+                'varlen_gather': (3,),
+                'delta_byte_array_lengths_batch': (5, 6, 7)}
 
     def __getattr__(self, name):
         if name in self._KERNELS:
@@ -399,4 +403,35 @@ def test_dryrun_delta_encodings(stub_decoder, tmp_path):
     assert out['s'] is not None and out['s'].n == 200
     assert 'delta_binary_packed_batch' in stub.calls
     assert 'delta_length_byte_array_batch' in stub.calls
+    assert not dec.cpu_assist_columns
+
+
+def test_dryrun_bool_bss_delta_ba(stub_decoder, tmp_path):
+    """BOOLEAN / BYTE_STREAM_SPLIT / DELTA_BYTE_ARRAY orchestration with
+    stubbed kernels: kernel selection and shape plumbing."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    dec, stub = stub_decoder
+    d = tmp_path / 'enc3'
+    d.mkdir()
+    n = 300
+    table = pa.Table.from_arrays(
+        [pa.array(np.random.rand(n).astype(np.float32)),
+         pa.array(np.arange(n) % 2 == 0),
+         pa.array(['p-%d' % (i // 5) for i in range(n)], pa.string())],
+        schema=pa.schema([pa.field('f', pa.float32(), nullable=False),
+                          pa.field('b', pa.bool_(), nullable=False),
+                          pa.field('s', pa.string(), nullable=False)]))
+    pq.write_table(table, str(d / 'p.parquet'), use_dictionary=False,
+                   compression='none',
+                   column_encoding={'f': 'BYTE_STREAM_SPLIT',
+                                    's': 'DELTA_BYTE_ARRAY',
+                                    'b': 'PLAIN'})
+    out, sch = _decode_all(dec, 'file://' + str(d), ['f', 'b', 's'])
+    assert out['f'] is not None and out['f'].shape == (n,)
+    assert out['b'] is not None and out['b'].dtype == torch.bool
+    assert out['s'] is not None and out['s'].n == n
+    assert 'byte_stream_split_batch' in stub.calls
+    assert 'bool_unpack_batch' in stub.calls
+    assert 'delta_byte_array_lengths_batch' in stub.calls
     assert not dec.cpu_assist_columns

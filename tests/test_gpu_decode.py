@@ -1309,3 +1309,45 @@ def test_delta_length_byte_array_with_nulls_gpu(ext, tmp_path):
             got.append(np.asarray(b.s, dtype=object))
         assert r.diagnostics['cpu_assist_columns'] == []
     assert np.concatenate(got).tolist() == svals
+
+
+def test_bool_bss_delta_byte_array_gpu(ext, tmp_path):
+    """BOOLEAN bit-unpack, BYTE_STREAM_SPLIT floats and DELTA_BYTE_ARRAY
+    (front-coded) strings decode on GPU, exact vs source."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'enc3'
+    d.mkdir()
+    n = 4000
+    rng = np.random.RandomState(5)
+    f32 = rng.rand(n).astype(np.float32)
+    f64 = rng.rand(n)
+    bools = (np.arange(n) % 3 == 0)
+    strs = ['shared-prefix-%06d/suffix-%s' % (i // 9, 'q' * (i % 13))
+            for i in range(n)]
+    table = pa.table({'id': pa.array(np.arange(n, dtype=np.int64)),
+                      'f': pa.array(f32), 'd': pa.array(f64),
+                      'b': pa.array(bools), 's': pa.array(strs)})
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=1000,
+                   use_dictionary=False, compression='snappy',
+                   column_encoding={'f': 'BYTE_STREAM_SPLIT',
+                                    'd': 'BYTE_STREAM_SPLIT',
+                                    's': 'DELTA_BYTE_ARRAY',
+                                    'id': 'PLAIN', 'b': 'PLAIN'},
+                   data_page_size=8 << 10)
+    got = {'id': [], 'f': [], 'd': [], 'b': [], 's': []}
+    with make_batch_reader('file://' + str(d), device='cuda',
+                           shuffle_row_groups=False) as r:
+        for batch in r:
+            for k in got:
+                v = getattr(batch, k)
+                got[k].append(v.cpu().numpy() if hasattr(v, 'cpu')
+                              else np.asarray(v, dtype=object))
+        assert r.diagnostics['cpu_assist_columns'] == []
+    ids = np.concatenate(got['id'])
+    np.testing.assert_array_equal(ids, np.arange(n))
+    np.testing.assert_array_equal(np.concatenate(got['f']), f32)
+    np.testing.assert_array_equal(np.concatenate(got['d']), f64)
+    np.testing.assert_array_equal(np.concatenate(got['b']), bools)
+    assert np.concatenate(got['s']).tolist() == strs
