@@ -540,6 +540,18 @@ class GpuRowGroupDecoder(object):
             return self._plain_fixed_fused(ext, dev, page_buf, p_start,
                                            p_end, page_nval, row0, max_def,
                                            n_rows, phys, ch['name'])
+        if data_enc == _ENC_PLAIN and phys == 'INT96':
+            sizes = uncomp_size if snappy else comp_size
+            page_nval = num_values[data_idx]
+            p_start = np.array([page_start[i] for i in data_idx],
+                               dtype=np.int64)
+            p_end = np.array([page_start[i] + sizes[i] for i in data_idx],
+                             dtype=np.int64)
+            row0 = np.zeros(len(data_idx), dtype=np.int64)
+            row0[1:] = np.cumsum(page_nval)[:-1]
+            return self._int96_timestamps(ext, dev, page_buf, p_start,
+                                          p_end, page_nval, row0, max_def,
+                                          ch['name'])
 
         # 3) locate per-page def-level and value sections
         val_start = np.empty(len(data_idx), dtype=np.int64)
@@ -639,6 +651,29 @@ class GpuRowGroupDecoder(object):
             return self._bool_plain(ext, dev, page_buf, val_start,
                                     page_nval)
         return self._cpu_assist_marker(ch['name'])
+
+    def _int96_timestamps(self, ext, dev, page_buf, p_start, p_end,
+                          page_nval, row0, max_def, name):
+        """PLAIN INT96 (legacy Spark timestamps): 8B nanos-in-day +
+        4B julian day -> int64 nanoseconds since the unix epoch; the
+        reader boundary surfaces datetime64[ns]."""
+        total = int(page_nval.sum())
+        out = torch.empty(total * 12 + _SLACK, dtype=torch.uint8,
+                          device=dev)
+        status = self._status(len(p_start))
+        empty8 = torch.empty(0, dtype=torch.uint8, device=dev)
+        empty64 = torch.empty(0, dtype=torch.int64, device=dev)
+        ext.plain_fixed_decode_batch(
+            page_buf, self._up(p_start), self._up(p_end),
+            self._up(page_nval.astype(np.int32)), self._up(row0),
+            1 if max_def > 0 else 0, 12, 0,
+            empty8, empty64, empty64, out, empty8, status)
+        self._check(status, 'int96:' + name)
+        m = out[:total * 12].view(total, 12)
+        nanos = m[:, :8].contiguous().view(torch.int64).reshape(total)
+        day = m[:, 8:12].contiguous().view(torch.int32).reshape(total) \
+            .to(torch.int64)
+        return (day - 2440588) * 86_400_000_000_000 + nanos
 
     def _bool_plain(self, ext, dev, page_buf, val_start, page_nval):
         """PLAIN BOOLEAN: bit-packed LSB-first -> bool tensor."""

@@ -486,6 +486,8 @@ class GpuBatchReader(object):
             col = pschema.column(i)
             if col.path.split('.')[0] != name:
                 continue
+            if col.physical_type == 'INT96':
+                return 'ns'  # legacy Spark timestamps decode to int64 ns
             lt = str(col.logical_type).lower()
             if lt.startswith('date'):
                 return 'D'

@@ -336,6 +336,10 @@ __device__ __forceinline__ void copy_elem(uint8_t* dst, const uint8_t* src,
     *(uint32_t*)(dst + 4) = hi;
   } else if (esize == 4) {
     *(uint32_t*)dst = load_u32_unaligned(src);
+  } else if (esize == 12) {  // INT96 (legacy Spark timestamps)
+    *(uint32_t*)dst = load_u32_unaligned(src);
+    *(uint32_t*)(dst + 4) = load_u32_unaligned(src + 4);
+    *(uint32_t*)(dst + 8) = load_u32_unaligned(src + 8);
   } else if (esize == 2) {
     dst[0] = src[0];
     dst[1] = src[1];

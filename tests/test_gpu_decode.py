@@ -1351,3 +1351,47 @@ def test_bool_bss_delta_byte_array_gpu(ext, tmp_path):
     np.testing.assert_array_equal(np.concatenate(got['d']), f64)
     np.testing.assert_array_equal(np.concatenate(got['b']), bools)
     assert np.concatenate(got['s']).tolist() == strs
+
+
+def test_int96_timestamps_gpu(ext, tmp_path):
+    """Legacy Spark INT96 timestamps decode on GPU to datetime64[ns],
+    matching the CPU route exactly."""
+    import datetime
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'int96'
+    d.mkdir()
+    n = 1000
+    base = datetime.datetime(2001, 3, 4, 5, 6, 7)
+    stamps = [base + datetime.timedelta(seconds=17 * i, microseconds=i % 997)
+              for i in range(n)]
+    table = pa.table({'id': pa.array(np.arange(n, dtype=np.int64)),
+                      'ts': pa.array(stamps, pa.timestamp('us'))})
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=250,
+                   use_dictionary=False, compression='snappy',
+                   use_deprecated_int96_timestamps=True)
+    pf = pq.ParquetFile(str(d / 'p.parquet'))
+    assert pf.metadata.row_group(0).column(1).physical_type == 'INT96'
+
+    def read_all(device):
+        got = {}
+        kwargs = {'device': device} if device else {}
+        with make_batch_reader('file://' + str(d), shuffle_row_groups=False,
+                               **kwargs) as r:
+            assists = None
+            for b in r:
+                ids = np.asarray(b.id.cpu() if hasattr(b.id, 'cpu')
+                                 else b.id)
+                ts = np.asarray(b.ts)
+                for i, rid in enumerate(ids):
+                    got[int(rid)] = ts[i]
+            assists = list(r.diagnostics.get('cpu_assist_columns', []))
+        return got, assists
+
+    gpu, assist = read_all('cuda')
+    cpu, _ = read_all(None)
+    assert 'ts' not in assist
+    for rid in range(n):
+        assert np.datetime64(gpu[rid], 'ns') == np.datetime64(cpu[rid], 'ns')
+        assert np.datetime64(gpu[rid], 'us') == np.datetime64(stamps[rid])
