@@ -204,7 +204,8 @@ class GpuRowGroupDecoder(object):
                            col.physical_type, col.compression,
                            col.num_values,
                            parquet_schema.column(ci).max_definition_level,
-                           parquet_schema.column(ci).max_repetition_level))
+                           parquet_schema.column(ci).max_repetition_level,
+                           getattr(parquet_schema.column(ci), 'length', 0)))
         # merge the requested chunk byte ranges into extents (gaps below 256
         # KiB are read through rather than seeking) so unrequested columns
         # between them are not read or uploaded
@@ -242,8 +243,8 @@ class GpuRowGroupDecoder(object):
             'chunks': [
                 dict(name=n, col_index=ci, offset=_rebase(s), length=ln,
                      physical=pt, compression=comp, num_values=nv,
-                     max_def=mdl, max_rep=mrl)
-                for (n, ci, s, ln, pt, comp, nv, mdl, mrl) in chunks],
+                     max_def=mdl, max_rep=mrl, type_length=tl)
+                for (n, ci, s, ln, pt, comp, nv, mdl, mrl, tl) in chunks],
         }
         return host, chunk_meta
 
@@ -650,7 +651,52 @@ class GpuRowGroupDecoder(object):
         if data_enc == _ENC_PLAIN and phys == 'BOOLEAN' and all_valid:
             return self._bool_plain(ext, dev, page_buf, val_start,
                                     page_nval)
+        if data_enc == _ENC_PLAIN and phys == 'FIXED_LEN_BYTE_ARRAY' and \
+                ch.get('type_length', 0) > 0:
+            return self._plain_flba(ext, dev, page_buf, host_buf,
+                                    val_start, page_nval,
+                                    nonnull_per_page, valid, host_visible,
+                                    schema, ch)
         return self._cpu_assist_marker(ch['name'])
+
+    def _plain_flba(self, ext, dev, page_buf, host_buf, val_start,
+                    page_nval, nonnull_per_page, valid, host_visible,
+                    schema, ch):
+        """PLAIN FIXED_LEN_BYTE_ARRAY: values are contiguous with a
+        fixed stride.  float16 (logical Float16) views the gathered bytes
+        directly as a half tensor; everything else becomes a fixed-stride
+        ByteArrayColumn (bytes at the boundary)."""
+        L = int(ch['type_length'])
+        counts = (nonnull_per_page if nonnull_per_page is not None
+                  else page_nval).astype(np.int64)
+        total = int(counts.sum())
+        all_valid = nonnull_per_page is None or \
+            bool((nonnull_per_page == page_nval).all())
+        field = schema.fields.get(ch['name'])
+        if field is not None and field.numpy_dtype is np.float16 and \
+                all_valid and L == 2:
+            out = torch.empty(total * L + _SLACK, dtype=torch.uint8,
+                              device=dev)
+            dst_off = np.zeros(len(counts), dtype=np.int64)
+            dst_off[1:] = np.cumsum(counts * L)[:-1]
+            ext.varlen_gather(page_buf, self._up(val_start),
+                              self._up(counts * L), out,
+                              self._up(dst_off))
+            return out[:total * L].view(torch.float16)
+        starts = np.concatenate(
+            [vs + np.arange(c, dtype=np.int64) * L
+             for vs, c in zip(val_start, counts)]) if total else \
+            np.zeros(0, dtype=np.int64)
+        col = ByteArrayColumn(page_buf, self._up(starts),
+                              torch.full((total,), L, dtype=torch.int32,
+                                         device=dev),
+                              host_buf if host_visible else None,
+                              starts if host_visible else None, total)
+        if host_visible:
+            col.host_val_len = np.full(total, L, dtype=np.int64)
+        if valid is not None and not all_valid:
+            col.valid = valid
+        return col
 
     def _int96_timestamps(self, ext, dev, page_buf, p_start, p_end,
                           page_nval, row0, max_def, name):

@@ -324,9 +324,14 @@ def _numpy_and_codec_from_arrow_type(arrow_type):
         return np.float64, (), None
     if pt.is_boolean(t):
         return np.bool_, (), None
+    if pt.is_float16(t):
+        return np.float16, (), None
     if pt.is_string(t) or pt.is_large_string(t):
         return np.str_, (), None
-    if pt.is_binary(t) or pt.is_large_binary(t):
+    if pt.is_binary(t) or pt.is_large_binary(t) or \
+            pt.is_fixed_size_binary(t):
+        # fixed_size_binary -> bytes, like the reference's np.string_
+        # mapping (unischema.py:491-493)
         return np.bytes_, (), None
     if pt.is_decimal(t):
         return Decimal, (), None

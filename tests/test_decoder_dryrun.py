@@ -435,3 +435,26 @@ def test_dryrun_bool_bss_delta_ba(stub_decoder, tmp_path):
     assert 'bool_unpack_batch' in stub.calls
     assert 'delta_byte_array_lengths_batch' in stub.calls
     assert not dec.cpu_assist_columns
+
+
+def test_dryrun_flba(stub_decoder, tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    dec, stub = stub_decoder
+    d = tmp_path / 'flba'
+    d.mkdir()
+    n = 100
+    table = pa.Table.from_arrays(
+        [pa.array(np.random.rand(n).astype(np.float16)),
+         pa.array([b'ab12'] * n, pa.binary(4))],
+        schema=pa.schema([pa.field('h', pa.float16(), nullable=False),
+                          pa.field('fb', pa.binary(4), nullable=False)]))
+    pq.write_table(table, str(d / 'p.parquet'), use_dictionary=False,
+                   compression='none')
+    out, sch = _decode_all(dec, 'file://' + str(d), ['h', 'fb'])
+    assert out['h'] is not None and out['h'].dtype == torch.float16
+    assert out['fb'] is not None and out['fb'].n == n
+    # uncompressed fixed binary is host-visible: exact bytes on CPU
+    got = dec.decode_string_column(out['fb'], sch.fields['fb'])
+    assert got.tolist() == [b'ab12'] * n
+    assert not dec.cpu_assist_columns

@@ -1395,3 +1395,34 @@ def test_int96_timestamps_gpu(ext, tmp_path):
     for rid in range(n):
         assert np.datetime64(gpu[rid], 'ns') == np.datetime64(cpu[rid], 'ns')
         assert np.datetime64(gpu[rid], 'us') == np.datetime64(stamps[rid])
+
+
+def test_flba_float16_and_fixed_binary_gpu(ext, tmp_path):
+    """FIXED_LEN_BYTE_ARRAY: float16 columns view directly as half
+    tensors; fixed-size binary surfaces bytes — exact values."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'flba'
+    d.mkdir()
+    n = 2000
+    rng = np.random.RandomState(7)
+    h = rng.rand(n).astype(np.float16)
+    fb = [bytes([i % 251, (i * 7) % 251, 3, 4]) for i in range(n)]
+    table = pa.table({'id': pa.array(np.arange(n, dtype=np.int64)),
+                      'h': pa.array(h),
+                      'fb': pa.array(fb, pa.binary(4))})
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=500,
+                   use_dictionary=False, compression='snappy')
+    got_h, got_fb, got_id = [], [], []
+    with make_batch_reader('file://' + str(d), device='cuda',
+                           shuffle_row_groups=False) as r:
+        for b in r:
+            got_id.append(b.id.cpu().numpy())
+            assert b.h.dtype == torch.float16
+            got_h.append(b.h.cpu().numpy())
+            got_fb.append(np.asarray(b.fb, dtype=object))
+        assert r.diagnostics['cpu_assist_columns'] == []
+    np.testing.assert_array_equal(np.concatenate(got_id), np.arange(n))
+    np.testing.assert_array_equal(np.concatenate(got_h), h)
+    assert np.concatenate(got_fb).tolist() == fb
