@@ -651,6 +651,20 @@ class GpuRowGroupDecoder(object):
         if data_enc == _ENC_PLAIN and phys == 'BOOLEAN' and all_valid:
             return self._bool_plain(ext, dev, page_buf, val_start,
                                     page_nval)
+        if data_enc == _ENC_RLE and phys == 'BOOLEAN' and all_valid:
+            # RLE-encoded booleans: u32 length prefix + hybrid runs (bw=1)
+            counts32 = page_nval.astype(np.int32)
+            total = int(counts32.sum())
+            out_off = np.zeros(len(counts32), dtype=np.int64)
+            out_off[1:] = np.cumsum(counts32)[:-1]
+            out32 = torch.empty(total, dtype=torch.int32, device=dev)
+            status = self._status(len(counts32))
+            ext.rle_hybrid_decode_batch(
+                page_buf, self._up(val_start + 4), self._up(val_end),
+                torch.ones(len(counts32), dtype=torch.int32, device=dev),
+                self._up(counts32), self._up(out_off), out32, status)
+            self._check(status, 'rlebool:' + ch['name'])
+            return out32.to(torch.bool)
         if data_enc == _ENC_PLAIN and phys == 'FIXED_LEN_BYTE_ARRAY' and \
                 ch.get('type_length', 0) > 0:
             return self._plain_flba(ext, dev, page_buf, host_buf,

@@ -1426,3 +1426,27 @@ def test_flba_float16_and_fixed_binary_gpu(ext, tmp_path):
     np.testing.assert_array_equal(np.concatenate(got_id), np.arange(n))
     np.testing.assert_array_equal(np.concatenate(got_h), h)
     assert np.concatenate(got_fb).tolist() == fb
+
+
+def test_rle_boolean_column_gpu(ext, tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'rlebool'
+    d.mkdir()
+    n = 3000
+    vals = (np.arange(n) % 17 < 5) | (np.arange(n) % 97 == 0)
+    table = pa.table({'id': pa.array(np.arange(n, dtype=np.int64)),
+                      'b': pa.array(vals)})
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=750,
+                   use_dictionary=False, compression='snappy',
+                   column_encoding={'b': 'RLE', 'id': 'PLAIN'})
+    got_b, got_id = [], []
+    with make_batch_reader('file://' + str(d), device='cuda',
+                           shuffle_row_groups=False) as r:
+        for batch in r:
+            got_id.append(batch.id.cpu().numpy())
+            got_b.append(batch.b.cpu().numpy())
+        assert r.diagnostics['cpu_assist_columns'] == []
+    np.testing.assert_array_equal(np.concatenate(got_id), np.arange(n))
+    np.testing.assert_array_equal(np.concatenate(got_b), vals)
