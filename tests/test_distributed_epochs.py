@@ -146,3 +146,29 @@ def test_uneven_rank_progress_does_not_deadlock(tmp_path, scalar_dataset):
     for p in procs:
         p.join(timeout=30)
     assert results[0] > 0 and results[1] > 0
+
+
+@pytest.mark.timeout(240)
+def test_sharded_readers_world4(tmp_path, scalar_dataset):
+    """4-process gloo coverage: disjoint shards, full union — the shape
+    of the 8-GPU driver run at half scale."""
+    ctx = mp.get_context('spawn')
+    q = ctx.Queue()
+    store = str(tmp_path / 'store4')
+    world = 4
+    procs = [ctx.Process(target=_worker_sharded_read,
+                         args=(r, world, store, scalar_dataset['url'], q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, ids = q.get(timeout=200)
+        results[rank] = ids
+    for p in procs:
+        p.join(timeout=30)
+    union = sorted(sum(results.values(), []))
+    assert union == sorted(list(range(500)) * 2)
+    for a in range(world):
+        for b in range(a + 1, world):
+            assert not (set(results[a]) & set(results[b]))
