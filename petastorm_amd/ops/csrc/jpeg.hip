@@ -203,6 +203,7 @@ __device__ __forceinline__ int receive_extend(BitReader& br, int s) {
 // register passed by reference (a pred[c] array would be runtime-indexed and
 // spill to scratch — guide rule: dynamic-indexed locals live in local
 // memory).  Returns 0 on success, an error code otherwise.
+template <bool NT>
 __device__ __forceinline__ int decode_block(BitReader& br,
                                             const JpegTables& tabs,
                                             const float* __restrict__ q,
@@ -213,7 +214,12 @@ __device__ __forceinline__ int decode_block(BitReader& br,
   if (t < 0 || t > 15) return 10;
   int diff = t ? receive_extend(br, t) : 0;
   pred += diff;
-  out[0] = (float)pred * q[0];
+  // Non-temporal stores: coefficients are written once here and read
+  // once (much later) by the IDCT kernel — keeping them OUT of L2
+  // preserves it for this kernel's latency-critical bitstream refills
+  // (85.7% L2 hit at the tuned config, profiles/r2_huffman_pmc.txt).
+  if (NT) __builtin_nontemporal_store((float)pred * q[0], out);
+  else out[0] = (float)pred * q[0];
   int kk = 1;
   while (kk < 64) {
     int rs = huff_decode(br, tabs, ac_t);
@@ -226,13 +232,16 @@ __device__ __forceinline__ int decode_block(BitReader& br,
       kk += r;
       if (kk > 63) return 12;
       int v = receive_extend(br, sz);
-      out[ZIGZAG_NAT[kk]] = (float)v * q[kk];
+      if (NT) __builtin_nontemporal_store((float)v * q[kk],
+                                          out + ZIGZAG_NAT[kk]);
+      else out[ZIGZAG_NAT[kk]] = (float)v * q[kk];
       ++kk;
     }
   }
   return 0;
 }
 
+template <bool NT>
 __global__ void jpeg_huffman_kernel(
     const uint8_t* __restrict__ data, JpegTables tabs, JpegGeom g,
     const int32_t* __restrict__ seg_img, const int64_t* __restrict__ seg_pos,
@@ -273,12 +282,12 @@ __global__ void jpeg_huffman_kernel(
     float* mcu_out = coef + (blk0 + (int64_t)(mcu0 + m) * bpm) * 64;
     int rc = 0;
     for (int r = 0; r < rep0 && !rc; ++r, mcu_out += 64)
-      rc = decode_block(br, tabs, q0, dc0, ac0, mcu_out, pred0);
+      rc = decode_block<NT>(br, tabs, q0, dc0, ac0, mcu_out, pred0);
     if (!rc && ncomp == 3) {
-      rc = decode_block(br, tabs, q1, dc1, ac1, mcu_out, pred1);
+      rc = decode_block<NT>(br, tabs, q1, dc1, ac1, mcu_out, pred1);
       mcu_out += 64;
       if (!rc)
-        rc = decode_block(br, tabs, q2, dc2, ac2, mcu_out, pred2);
+        rc = decode_block<NT>(br, tabs, q2, dc2, ac2, mcu_out, pred2);
     }
     if (rc) { status[s] = rc; return; }
   }
@@ -579,15 +588,34 @@ void jpeg_decode_batch(torch::Tensor data, py::dict meta, torch::Tensor coef,
   int64_t block_total = coef.numel() / 64;
   hipStream_t stream = c10::hip::getCurrentHIPStream();
 
-  hipLaunchKernelGGL(jpeg_huffman_kernel,
-                     dim3((n_segs + 63) / 64), dim3(64), 0, stream,
-                     data.data_ptr<uint8_t>(), tabs, g,
-                     seg_img.data_ptr<int32_t>(),
-                     seg_pos.data_ptr<int64_t>(),
-                     seg_end.data_ptr<int64_t>(),
-                     seg_mcu0.data_ptr<int32_t>(),
-                     seg_nmcu.data_ptr<int32_t>(), coef.data_ptr<float>(),
-                     status.data_ptr<int32_t>(), n_segs);
+  // PSA_JPEG_NT=1 enables non-temporal coefficient stores.  MEASURED
+  // WORSE (same-box A/B, r2: 485k vs 615k samples/s): the scattered
+  // 4-byte stores lose L2 write-combining and hit HBM as partial lines.
+  // Kept as a knob because the result is instructive; default OFF.
+  static const bool use_nt = [] {
+    const char* e = getenv("PSA_JPEG_NT");
+    return e && e[0] == '1';
+  }();
+  if (use_nt)
+    hipLaunchKernelGGL(jpeg_huffman_kernel<true>,
+                       dim3((n_segs + 63) / 64), dim3(64), 0, stream,
+                       data.data_ptr<uint8_t>(), tabs, g,
+                       seg_img.data_ptr<int32_t>(),
+                       seg_pos.data_ptr<int64_t>(),
+                       seg_end.data_ptr<int64_t>(),
+                       seg_mcu0.data_ptr<int32_t>(),
+                       seg_nmcu.data_ptr<int32_t>(), coef.data_ptr<float>(),
+                       status.data_ptr<int32_t>(), n_segs);
+  else
+    hipLaunchKernelGGL(jpeg_huffman_kernel<false>,
+                       dim3((n_segs + 63) / 64), dim3(64), 0, stream,
+                       data.data_ptr<uint8_t>(), tabs, g,
+                       seg_img.data_ptr<int32_t>(),
+                       seg_pos.data_ptr<int64_t>(),
+                       seg_end.data_ptr<int64_t>(),
+                       seg_mcu0.data_ptr<int32_t>(),
+                       seg_nmcu.data_ptr<int32_t>(), coef.data_ptr<float>(),
+                       status.data_ptr<int32_t>(), n_segs);
 
   hipLaunchKernelGGL(jpeg_idct_kernel,
                      dim3((unsigned)((block_total + 255) / 256)), dim3(256),
@@ -628,15 +656,34 @@ void jpeg_decode_fused_batch(torch::Tensor data, py::dict meta,
   int64_t block_total = coef.numel() / 64;
   hipStream_t stream = c10::hip::getCurrentHIPStream();
 
-  hipLaunchKernelGGL(jpeg_huffman_kernel,
-                     dim3((n_segs + 63) / 64), dim3(64), 0, stream,
-                     data.data_ptr<uint8_t>(), tabs, g,
-                     seg_img.data_ptr<int32_t>(),
-                     seg_pos.data_ptr<int64_t>(),
-                     seg_end.data_ptr<int64_t>(),
-                     seg_mcu0.data_ptr<int32_t>(),
-                     seg_nmcu.data_ptr<int32_t>(), coef.data_ptr<float>(),
-                     status.data_ptr<int32_t>(), n_segs);
+  // PSA_JPEG_NT=1 enables non-temporal coefficient stores.  MEASURED
+  // WORSE (same-box A/B, r2: 485k vs 615k samples/s): the scattered
+  // 4-byte stores lose L2 write-combining and hit HBM as partial lines.
+  // Kept as a knob because the result is instructive; default OFF.
+  static const bool use_nt = [] {
+    const char* e = getenv("PSA_JPEG_NT");
+    return e && e[0] == '1';
+  }();
+  if (use_nt)
+    hipLaunchKernelGGL(jpeg_huffman_kernel<true>,
+                       dim3((n_segs + 63) / 64), dim3(64), 0, stream,
+                       data.data_ptr<uint8_t>(), tabs, g,
+                       seg_img.data_ptr<int32_t>(),
+                       seg_pos.data_ptr<int64_t>(),
+                       seg_end.data_ptr<int64_t>(),
+                       seg_mcu0.data_ptr<int32_t>(),
+                       seg_nmcu.data_ptr<int32_t>(), coef.data_ptr<float>(),
+                       status.data_ptr<int32_t>(), n_segs);
+  else
+    hipLaunchKernelGGL(jpeg_huffman_kernel<false>,
+                       dim3((n_segs + 63) / 64), dim3(64), 0, stream,
+                       data.data_ptr<uint8_t>(), tabs, g,
+                       seg_img.data_ptr<int32_t>(),
+                       seg_pos.data_ptr<int64_t>(),
+                       seg_end.data_ptr<int64_t>(),
+                       seg_mcu0.data_ptr<int32_t>(),
+                       seg_nmcu.data_ptr<int32_t>(), coef.data_ptr<float>(),
+                       status.data_ptr<int32_t>(), n_segs);
 
   hipLaunchKernelGGL(jpeg_idct_kernel,
                      dim3((unsigned)((block_total + 255) / 256)), dim3(256),
