@@ -28,6 +28,7 @@ setup(
             'petastorm-amd-generate-metadata = '
             'petastorm_amd.etl.petastorm_generate_metadata:main',
             'petastorm-amd-metadata-util = petastorm_amd.etl.metadata_util:main',
+            'petastorm-amd-reencode = petastorm_amd.tools.reencode_dataset:main',
         ],
     },
 )
