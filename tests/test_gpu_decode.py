@@ -1450,3 +1450,52 @@ def test_rle_boolean_column_gpu(ext, tmp_path):
         assert r.diagnostics['cpu_assist_columns'] == []
     np.testing.assert_array_equal(np.concatenate(got_id), np.arange(n))
     np.testing.assert_array_equal(np.concatenate(got_b), vals)
+
+
+def test_batched_dataloader_over_gpu_reader(ext, tmp_path):
+    """BatchedDataLoader + on-device shuffling pool over the GPU reader:
+    all rows delivered once per epoch, tensors stay on device."""
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.pytorch import BatchedDataLoader
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'ld')
+    create_scalar_dataset(url, num_rows=3000, rowgroup_size=500)
+    reader = make_batch_reader(url, device='cuda', num_epochs=1, seed=11,
+                               shuffle_row_groups=True,
+                               schema_fields=['id', 'f0'])
+    loader = BatchedDataLoader(reader, batch_size=256,
+                               shuffling_queue_capacity=1024, seed=5)
+    ids = []
+    for b in loader:
+        assert b['id'].is_cuda
+        ids.extend(b['id'].cpu().numpy().tolist())
+    reader.stop()
+    reader.join()
+    assert sorted(ids) == list(range(3000))
+    assert ids != sorted(ids)  # the pool actually shuffled
+
+
+def test_inmem_batched_dataloader_gpu(ext, tmp_path):
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.pytorch import InMemBatchedDataLoader
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'inmem')
+    create_scalar_dataset(url, num_rows=1000, rowgroup_size=250)
+    reader = make_batch_reader(url, device='cuda', num_epochs=1,
+                               shuffle_row_groups=False,
+                               schema_fields=['id', 'i0'])
+    loader = InMemBatchedDataLoader(reader, batch_size=128, num_epochs=2,
+                                    rows_capacity=1000, shuffle=True,
+                                    seed=3)
+    epochs_ids = []
+    for _ in range(2):  # one iter() per epoch (reference pytorch.py:437+)
+        ids = []
+        for b in loader:
+            ids.extend(b['id'].cpu().numpy().tolist())
+        epochs_ids.append(ids)
+    reader.stop()
+    reader.join()
+    assert len(epochs_ids) == 2
+    assert sorted(epochs_ids[0]) == list(range(1000))
+    assert sorted(epochs_ids[1]) == list(range(1000))
+    assert epochs_ids[0] != epochs_ids[1]  # per-epoch reshuffle
