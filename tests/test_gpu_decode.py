@@ -1499,3 +1499,38 @@ def test_inmem_batched_dataloader_gpu(ext, tmp_path):
     assert sorted(epochs_ids[0]) == list(range(1000))
     assert sorted(epochs_ids[1]) == list(range(1000))
     assert epochs_ids[0] != epochs_ids[1]  # per-epoch reshuffle
+
+
+def test_gpu_reader_reset_and_string_predicate(ext, tmp_path):
+    """reset() restarts epoch iteration on the GPU route, and predicates
+    over STRING fields (host-evaluated mask on device columns) filter
+    correctly — incl. the upstream-exact pseudorandom split."""
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.predicates import in_lambda, in_pseudorandom_split
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'resetp')
+    cols = create_scalar_dataset(url, num_rows=1000, rowgroup_size=250)
+
+    pred = in_lambda(['name'], lambda v: np.char.endswith(
+        np.asarray(v['name'], dtype=np.str_), '7'))
+    with make_batch_reader(url, device='cuda', num_epochs=1,
+                           shuffle_row_groups=False, predicate=pred,
+                           schema_fields=['id', 'name']) as r:
+        ids1 = np.sort(np.concatenate(
+            [b.id.cpu().numpy() for b in r]))
+        r.reset()
+        ids2 = np.sort(np.concatenate(
+            [b.id.cpu().numpy() for b in r]))
+    expected = np.array([i for i in range(1000)
+                         if cols['name'][i].endswith('7')])
+    np.testing.assert_array_equal(ids1, expected)
+    np.testing.assert_array_equal(ids2, expected)
+
+    split = in_pseudorandom_split([0.6, 0.4], 0, 'name')
+    with make_batch_reader(url, device='cuda', num_epochs=1,
+                           shuffle_row_groups=False, predicate=split,
+                           schema_fields=['id', 'name']) as r:
+        got = np.sort(np.concatenate([b.id.cpu().numpy() for b in r]))
+    keep = np.array([i for i in range(1000)
+                     if split.do_include({'name': cols['name'][i]})])
+    np.testing.assert_array_equal(got, keep)
