@@ -97,3 +97,39 @@ def test_batched_random_min_after():
     assert not b.can_retrieve()
     b.add_many(_cols(16, 48))
     assert b.can_retrieve()
+
+
+# ---------------------------------------------------------------------------
+# BatchingQueue (reference pyarrow_helpers/batching_table_queue.py:20-79)
+# ---------------------------------------------------------------------------
+
+def test_batching_queue_rechunks_exact_batches():
+    import numpy as np
+    from petastorm_amd.reader_impl.batching_queue import BatchingQueue
+    q = BatchingQueue(batch_size=10)
+    total = 0
+    for n in (3, 9, 25, 2, 14):  # uneven puts
+        q.put({'a': np.arange(total, total + n),
+               'b': np.arange(total, total + n) * 2.0})
+        total += n
+    got = []
+    while not q.empty():
+        b = q.get()
+        assert len(b['a']) == 10
+        np.testing.assert_array_equal(b['b'], b['a'] * 2.0)
+        got.extend(b['a'].tolist())
+    assert got == list(range((total // 10) * 10))
+    assert q.size == total % 10
+    assert q.get() is None  # not enough rows buffered
+
+
+def test_batching_queue_torch_tensors():
+    import numpy as np
+    import torch
+    from petastorm_amd.reader_impl.batching_queue import BatchingQueue
+    q = BatchingQueue(batch_size=4)
+    q.put({'x': torch.arange(3)})
+    q.put({'x': torch.arange(3, 9)})
+    b = q.get()
+    assert isinstance(b['x'], torch.Tensor)
+    np.testing.assert_array_equal(b['x'].numpy(), [0, 1, 2, 3])
