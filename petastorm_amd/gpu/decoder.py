@@ -464,6 +464,7 @@ class GpuRowGroupDecoder(object):
             return torch.empty(0, device=dev)
         if any(page_type[i] == _PAGE_DATA_V2 for i in data_idx_early):
             return self._decode_v2_chunk(ext, dev, dbuf, host_buf, ch,
+                                         schema,
                                          pages, data_idx_early, n_rows)
 
         # 1) page payload location: either in dbuf directly, or in a
@@ -875,16 +876,22 @@ class GpuRowGroupDecoder(object):
         self._check(status, 'plainfixed:' + name)
         return out[:total * esize].view(dtype)
 
-    def _decode_v2_chunk(self, ext, dev, dbuf, host_buf, ch, pages,
-                         data_idx, n_rows):
+    def _decode_v2_chunk(self, ext, dev, dbuf, host_buf, ch, schema,
+                         pages, data_idx, n_rows):
         """DataPageV2: levels are stored uncompressed with explicit byte
         lengths; only the values section is compressed (Parquet format
         spec).  Supported for PLAIN fixed-width columns; everything else
         takes the CPU assist."""
         phys = ch['physical']
         enc = pages['encoding'].numpy()
-        if phys not in _PHYS_TO_TORCH or \
-                any(enc[i] != _ENC_PLAIN for i in data_idx):
+        supported_enc = {_ENC_PLAIN, _ENC_DELTA_BINARY,
+                         _ENC_DELTA_LENGTH_BA, _ENC_DELTA_BA,
+                         _ENC_BYTE_STREAM_SPLIT}
+        supported_phys = set(_PHYS_TO_TORCH) | {'BYTE_ARRAY', 'BOOLEAN',
+                                                'FIXED_LEN_BYTE_ARRAY'}
+        if phys not in supported_phys or \
+                any(enc[i] not in supported_enc for i in data_idx) or \
+                len({enc[i] for i in data_idx}) != 1:
             return self._cpu_assist_marker(ch['name'])
         data_off = pages['data_off'].numpy()
         comp_size = pages['comp_size'].numpy()
@@ -960,21 +967,78 @@ class GpuRowGroupDecoder(object):
             val_end = (data_off[idx] + comp_size[idx]).astype(np.int64)
         row0 = np.zeros(n, dtype=np.int64)
         row0[1:] = np.cumsum(page_nval)[:-1]
-        dtype, esize = _PHYS_TO_TORCH[phys]
-        total = int(page_nval.sum())
-        out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
-                          device=dev)
-        status2 = self._status(n)
-        has_def = 2 if ch['max_def'] > 0 else 0
-        empty8 = torch.empty(0, dtype=torch.uint8, device=dev)
-        ext.plain_fixed_decode_batch(
-            val_buf, self._up(val_start), self._up(val_end),
-            self._up(page_nval.astype(np.int32)), self._up(row0),
-            has_def, esize, self._FILL_PATTERNS[phys],
-            dbuf, self._up(lev_start), self._up(lev_len),
-            out, empty8, status2)
-        self._check(status2, 'v2-plainfixed:' + ch['name'])
-        return out[:total * esize].view(dtype)
+        data_enc = enc[idx[0]]
+        if data_enc == _ENC_PLAIN and phys in _PHYS_TO_TORCH:
+            dtype, esize = _PHYS_TO_TORCH[phys]
+            total = int(page_nval.sum())
+            out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
+                              device=dev)
+            status2 = self._status(n)
+            has_def = 2 if ch['max_def'] > 0 else 0
+            empty8 = torch.empty(0, dtype=torch.uint8, device=dev)
+            ext.plain_fixed_decode_batch(
+                val_buf, self._up(val_start), self._up(val_end),
+                self._up(page_nval.astype(np.int32)), self._up(row0),
+                has_def, esize, self._FILL_PATTERNS[phys],
+                dbuf, self._up(lev_start), self._up(lev_len),
+                out, empty8, status2)
+            self._check(status2, 'v2-plainfixed:' + ch['name'])
+            return out[:total * esize].view(dtype)
+
+        # other encodings share the V1 helpers; V2 levels (uncompressed,
+        # no length prefix) decode here to validity/non-null counts
+        valid = None
+        nonnull_per_page = None
+        if ch['max_def'] > 0:
+            md = int(ch['max_def'])
+            bw = max(1, int(np.ceil(np.log2(md + 1))))
+            lv_off = np.zeros(n + 1, dtype=np.int64)
+            lv_off[1:] = np.cumsum(page_nval)
+            levels = torch.empty(int(lv_off[-1]), dtype=torch.int32,
+                                 device=dev)
+            lst = self._status(n)
+            ext.rle_hybrid_decode_batch(
+                dbuf, self._up(lev_start), self._up(lev_start + lev_len),
+                torch.full((n,), bw, dtype=torch.int32, device=dev),
+                self._up(page_nval.astype(np.int32)),
+                self._up(lv_off[:-1]), levels, lst)
+            self._check(lst, 'v2-deflevels:' + ch['name'])
+            valid = levels == md
+            vmat = valid.split([int(x) for x in page_nval])
+            nonnull_per_page = np.array(
+                [int(v.sum().item()) for v in vmat], dtype=np.int64)
+        all_valid = nonnull_per_page is None or \
+            bool((nonnull_per_page == page_nval).all())
+        if data_enc == _ENC_PLAIN and phys == 'BYTE_ARRAY':
+            self._plan_entry = {}
+            return self._plain_byte_array(
+                ext, dev, val_buf, host_buf, val_start, val_end, page_nval,
+                nonnull_per_page, valid, False, ch)
+        if data_enc == _ENC_PLAIN and phys == 'BOOLEAN' and all_valid:
+            return self._bool_plain(ext, dev, val_buf, val_start,
+                                    page_nval)
+        if data_enc == _ENC_PLAIN and phys == 'FIXED_LEN_BYTE_ARRAY' and \
+                ch.get('type_length', 0) > 0:
+            return self._plain_flba(ext, dev, val_buf, host_buf, val_start,
+                                    page_nval, nonnull_per_page, valid,
+                                    False, schema, ch)
+        if data_enc == _ENC_DELTA_BINARY and phys in ('INT32', 'INT64') \
+                and all_valid:
+            return self._delta_fixed(ext, dev, val_buf, val_start, val_end,
+                                     page_nval, phys, ch)
+        if data_enc == _ENC_DELTA_LENGTH_BA and phys == 'BYTE_ARRAY':
+            return self._delta_length_byte_array(
+                ext, dev, val_buf, val_start, val_end, page_nval,
+                nonnull_per_page, valid, ch)
+        if data_enc == _ENC_DELTA_BA and phys == 'BYTE_ARRAY':
+            return self._delta_byte_array(
+                ext, dev, val_buf, val_start, val_end, page_nval,
+                nonnull_per_page, valid, ch)
+        if data_enc == _ENC_BYTE_STREAM_SPLIT and phys in _PHYS_TO_TORCH \
+                and all_valid:
+            return self._byte_stream_split(ext, dev, val_buf, val_start,
+                                           page_nval, phys)
+        return self._cpu_assist_marker(ch['name'])
 
     def _plain_byte_array(self, ext, dev, page_buf, host_buf, val_start,
                           val_end, page_nval, nonnull_per_page, valid,

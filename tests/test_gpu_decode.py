@@ -1534,3 +1534,44 @@ def test_gpu_reader_reset_and_string_predicate(ext, tmp_path):
     keep = np.array([i for i in range(1000)
                      if split.do_include({'name': cols['name'][i]})])
     np.testing.assert_array_equal(got, keep)
+
+
+def test_v2_pages_delta_and_bool_gpu(ext, tmp_path):
+    """DataPageV2 with the widened encodings (delta ints, delta-length
+    strings, byte-stream-split floats, plain bools): exact values,
+    snappy-compressed values sections."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'v2enc'
+    d.mkdir()
+    n = 3000
+    rng = np.random.RandomState(9)
+    i64 = rng.randint(-2**50, 2**50, n)
+    f32 = rng.rand(n).astype(np.float32)
+    strs = ['v2-%d-%s' % (i, 'z' * (i % 11)) for i in range(n)]
+    bools = np.arange(n) % 5 < 2
+    table = pa.table({'id': pa.array(np.arange(n, dtype=np.int64)),
+                      'i64': pa.array(i64), 'f': pa.array(f32),
+                      's': pa.array(strs), 'b': pa.array(bools)})
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=750,
+                   use_dictionary=False, compression='snappy',
+                   data_page_version='2.0', data_page_size=16 << 10,
+                   column_encoding={'i64': 'DELTA_BINARY_PACKED',
+                                    'f': 'BYTE_STREAM_SPLIT',
+                                    's': 'DELTA_LENGTH_BYTE_ARRAY',
+                                    'id': 'PLAIN', 'b': 'PLAIN'})
+    got = {k: [] for k in ('id', 'i64', 'f', 's', 'b')}
+    with make_batch_reader('file://' + str(d), device='cuda',
+                           shuffle_row_groups=False) as r:
+        for batch in r:
+            for k in got:
+                v = getattr(batch, k)
+                got[k].append(v.cpu().numpy() if hasattr(v, 'cpu')
+                              else np.asarray(v, dtype=object))
+        assert r.diagnostics['cpu_assist_columns'] == []
+    np.testing.assert_array_equal(np.concatenate(got['id']), np.arange(n))
+    np.testing.assert_array_equal(np.concatenate(got['i64']), i64)
+    np.testing.assert_array_equal(np.concatenate(got['f']), f32)
+    np.testing.assert_array_equal(np.concatenate(got['b']), bools)
+    assert np.concatenate(got['s']).tolist() == strs
