@@ -562,15 +562,28 @@ class GpuBatchReader(object):
                 else:
                     columns[name] = decoded
             else:
-                if field is not None and field.numpy_dtype is np.datetime64 \
-                        and isinstance(col, torch.Tensor):
-                    # DATE/TIMESTAMP: physical ints decoded on device;
-                    # materialize numpy datetime64 at the boundary (torch
-                    # has no datetime dtype — CPU-route type parity,
-                    # reference unischema.py:467-502 mapping)
-                    unit = self._datetime_unit(piece.path, name) or 'us'
-                    col = col.cpu().numpy().astype(
-                        'datetime64[{}]'.format(unit))
+                dt = field.numpy_dtype if field is not None else None
+                if isinstance(col, torch.Tensor):
+                    if dt is np.datetime64:
+                        # DATE/TIMESTAMP: physical ints decoded on device;
+                        # materialize numpy datetime64 at the boundary
+                        # (torch has no datetime dtype — CPU-route type
+                        # parity, reference unischema.py:467-502 mapping)
+                        unit = self._datetime_unit(piece.path, name) or 'us'
+                        col = col.cpu().numpy().astype(
+                            'datetime64[{}]'.format(unit))
+                    # unsigned logical types over signed physical storage:
+                    # same widening convention as the ndarray codec path
+                    # (reference pytorch.py:40-70 sanitization)
+                    elif dt is np.uint8:
+                        col = col.to(torch.uint8)
+                    elif dt is np.uint16:
+                        col = col.to(torch.int32) & 0xFFFF
+                    elif dt is np.uint32:
+                        col = col.to(torch.int64) & 0xFFFFFFFF
+                    elif dt is np.uint64:
+                        # no torch uint64: numpy at the boundary
+                        col = col.cpu().numpy().view(np.uint64)
                 columns[name] = col
         if assist:
             columns.update(self._cpu_assist(piece, assist))

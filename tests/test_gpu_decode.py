@@ -1575,3 +1575,46 @@ def test_v2_pages_delta_and_bool_gpu(ext, tmp_path):
     np.testing.assert_array_equal(np.concatenate(got['f']), f32)
     np.testing.assert_array_equal(np.concatenate(got['b']), bools)
     assert np.concatenate(got['s']).tolist() == strs
+
+
+def test_unsigned_logical_types_gpu(ext, tmp_path):
+    """Unsigned logical types over signed physical storage: values exact
+    at full range (uint8/16/32/64), widened per the framework convention
+    (u16->i32, u32->i64, u64->numpy)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'uns'
+    d.mkdir()
+    n = 300
+    rng = np.random.RandomState(13)
+    u8 = rng.randint(0, 256, n).astype(np.uint8)
+    u16 = rng.randint(0, 2**16, n).astype(np.uint16)
+    u32 = (rng.randint(0, 2**31, n).astype(np.uint64) * 2 + 1) \
+        .astype(np.uint32)
+    u64 = (rng.randint(0, 2**62, n).astype(np.uint64) * 4 + 3)
+    table = pa.table({'id': pa.array(np.arange(n, dtype=np.int64)),
+                      'u8': pa.array(u8), 'u16': pa.array(u16),
+                      'u32': pa.array(u32), 'u64': pa.array(u64)})
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=100,
+                   use_dictionary=False, compression='snappy')
+    got = {k: [] for k in ('id', 'u8', 'u16', 'u32', 'u64')}
+    with make_batch_reader('file://' + str(d), device='cuda',
+                           shuffle_row_groups=False) as r:
+        for b in r:
+            assert b.u8.dtype == torch.uint8
+            assert b.u16.dtype == torch.int32
+            assert b.u32.dtype == torch.int64
+            assert isinstance(b.u64, np.ndarray) and \
+                b.u64.dtype == np.uint64
+            for k in got:
+                v = getattr(b, k)
+                got[k].append(v.cpu().numpy() if hasattr(v, 'cpu')
+                              else np.asarray(v))
+        assert r.diagnostics['cpu_assist_columns'] == []
+    np.testing.assert_array_equal(np.concatenate(got['u8']), u8)
+    np.testing.assert_array_equal(np.concatenate(got['u16']),
+                                  u16.astype(np.int32))
+    np.testing.assert_array_equal(np.concatenate(got['u32']),
+                                  u32.astype(np.int64))
+    np.testing.assert_array_equal(np.concatenate(got['u64']), u64)
