@@ -577,6 +577,10 @@ class GpuBatchReader(object):
                     # (reference pytorch.py:40-70 sanitization)
                     elif dt is np.uint8:
                         col = col.to(torch.uint8)
+                    elif dt is np.int8:
+                        col = col.to(torch.int8)
+                    elif dt is np.int16:
+                        col = col.to(torch.int16)
                     elif dt is np.uint16:
                         col = col.to(torch.int32) & 0xFFFF
                     elif dt is np.uint32:
