@@ -440,7 +440,7 @@ class GpuRowGroupDecoder(object):
         pages = plan_entry['pages']
         if ch['compression'] == 'ZSTD':
             z = plan_entry.get('zstd')
-            if z is None:  # V2+ZSTD (values-section framing): assist path
+            if z is None:  # defensive: plan missing (never in normal flow)
                 return self._cpu_assist_marker(ch['name'])
             host_buf = z['host_buf']
             dbuf = host_buf.to(dev, non_blocking=True)
