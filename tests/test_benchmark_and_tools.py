@@ -174,3 +174,21 @@ def test_reencode_dataset_adds_restart_markers(tmp_path):
             diff = np.abs(a.image.astype(np.int16) -
                           b.image.astype(np.int16))
             assert diff.mean() < 6  # one extra jpeg generation at q95
+
+
+def test_examples_run(tmp_path):
+    """The example scripts execute end to end on CPU (reference keeps
+    per-example tests, examples/*/tests)."""
+    import os
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for script, args in (
+            ('examples/hello_world/main.py',
+             ['file://' + str(tmp_path / 'hw_ex')]),
+            ('examples/hello_world/external_dataset.py', []),
+    ):
+        out = subprocess.run([sys.executable, os.path.join(root, script)]
+                             + args, capture_output=True, text=True,
+                             timeout=240, cwd=root)
+        assert out.returncode == 0, (script, out.stderr[-1500:])
