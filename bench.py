@@ -188,11 +188,19 @@ def bench_scalar(args, rank, world, device, dist):
     from petastorm_amd.pytorch import BatchedDataLoader
     from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
 
-    n_rows = args.rows or 2_000_000
+    # rowgroup/page sweep on MI355X (profiles/RESULTS.md r2.7): the
+    # config was host-dispatch-bound at 62.5k-row groups; 1M-row groups
+    # with 16 KiB pages reach ~206M rows/s (24 groups -> 3 per shard at
+    # 8 GPUs)
+    n_rows = args.rows or 24_000_000
     comp = os.environ.get('PSA_SCALAR_COMPRESSION', 'snappy')
-    url = _dataset_dir('scalar_{}_{}'.format(n_rows, comp), rank, dist,
+    page_kb = os.environ.get('PSA_SCALAR_PAGE_KB', '16')
+    rg_rows = int(os.environ.get('PSA_SCALAR_RG', '1000000'))
+    url = _dataset_dir('scalar_{}_{}_p{}_g{}'.format(n_rows, comp, page_kb,
+                                                     rg_rows),
+                       rank, dist,
                        lambda u: create_scalar_dataset(
-                           u, num_rows=n_rows, rowgroup_size=62500,
+                           u, num_rows=n_rows, rowgroup_size=rg_rows,
                            compression=comp))
     reader = make_batch_reader(
         url, device=str(device), num_epochs=None, shuffle_row_groups=True,
@@ -267,10 +275,13 @@ def bench_ngram(args, rank, world, device, dist):
     from petastorm_amd.pytorch import BatchedDataLoader
     from petastorm_amd.test_util.dataset_gen import create_sequence_dataset
 
-    n_rows = args.rows or 100_000
-    url = _dataset_dir('seq_{}'.format(n_rows), rank, dist,
+    # rowgroup sweep r2.7: 25k-row groups beat 6.25k by ~16%; 16 groups
+    # keep 8-GPU shards equal (2 each)
+    n_rows = args.rows or 400_000
+    seq_rg = int(os.environ.get('PSA_SEQ_RG', '25000'))
+    url = _dataset_dir('seq_{}_g{}'.format(n_rows, seq_rg), rank, dist,
                        lambda u: create_sequence_dataset(
-                           u, num_rows=n_rows, rows_per_rowgroup=6250))
+                           u, num_rows=n_rows, rows_per_rowgroup=seq_rg))
     pred = in_lambda(['source'], lambda v: v['source'] != 3)  # keep 3/4
     reader = make_batch_reader(
         url, device=str(device), num_epochs=None, shuffle_row_groups=True,

@@ -172,9 +172,12 @@ def create_scalar_dataset(url, num_rows=1000, num_float_cols=8,
     fs.makedirs(path, exist_ok=True)
     # 32 KiB pages: the page is the decompression-parallelism unit on the
     # GPU (one wave per snappy stream); small pages keep 256 CUs fed
+    # (sweepable via PSA_SCALAR_PAGE_KB)
+    import os
+    page_kb = int(os.environ.get('PSA_SCALAR_PAGE_KB', '32'))
     pq.write_table(table, path + '/data-00000.parquet',
                    row_group_size=rowgroup_size, compression=compression,
-                   use_dictionary=False, data_page_size=32 << 10)
+                   use_dictionary=False, data_page_size=page_kb << 10)
     return cols
 
 
