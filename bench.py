@@ -279,7 +279,9 @@ def bench_ngram(args, rank, world, device, dist):
     # keep 8-GPU shards equal (2 each)
     n_rows = args.rows or 400_000
     seq_rg = int(os.environ.get('PSA_SEQ_RG', '25000'))
-    url = _dataset_dir('seq_{}_g{}'.format(n_rows, seq_rg), rank, dist,
+    wpk = os.environ.get('PSA_WRITER_PAGE_KB', '256')
+    url = _dataset_dir('seq_{}_g{}_w{}'.format(n_rows, seq_rg, wpk), rank,
+                       dist,
                        lambda u: create_sequence_dataset(
                            u, num_rows=n_rows, rows_per_rowgroup=seq_rg))
     pred = in_lambda(['source'], lambda v: v['source'] != 3)  # keep 3/4

@@ -142,7 +142,9 @@ class DatasetWriter(object):
                 # 256 KiB pages: the page is the GPU decode-parallelism unit
                 # (one wave per page), so smaller-than-arrow-default pages
                 # keep the 256-CU chip fed on few-column datasets
-                data_page_size=256 << 10)
+                # (sweepable via PSA_WRITER_PAGE_KB)
+                data_page_size=int(os.environ.get(
+                    'PSA_WRITER_PAGE_KB', '256')) << 10)
         self._writer.write_table(table)
         self._buffer = []
         self._buffer_bytes = 0
