@@ -1651,3 +1651,83 @@ def test_v2_zstd_delta_gpu(ext, tmp_path):
     np.testing.assert_array_equal(np.concatenate(got['id']), np.arange(n))
     np.testing.assert_array_equal(np.concatenate(got['i64']), i64)
     assert np.concatenate(got['s']).tolist() == strs
+
+
+def test_corrupt_jpeg_fails_loudly_gpu(ext, tmp_path):
+    """A corrupted jpeg payload must raise a decode error (status
+    machinery), never hang or return garbage silently."""
+    import glob
+    import pyarrow.parquet as pq
+    import pyarrow as pa
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_imagenet_dataset
+    url_dir = str(tmp_path / 'corrupt')
+    create_imagenet_dataset('file://' + url_dir, num_rows=8,
+                            rowgroup_size_mb=8)
+    f = glob.glob(url_dir + '/*.parquet')[0]
+    t = pq.read_table(f)
+    imgs = t.column('image').to_pylist()
+    bad = bytearray(imgs[3])
+    # corrupt the header structure (invalid marker lengths).  NB: stomping
+    # ENTROPY bytes instead hits the graceful truncated-stream path — an
+    # unstuffed 0xFF reads as end-of-segment and the remaining MCUs decode
+    # as zero-padded data, libjpeg's incomplete-stream behavior.
+    sos = bytes(bad).find(b'\xff\xda')
+    assert sos > 0
+    bad[2:sos] = bytes(len(bad[2:sos]))  # zero every header segment
+    imgs[3] = bytes(bad)
+    t2 = t.set_column(t.schema.get_field_index('image'), 'image',
+                      pa.array(imgs, pa.binary()))
+    pq.write_table(t2, f, row_group_size=8, use_dictionary=False,
+                   compression='none')
+    with pytest.raises(Exception) as ei:
+        with make_batch_reader('file://' + url_dir, device='cuda',
+                               shuffle_row_groups=False) as r:
+            list(r)
+    msg = str(ei.value).lower()
+    assert 'jpeg' in msg or 'decode' in msg or 'marker' in msg or \
+        'image' in msg
+
+
+def test_two_concurrent_gpu_readers(ext, tmp_path):
+    """Two independent GPU readers in one process (separate stream pools)
+    interleave without interference."""
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    u1 = 'file://' + str(tmp_path / 'a')
+    u2 = 'file://' + str(tmp_path / 'b')
+    create_scalar_dataset(u1, num_rows=2000, rowgroup_size=500, seed=1)
+    create_scalar_dataset(u2, num_rows=1500, rowgroup_size=500, seed=2)
+    r1 = make_batch_reader(u1, device='cuda', num_epochs=1,
+                           shuffle_row_groups=False, schema_fields=['id'])
+    r2 = make_batch_reader(u2, device='cuda', num_epochs=1,
+                           shuffle_row_groups=False, schema_fields=['id'])
+    it1, it2 = iter(r1), iter(r2)
+    got1, got2 = [], []
+    while True:
+        done = 0
+        for it, acc in ((it1, got1), (it2, got2)):
+            try:
+                acc.append(next(it).id.cpu().numpy())
+            except StopIteration:
+                done += 1
+        if done == 2:
+            break
+    for r in (r1, r2):
+        r.stop()
+        r.join()
+    assert sorted(np.concatenate(got1).tolist()) == list(range(2000))
+    assert sorted(np.concatenate(got2).tolist()) == list(range(1500))
+
+
+def test_throughput_harness_gpu(ext, tmp_path):
+    """The reference-shaped benchmark harness runs against the GPU batch
+    reader (reference throughput.py:112-172 with device='cuda')."""
+    from petastorm_amd.benchmark.throughput import reader_throughput
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'thr')
+    create_scalar_dataset(url, num_rows=5000, rowgroup_size=1000)
+    res = reader_throughput(url, warmup_cycles_count=2,
+                            measure_cycles_count=5, read_method='batch',
+                            device='cuda')
+    assert res.samples_per_second > 0
