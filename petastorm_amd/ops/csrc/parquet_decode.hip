@@ -712,7 +712,8 @@ __global__ void delta_length_byte_array_kernel(
   if (lane == 0) {
     int64_t byte_pos = bytes0;
     for (int32_t i = 0; i < want; ++i) {
-      po[i] = byte_pos;
+      if (pl[i] < 0) { status[page] = 49; return; }  // corrupt length:
+      po[i] = byte_pos;        // negatives would send gathers out of bounds
       byte_pos += pl[i];
     }
     if (byte_pos > end[page]) status[page] = 49;

@@ -1731,3 +1731,32 @@ def test_throughput_harness_gpu(ext, tmp_path):
                             measure_cycles_count=5, read_method='batch',
                             device='cuda')
     assert res.samples_per_second > 0
+
+
+def test_corrupt_delta_page_fails_loudly_gpu(ext, tmp_path):
+    """Garbage bytes in a DELTA_LENGTH_BYTE_ARRAY page must raise, never
+    gather out of bounds or hang."""
+    import glob
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'cdelta'
+    d.mkdir()
+    n = 500
+    table = pa.table({'s': pa.array(['x%04d' % i for i in range(n)])})
+    pq.write_table(table, str(d / 'p.parquet'), use_dictionary=False,
+                   compression='none',
+                   column_encoding={'s': 'DELTA_LENGTH_BYTE_ARRAY'})
+    f = glob.glob(str(d) + '/*.parquet')[0]
+    raw = bytearray(open(f, 'rb').read())
+    pf = pq.ParquetFile(f)
+    col = pf.metadata.row_group(0).column(0)
+    off = col.data_page_offset
+    # stomp the delta header + first miniblocks with 0xAA garbage
+    for i in range(off + 12, off + 200):
+        raw[i] = 0xAA
+    open(f, 'wb').write(bytes(raw))
+    with pytest.raises(Exception):
+        with make_batch_reader('file://' + str(d), device='cuda',
+                               shuffle_row_groups=False) as r:
+            list(r)
