@@ -724,17 +724,26 @@ class GpuRowGroupDecoder(object):
         status = self._status(len(p_start))
         empty8 = torch.empty(0, dtype=torch.uint8, device=dev)
         empty64 = torch.empty(0, dtype=torch.int64, device=dev)
+        valid_out = torch.empty(total if max_def > 0 else 0,
+                                dtype=torch.uint8, device=dev)
         ext.plain_fixed_decode_batch(
             page_buf, self._up(p_start), self._up(p_end),
             self._up(page_nval.astype(np.int32)), self._up(row0),
             1 if max_def > 0 else 0, 12, 0,
-            empty8, empty64, empty64, out, empty8, status)
+            empty8, empty64, empty64, out, valid_out, status)
         self._check(status, 'int96:' + name)
         m = out[:total * 12].view(total, 12)
         nanos = m[:, :8].contiguous().view(torch.int64).reshape(total)
         day = m[:, 8:12].contiguous().view(torch.int32).reshape(total) \
             .to(torch.int64)
-        return (day - 2440588) * 86_400_000_000_000 + nanos
+        ns = (day - 2440588) * 86_400_000_000_000 + nanos
+        if max_def > 0:
+            # null rows -> int64 min == numpy NaT after the boundary's
+            # datetime64 view (CPU-route parity)
+            ns = torch.where(valid_out.bool(), ns,
+                             torch.tensor(np.iinfo(np.int64).min,
+                                          dtype=torch.int64, device=dev))
+        return ns
 
     def _bool_plain(self, ext, dev, page_buf, val_start, page_nval):
         """PLAIN BOOLEAN: bit-packed LSB-first -> bool tensor."""

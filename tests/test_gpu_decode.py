@@ -1760,3 +1760,37 @@ def test_corrupt_delta_page_fails_loudly_gpu(ext, tmp_path):
         with make_batch_reader('file://' + str(d), device='cuda',
                                shuffle_row_groups=False) as r:
             list(r)
+
+
+def test_int96_nullable_nat_gpu(ext, tmp_path):
+    """Nullable INT96 with actual nulls: null rows surface as NaT (CPU
+    route parity)."""
+    import datetime
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'int96n'
+    d.mkdir()
+    n = 200
+    base = datetime.datetime(2015, 6, 1)
+    stamps = [None if i % 7 == 0 else
+              base + datetime.timedelta(minutes=i) for i in range(n)]
+    table = pa.table({'id': pa.array(np.arange(n, dtype=np.int64)),
+                      'ts': pa.array(stamps, pa.timestamp('us'))})
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=50,
+                   use_dictionary=False, compression='snappy',
+                   use_deprecated_int96_timestamps=True)
+    got = {}
+    with make_batch_reader('file://' + str(d), device='cuda',
+                           shuffle_row_groups=False) as r:
+        for b in r:
+            ids = b.id.cpu().numpy()
+            ts = np.asarray(b.ts)
+            for i, rid in enumerate(ids):
+                got[int(rid)] = ts[i]
+        assert 'ts' not in r.diagnostics['cpu_assist_columns']
+    for i in range(n):
+        if stamps[i] is None:
+            assert np.isnat(got[i]), (i, got[i])
+        else:
+            assert np.datetime64(got[i], 'us') == np.datetime64(stamps[i])
