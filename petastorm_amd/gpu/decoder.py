@@ -539,9 +539,11 @@ class GpuRowGroupDecoder(object):
                              dtype=np.int64)
             row0 = np.zeros(len(data_idx), dtype=np.int64)
             row0[1:] = np.cumsum(page_nval)[:-1]
+            fill = self._fill_for(schema, ch['name'], phys)
             return self._plain_fixed_fused(ext, dev, page_buf, p_start,
                                            p_end, page_nval, row0, max_def,
-                                           n_rows, phys, ch['name'])
+                                           n_rows, phys, ch['name'],
+                                           fill=fill)
         if data_enc == _ENC_PLAIN and phys == 'INT96':
             sizes = uncomp_size if snappy else comp_size
             page_nval = num_values[data_idx]
@@ -867,8 +869,21 @@ class GpuRowGroupDecoder(object):
         'INT32': 0, 'INT64': 0,
     }
 
+    def _fill_for(self, schema, name, phys):
+        """Null-slot fill pattern: NaN for floats; the int-min NaT
+        sentinel for datetime fields (boundary maps it to NaT); 0
+        otherwise."""
+        field = schema.fields.get(name) if schema is not None else None
+        if field is not None and field.numpy_dtype is np.datetime64:
+            return {'INT32': 0x80000000,
+                    'INT64': -0x8000000000000000}.get(phys,
+                                                     self._FILL_PATTERNS
+                                                     .get(phys, 0))
+        return self._FILL_PATTERNS.get(phys, 0)
+
     def _plain_fixed_fused(self, ext, dev, page_buf, p_start, p_end,
-                           page_nval, row0, max_def, n_rows, phys, name):
+                           page_nval, row0, max_def, n_rows, phys, name,
+                           fill=None):
         dtype, esize = _PHYS_TO_TORCH[phys]
         total = int(page_nval.sum())
         out = torch.empty(total * esize + _SLACK, dtype=torch.uint8,
@@ -879,7 +894,8 @@ class GpuRowGroupDecoder(object):
         ext.plain_fixed_decode_batch(
             page_buf, self._up(p_start), self._up(p_end),
             self._up(page_nval.astype(np.int32)), self._up(row0),
-            1 if max_def > 0 else 0, esize, self._FILL_PATTERNS[phys],
+            1 if max_def > 0 else 0, esize,
+            self._FILL_PATTERNS[phys] if fill is None else fill,
             empty8, empty64, empty64,
             out, empty8, status)
         self._check(status, 'plainfixed:' + name)
@@ -988,7 +1004,7 @@ class GpuRowGroupDecoder(object):
             ext.plain_fixed_decode_batch(
                 val_buf, self._up(val_start), self._up(val_end),
                 self._up(page_nval.astype(np.int32)), self._up(row0),
-                has_def, esize, self._FILL_PATTERNS[phys],
+                has_def, esize, self._fill_for(schema, ch['name'], phys),
                 dbuf, self._up(lev_start), self._up(lev_len),
                 out, empty8, status2)
             self._check(status2, 'v2-plainfixed:' + ch['name'])

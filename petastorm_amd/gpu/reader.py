@@ -568,10 +568,14 @@ class GpuBatchReader(object):
                         # DATE/TIMESTAMP: physical ints decoded on device;
                         # materialize numpy datetime64 at the boundary
                         # (torch has no datetime dtype — CPU-route type
-                        # parity, reference unischema.py:467-502 mapping)
+                        # parity, reference unischema.py:467-502 mapping).
+                        # Null slots carry the int-min sentinel -> NaT.
                         unit = self._datetime_unit(piece.path, name) or 'us'
-                        col = col.cpu().numpy().astype(
-                            'datetime64[{}]'.format(unit))
+                        arr = col.cpu().numpy()
+                        nat = arr == np.iinfo(arr.dtype).min
+                        col = arr.astype('datetime64[{}]'.format(unit))
+                        if nat.any():
+                            col[nat] = np.datetime64('NaT')
                     # unsigned logical types over signed physical storage:
                     # same widening convention as the ndarray codec path
                     # (reference pytorch.py:40-70 sanitization)

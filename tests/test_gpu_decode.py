@@ -1794,3 +1794,44 @@ def test_int96_nullable_nat_gpu(ext, tmp_path):
             assert np.isnat(got[i]), (i, got[i])
         else:
             assert np.datetime64(got[i], 'us') == np.datetime64(stamps[i])
+
+
+def test_nullable_datetime_nat_gpu(ext, tmp_path):
+    """Nullable DATE/TIMESTAMP (INT32/INT64 physical) with actual nulls:
+    NaT on the GPU route (sentinel fill + boundary conversion)."""
+    import datetime
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader
+    d = tmp_path / 'dtnull'
+    d.mkdir()
+    n = 120
+    dates = [None if i % 5 == 0 else
+             datetime.date(2020, 1, 1) + datetime.timedelta(days=i)
+             for i in range(n)]
+    stamps = [None if i % 4 == 0 else
+              datetime.datetime(2021, 2, 3) + datetime.timedelta(seconds=i)
+              for i in range(n)]
+    table = pa.table({'id': pa.array(np.arange(n, dtype=np.int64)),
+                      'd': pa.array(dates, pa.date32()),
+                      'ts': pa.array(stamps, pa.timestamp('us'))})
+    pq.write_table(table, str(d / 'p.parquet'), row_group_size=40,
+                   use_dictionary=False, compression='snappy')
+    got = {}
+    with make_batch_reader('file://' + str(d), device='cuda',
+                           shuffle_row_groups=False) as r:
+        for b in r:
+            ids = b.id.cpu().numpy()
+            for i, rid in enumerate(ids):
+                got[int(rid)] = (np.asarray(b.d)[i], np.asarray(b.ts)[i])
+        assert r.diagnostics['cpu_assist_columns'] == []
+    for i in range(n):
+        gd, gts = got[i]
+        if dates[i] is None:
+            assert np.isnat(gd)
+        else:
+            assert np.datetime64(gd, 'D') == np.datetime64(dates[i], 'D')
+        if stamps[i] is None:
+            assert np.isnat(gts)
+        else:
+            assert np.datetime64(gts, 'us') == np.datetime64(stamps[i])
