@@ -30,7 +30,6 @@ import os
 import numpy as np
 import torch
 
-from petastorm_amd import codecs as _codecs
 from petastorm_amd import ops
 from petastorm_amd.codecs import (CompressedImageCodec, NdarrayCodec)
 

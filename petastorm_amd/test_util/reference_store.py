@@ -13,7 +13,6 @@ import io
 import pickle
 import sys
 import types
-import zlib
 from collections import OrderedDict, namedtuple
 
 import numpy as np

@@ -29,9 +29,7 @@ import time
 import traceback
 
 from petastorm_amd.workers_pool import (EmptyResultError,
-                                        TimeoutWaitingForResultError,
-                                        VentilatedItemProcessedMessage,
-                                        WorkerExceptionMessage)
+                                        TimeoutWaitingForResultError)
 from petastorm_amd.reader_impl.serializers import PickleSerializer
 
 _STOP = '__petastorm_amd_stop__'
