@@ -47,3 +47,30 @@ def test_disk_cache_cleanup(tmp_path):
     assert os.path.exists(path)
     c.cleanup()
     assert not os.path.exists(path)
+
+
+# ---------------------------------------------------------------------------
+# HbmCache eviction logic (CPU-tensor unit tests; GPU usage covered by
+# tests/test_gpu_decode.py::test_gpu_batch_reader_hbm_cache_and_epochs)
+# ---------------------------------------------------------------------------
+
+def test_hbm_cache_lru_eviction_and_budget():
+    import torch
+    from petastorm_amd.gpu.hbm_cache import HbmCache
+    one_mb = torch.zeros(1 << 20, dtype=torch.uint8)
+    cache = HbmCache(size_limit_bytes=3 << 20)
+    for k in 'abc':
+        cache.get(k, lambda: {'x': one_mb.clone()})
+    assert cache.size_bytes == 3 << 20
+    cache.get('a', lambda: (_ for _ in ()).throw(AssertionError('hit!')))
+    # inserting d evicts the LRU entry: 'b' (a was refreshed)
+    cache.get('d', lambda: {'x': one_mb.clone()})
+    assert 'b' not in cache._store and 'a' in cache._store
+    assert cache.size_bytes == 3 << 20
+    # value over budget: served but never cached
+    big = {'x': torch.zeros(4 << 20, dtype=torch.uint8)}
+    cache.get('huge', lambda: big)
+    assert 'huge' not in cache._store
+    assert cache.hits == 1 and cache.misses == 5
+    cache.cleanup()
+    assert cache.size_bytes == 0
