@@ -315,6 +315,9 @@ class Reader(object):
             random_seed=seed,
             max_ventilation_queue_size=max_q)
         self._ventilator._ventilate_fn = self._workers_pool.ventilate
+        # hold ventilation until first consumption so load_state_dict can
+        # fast-forward the item cursor before any work flows
+        self._ventilator.hold()
         self._workers_pool.start(worker_class, worker_args,
                                  ventilator=self._ventilator)
         self._cache = cache
@@ -325,6 +328,16 @@ class Reader(object):
         self._any_shuffle = bool(shuffle_row_groups or shuffle_rows or
                                  (shuffle_row_drop_partitions or 1) > 1)
         self._rows_consumed = 0
+        self._consumption_started = False
+        self._num_epochs = num_epochs
+        self._shuffle_row_groups_flag = bool(shuffle_row_groups)
+        # per-item row counts in VENTILATION ORDER (the O(1) cursor math;
+        # exact only without predicate / row-drop / ngram)
+        self._item_rows = [self._pieces[it['piece_index']].num_rows
+                           for it in items]
+        self._fast_skip_ok = (predicate is None and
+                              (shuffle_row_drop_partitions or 1) == 1 and
+                              self.ngram is None)
 
     # ------------------------------------------------------------------
     def _push_down_partition_predicate(self, predicate, selected):
@@ -403,6 +416,9 @@ class Reader(object):
         """reference :708-718"""
         if self._stopped:
             raise StopIteration
+        if not self._consumption_started:
+            self._consumption_started = True
+            self._ventilator.release()
         try:
             if self.batched_output:
                 columns = self._workers_pool.get_results()
@@ -453,16 +469,64 @@ class Reader(object):
 
     def load_state_dict(self, state):
         """Fast-forward a FRESH reader (same constructor arguments) to a
-        :meth:`state_dict` position by consuming and discarding.  Costs the
-        re-decode of the skipped span — the price of exactness with
-        worker-pool parallelism; prefer the GPU reader's O(1) cursor resume
-        for large skips."""
+        :meth:`state_dict` position.
+
+        Plain configurations (no predicate, no shuffle_row_drop, no
+        NGram) restore with an O(1) ITEM CURSOR: whole row groups before
+        the checkpoint are never ventilated, decoded or read — only the
+        partially-consumed row group (row path) is decoded once to
+        discard its leading rows.  Other configurations fall back to
+        deterministic consume-and-discard replay."""
         self._check_deterministic()
-        if self._rows_consumed:
+        if self._rows_consumed or self._consumption_started:
             raise RuntimeError('load_state_dict requires a fresh reader')
         if state.get('seed') != self._seed:
             raise ValueError('state was captured with a different seed')
         target = int(state['rows_consumed'])
+        if target and self._fast_skip_ok:
+            if self.batched_output:
+                # units are row-group batches: one per ventilated item
+                per_epoch = len(self._item_rows)
+                epoch = target // per_epoch if per_epoch else 0
+                skip_items, rem = (target % per_epoch if per_epoch else 0,
+                                   0)
+            else:
+                per_epoch = sum(self._item_rows)
+                epoch = target // per_epoch if per_epoch else 0
+                rem_rows = target % per_epoch if per_epoch else 0
+                order = list(range(len(self._item_rows)))
+                if self._shuffle_row_groups_flag:
+                    # EXACTLY the ventilator's per-epoch permutation
+                    import random as _random
+                    _random.Random(
+                        None if self._seed is None
+                        else self._seed + epoch).shuffle(order)
+                skip_items = 0
+                for idx in order:
+                    n = self._item_rows[idx]
+                    if rem_rows < n:
+                        break
+                    rem_rows -= n
+                    skip_items += 1
+                rem = rem_rows
+            if self._num_epochs is not None and \
+                    epoch >= self._num_epochs:
+                # checkpoint at/after the end: nothing left to read
+                self._ventilator.fast_forward(self._num_epochs, 0)
+                self._rows_consumed = target
+                self._ventilator.release()
+                self._consumption_started = True
+                return self
+            self._ventilator.fast_forward(epoch, skip_items)
+            self._rows_consumed = target - rem
+            for _ in range(rem):
+                try:
+                    next(self)
+                except StopIteration:
+                    break
+            self._consumption_started = True
+            self._ventilator.release()
+            return self
         for _ in range(target):
             try:
                 next(self)

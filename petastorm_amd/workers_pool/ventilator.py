@@ -55,9 +55,17 @@ class ConcurrentVentilator(Ventilator):
         self._completed = threading.Event()
         self._thread = None
         self._epoch = 0
+        # deferred-start support (reader checkpoint fast-forward): while
+        # held, start() only records the request; release() performs it
+        self._hold = False
+        self._start_requested = False
+        self._skip_once = 0
 
     # ------------------------------------------------------------------
     def start(self):
+        if self._hold:
+            self._start_requested = True
+            return
         if self._thread is not None and self._thread.is_alive():
             raise RuntimeError('Ventilator is already running')
         self._stop_event.clear()
@@ -65,6 +73,32 @@ class ConcurrentVentilator(Ventilator):
         self._thread = threading.Thread(target=self._ventilate, daemon=True,
                                         name='petastorm-amd-ventilator')
         self._thread.start()
+
+    def hold(self):
+        """Defer the next start() until release() — lets a reader position
+        the ventilator (fast_forward) before any item flows."""
+        self._hold = True
+
+    def release(self):
+        self._hold = False
+        if self._start_requested:
+            self._start_requested = False
+            self.start()
+
+    def fast_forward(self, epoch, skip_items):
+        """Position BEFORE any ventilation: begin at ``epoch`` (the
+        per-epoch permutation uses random_seed+epoch, and the remaining
+        iteration budget shrinks accordingly) and skip the first
+        ``skip_items`` items of that epoch.  Only valid while held /
+        not yet running."""
+        if self._thread is not None and self._thread.is_alive():
+            raise RuntimeError('fast_forward requires a not-yet-running '
+                               'ventilator')
+        self._epoch = epoch
+        if self._iterations_remaining is not None:
+            self._iterations_remaining = max(
+                0, self._iterations_orig - epoch)
+        self._skip_once = int(skip_items)
 
     def _ventilate(self):
         while not self._stop_event.is_set():
@@ -76,6 +110,9 @@ class ConcurrentVentilator(Ventilator):
                     None if self._random_seed is None
                     else self._random_seed + self._epoch)
                 rng.shuffle(items)
+            if self._skip_once:
+                items = items[self._skip_once:]
+                self._skip_once = 0
             for item in items:
                 # backpressure (reference ventilator.py:155-157)
                 with self._in_flight_cv:
