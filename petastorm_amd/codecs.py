@@ -197,13 +197,24 @@ class CompressedNdarrayCodec(DataframeColumnCodec):
     identical.
     """
 
-    def __init__(self, level=6):
+    def __init__(self, level=6, container='zlib'):
+        """``container='zlib'`` (default) writes this framework's raw
+        zlib(npy) framing; ``container='npz'`` writes np.savez_compressed
+        exactly like upstream petastorm (codecs.py:193) for stores that
+        must remain readable by BOTH frameworks.  Decode accepts either
+        transparently."""
+        if container not in ('zlib', 'npz'):
+            raise ValueError('container must be zlib or npz')
         self.level = level
+        self.container = container
 
     def encode(self, unischema_field, value):
         _check_dtype(unischema_field, value)
         _check_shape_compliance(unischema_field, value)
         memfile = io.BytesIO()
+        if self.container == 'npz':
+            np.savez_compressed(memfile, value)
+            return memfile.getvalue()
         np.save(memfile, value)
         return zlib.compress(memfile.getvalue(), self.level)
 
@@ -223,7 +234,8 @@ class CompressedNdarrayCodec(DataframeColumnCodec):
         return pa.binary()
 
     def to_dict(self):
-        return {'type': 'CompressedNdarrayCodec', 'level': self.level}
+        return {'type': 'CompressedNdarrayCodec', 'level': self.level,
+                'container': self.container}
 
 
 class CompressedImageCodec(DataframeColumnCodec):
@@ -308,7 +320,8 @@ class CompressedImageCodec(DataframeColumnCodec):
 _CODEC_REGISTRY = {
     'ScalarCodec': lambda d: ScalarCodec(),
     'NdarrayCodec': lambda d: NdarrayCodec(),
-    'CompressedNdarrayCodec': lambda d: CompressedNdarrayCodec(d.get('level', 6)),
+    'CompressedNdarrayCodec': lambda d: CompressedNdarrayCodec(
+        d.get('level', 6), d.get('container', 'zlib')),
     'CompressedImageCodec': lambda d: CompressedImageCodec(
         d.get('image_codec', 'png'), d.get('quality', 80)),
 }

@@ -117,3 +117,25 @@ def test_codec_json_roundtrip():
     assert codec_from_dict(None) is None
     assert codec_from_dict(codec_to_dict(
         CompressedImageCodec('jpeg', 75))).quality == 75
+
+
+def test_compressed_ndarray_npz_container_upstream_compatible():
+    """container='npz' writes np.savez_compressed payloads byte-compatible
+    with upstream petastorm's CompressedNdarrayCodec; schema JSON keeps
+    the container through a round trip."""
+    import io
+    import numpy as np
+    from petastorm_amd.codecs import (CompressedNdarrayCodec, codec_from_dict)
+    from petastorm_amd.unischema import UnischemaField
+    arr = np.arange(24, dtype=np.int32).reshape(4, 6)
+    field = UnischemaField('m', np.int32, (4, 6),
+                           CompressedNdarrayCodec(container='npz'), False)
+    blob = field.codec.encode(field, arr)
+    assert blob[:2] == b'PK'  # npz (zip) container, as upstream writes
+    with np.load(io.BytesIO(blob), allow_pickle=False) as npz:
+        np.testing.assert_array_equal(npz[npz.files[0]], arr)
+    np.testing.assert_array_equal(field.codec.decode(field, blob), arr)
+    back = codec_from_dict(field.codec.to_dict())
+    assert back.container == 'npz'
+    np.testing.assert_array_equal(back.decode(field, back.encode(field, arr)),
+                                  arr)
