@@ -187,6 +187,7 @@ def test_examples_run(tmp_path):
             ('examples/hello_world/main.py',
              ['file://' + str(tmp_path / 'hw_ex')]),
             ('examples/hello_world/external_dataset.py', []),
+            ('examples/mnist/main.py', []),
     ):
         out = subprocess.run([sys.executable, os.path.join(root, script)]
                              + args, capture_output=True, text=True,
