@@ -115,3 +115,33 @@ def get_filesystem_and_path_or_paths(url_or_urls, storage_options=None,
     if isinstance(url_or_urls, list):
         return fs, paths
     return fs, paths[0]
+
+
+def get_dataset_path(parsed_url):
+    """Filesystem-facing path of a parsed dataset URL: s3-like filesystems
+    want the bucket inside the path (reference fs_utils.py:28-38)."""
+    if (parsed_url.scheme or 'file').lower() in ('file', 'hdfs'):
+        return parsed_url.path
+    return parsed_url.netloc + parsed_url.path
+
+
+class FilesystemResolver(object):
+    """Drop-in-shaped resolver over the fsspec standardization
+    (reference fs_utils.py:41-177 resolves via pyarrow/libhdfs dispatch;
+    here every scheme goes through fsspec + RetryingFilesystem — see the
+    module docstring).  Provides the reference's accessor surface."""
+
+    def __init__(self, dataset_url, storage_options=None, **_compat_kwargs):
+        self._dataset_url = normalize_dir_url(dataset_url)
+        self._parsed = urlparse(self._dataset_url)
+        self._filesystem, self._path = get_filesystem_and_path_or_paths(
+            self._dataset_url, storage_options)
+
+    def filesystem(self):
+        return self._filesystem
+
+    def get_dataset_path(self):
+        return self._path
+
+    def parsed_dataset_url(self):
+        return self._parsed

@@ -258,6 +258,18 @@ def vector_to_array(column, dtype='float64'):
     return _VectorToArray(column, dtype)
 
 
+class Row(dict):
+    """Minimal pyspark.sql.Row: keyword construction, attribute access,
+    alphabetical field order preserved by the caller."""
+
+    def __init__(self, **kwargs):
+        super(Row, self).__init__(kwargs)
+        self.__dict__.update(kwargs)
+
+    def asDict(self):
+        return dict(self)
+
+
 def build_modules():
     """Return {module_name: module} shaped like the pyspark package tree."""
     pyspark = types.ModuleType('pyspark')
@@ -276,6 +288,7 @@ def build_modules():
     ml_linalg.DenseVector = DenseVector
     ml_functions.vector_to_array = vector_to_array
 
+    sql.Row = Row
     pyspark.sql = sql
     sql.types = sql_types
     sql.functions = sql_functions

@@ -214,3 +214,10 @@ def make_petastorm_dataset(reader):
         return nt(*out)
 
     return dataset.map(set_shapes)
+
+
+def date_to_nsec_from_epoch(dt):
+    """Seconds-resolution datetime/date -> int64 ns since epoch
+    (reference tf_utils.py:50-54)."""
+    from calendar import timegm
+    return timegm(dt.timetuple()) * 1_000_000_000

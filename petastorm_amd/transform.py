@@ -143,3 +143,9 @@ def transform_schema(schema, transform_spec):
         fields = {k: v for k, v in fields.items()
                   if k in set(transform_spec.selected_fields)}
     return Unischema(schema._name + '_transformed', list(fields.values()))
+
+
+def edit_field(name, numpy_dtype, shape, nullable=False):
+    """Helper building one ``edit_fields`` entry (reference
+    transform.py:19-25)."""
+    return (name, numpy_dtype, shape, nullable)

@@ -416,3 +416,16 @@ def dict_to_encoded_row(unischema, row_dict):
         else:
             encoded[name] = _codecs.effective_codec(field).encode(field, value)
     return encoded
+
+
+def dict_to_spark_row(unischema, row_dict):
+    """Encode a row dict into a ``pyspark.sql.Row`` for Spark-side dataset
+    writes (reference unischema.py:359-406: codecs encode each field,
+    explicit nulls are inserted, and Row fields sort alphabetically —
+    Spark matches schema to Row by POSITION, reference's warning at
+    :225-227).  Requires pyspark."""
+    from collections import OrderedDict
+
+    from pyspark.sql import Row
+    encoded = dict_to_encoded_row(unischema, row_dict)
+    return Row(**OrderedDict(sorted(encoded.items(), key=lambda kv: kv[0])))

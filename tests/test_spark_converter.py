@@ -160,3 +160,23 @@ def test_dataset_as_rdd(spark, tmp_path):
     for src in rows:
         np.testing.assert_allclose(by_id[int(src['id'])].matrix,
                                    src['matrix'], rtol=1e-6)
+
+
+def test_dict_to_spark_row(spark):
+    """Write-path compat: dict_to_spark_row encodes through the codecs and
+    yields an alphabetically-ordered Row (reference unischema.py:359-406)."""
+    from petastorm_amd.codecs import NdarrayCodec, ScalarCodec
+    from petastorm_amd.unischema import (Unischema, UnischemaField,
+                                         dict_to_spark_row)
+    schema = Unischema('R', [
+        UnischemaField('zz', np.int64, (), ScalarCodec(), False),
+        UnischemaField('aa', np.float32, (3,), NdarrayCodec(), False),
+        UnischemaField('mm', np.str_, (), ScalarCodec(), True),
+    ])
+    row = dict_to_spark_row(schema, {'zz': np.int64(7),
+                                     'aa': np.ones(3, np.float32)})
+    assert list(row.asDict()) == ['aa', 'mm', 'zz']  # alphabetical
+    assert row.zz == 7 and row.mm is None
+    assert isinstance(row.aa, bytes)  # npy-encoded by the codec
+    back = np.load(__import__('io').BytesIO(row.aa))
+    np.testing.assert_array_equal(back, np.ones(3, np.float32))
