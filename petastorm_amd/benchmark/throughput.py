@@ -19,6 +19,9 @@ BenchmarkResult = namedtuple('BenchmarkResult',
                              ['time_mean', 'samples_per_second',
                               'memory_info', 'cpu'])
 
+#: Reference-name parity (throughput.py): read_method values.
+ReadMethod = namedtuple('ReadMethods', ['PYTHON', 'TF'])('python', 'tf')
+
 WorkerPoolType = namedtuple('WorkerPoolTypes', ['THREAD', 'PROCESS', 'NONE'])(
     'thread', 'process', 'dummy')
 

@@ -30,3 +30,12 @@ class GpuExtensionNotAvailable(PetastormAmdError):
     The HIP extension must fail loudly on a GPU box rather than silently
     falling back to an eager/CPU path.
     """
+
+
+class PetastormMetadataError(PetastormAmdError):
+    """Missing/invalid dataset metadata (reference etl/dataset_metadata.py
+    exception of the same name)."""
+
+
+class PetastormMetadataGenerationError(PetastormAmdError):
+    """Metadata (re)generation failed (reference name parity)."""

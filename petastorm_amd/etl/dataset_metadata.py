@@ -24,6 +24,8 @@ import posixpath
 from collections import namedtuple
 from contextlib import contextmanager
 
+from petastorm_amd.errors import (PetastormMetadataError,  # noqa: F401
+                                  PetastormMetadataGenerationError)
 from petastorm_amd.fs_utils import get_filesystem_and_path_or_paths
 from petastorm_amd.unischema import Unischema, dict_to_encoded_row
 

@@ -46,3 +46,7 @@ def main(args=None):
 
 if __name__ == '__main__':
     sys.exit(main())
+
+
+#: Reference-name alias (petastorm_generate_metadata.py:47).
+generate_petastorm_metadata = generate_metadata

@@ -182,3 +182,15 @@ def load_rowgroup_indexes(fs, path_or_paths):
                 out[name] = _INDEXER_TYPES[d['type']].from_dict(d)
             return out
     return {}
+
+
+def get_row_group_indexes(dataset_url_or_fs, path_or_paths=None):
+    """Load the dataset's persisted rowgroup indexes (reference
+    rowgroup_indexing.py:136-158).  Accepts a URL, or (fs, path)."""
+    if path_or_paths is None:
+        from petastorm_amd.fs_utils import get_filesystem_and_path_or_paths
+        fs, path_or_paths = get_filesystem_and_path_or_paths(
+            dataset_url_or_fs)
+    else:
+        fs = dataset_url_or_fs
+    return load_rowgroup_indexes(fs, path_or_paths)
