@@ -181,9 +181,8 @@ class SparkDatasetConverter(object):
 
     def delete(self):
         """Delete the materialized files (reference :287-294)."""
-        fs, paths = get_filesystem_and_path_or_paths(self.cache_dir_url)
         try:
-            fs.rm(paths, recursive=True)
+            _delete_dir_handler(self.cache_dir_url)
         except Exception:  # noqa: BLE001 - best-effort cleanup
             logger.warning('Failed to delete cache dir %s',
                            self.cache_dir_url, exc_info=True)
@@ -250,10 +249,26 @@ def _materialize_df(df, parent_cache_dir_url, row_group_size_mb,
     return subdir, sorted(files)
 
 
+def _default_delete_dir_handler(url):
+    fs, path = get_filesystem_and_path_or_paths(url)
+    fs.rm(path, recursive=True)
+
+
+_delete_dir_handler = _default_delete_dir_handler
+
+
+def register_delete_dir_handler(handler):
+    """Plug a custom directory-delete function for materialization
+    cleanup (reference spark_dataset_converter.py:102-114; Databricks
+    environments override this).  ``None`` restores the default."""
+    global _delete_dir_handler
+    _delete_dir_handler = handler if handler is not None \
+        else _default_delete_dir_handler
+
+
 def _best_effort_delete(url):
     try:
-        fs, path = get_filesystem_and_path_or_paths(url)
-        fs.rm(path, recursive=True)
+        _delete_dir_handler(url)
     except Exception:  # noqa: BLE001
         pass
 

@@ -180,3 +180,16 @@ def test_dict_to_spark_row(spark):
     assert isinstance(row.aa, bytes)  # npy-encoded by the codec
     back = np.load(__import__('io').BytesIO(row.aa))
     np.testing.assert_array_equal(back, np.ones(3, np.float32))
+
+
+def test_register_delete_dir_handler(spark):
+    """Custom delete handlers plug into materialization cleanup
+    (reference :102-114)."""
+    deleted = []
+    sdc.register_delete_dir_handler(lambda url: deleted.append(url))
+    try:
+        conv = sdc.make_spark_converter(_make_df(spark, source_id='del'))
+        conv.delete()
+        assert deleted == [conv.cache_dir_url]
+    finally:
+        sdc.register_delete_dir_handler(None)
