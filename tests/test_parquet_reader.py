@@ -185,3 +185,21 @@ def test_many_columns_store(tmp_path):
                                           'col_999']) as r:
         b = next(iter(r))
         assert sorted(b._fields) == ['col_1', 'col_42', 'col_999']
+
+
+def test_weighted_sampling_over_batch_readers(scalar_dataset, tmp_path):
+    """WeightedSamplingReader mixes batched readers too (reference
+    weighted_sampling_reader.py checks batched_output compatibility)."""
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.weighted_sampling_reader import WeightedSamplingReader
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url2 = 'file://' + str(tmp_path / 'mix2')
+    create_scalar_dataset(url2, num_rows=400, rowgroup_size=100, seed=9)
+    r1 = make_batch_reader(scalar_dataset['url'], num_epochs=None,
+                           schema_fields=['id'], shuffle_row_groups=False)
+    r2 = make_batch_reader(url2, num_epochs=None, schema_fields=['id'],
+                           shuffle_row_groups=False)
+    mixed = WeightedSamplingReader([r1, r2], [0.5, 0.5])
+    batches = [next(mixed) for _ in range(20)]
+    assert all(hasattr(b, 'id') for b in batches)
+    r1.stop(); r1.join(); r2.stop(); r2.join()
