@@ -151,6 +151,16 @@ def test_too_many_shards_raises(test_dataset):
                     cur_shard=0, shard_count=10000)
 
 
+@pytest.mark.parametrize('pool', POOLS)
+def test_num_epochs_pools(test_dataset, pool):
+    with make_reader(test_dataset['url'], reader_pool_type=pool,
+                     num_epochs=3, shuffle_row_groups=False,
+                     schema_fields=['id']) as r:
+        ids = [int(row.id) for row in r]
+    assert sorted(ids) == sorted(
+        [int(x['id']) for x in test_dataset['rows']] * 3)
+
+
 def test_num_epochs(test_dataset):
     with make_reader(test_dataset['url'], reader_pool_type='thread',
                      num_epochs=3, shuffle_row_groups=False) as r:
@@ -166,6 +176,18 @@ def test_reset_after_exhaustion(test_dataset):
         r.reset()
         second = [int(row.id) for row in r]
     assert sorted(first) == sorted(second)
+
+
+@pytest.mark.parametrize('pool', POOLS)
+def test_transform_spec_row_pools(test_dataset, pool):
+    from petastorm_amd.transform import TransformSpec
+    ts = TransformSpec(lambda row: dict(row, id=row['id'] * 10),
+                       edit_fields=[('id', np.int64, (), False)])
+    with make_reader(test_dataset['url'], reader_pool_type=pool,
+                     schema_fields=['id'], transform_spec=ts,
+                     shuffle_row_groups=False) as r:
+        ids = sorted(int(row.id) for row in r)
+    assert ids == sorted(int(x['id']) * 10 for x in test_dataset['rows'])
 
 
 def test_transform_spec_row(test_dataset):
