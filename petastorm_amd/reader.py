@@ -332,12 +332,17 @@ class Reader(object):
         self._num_epochs = num_epochs
         self._shuffle_row_groups_flag = bool(shuffle_row_groups)
         # per-item row counts in VENTILATION ORDER (the O(1) cursor math;
-        # exact only without predicate / row-drop / ngram)
+        # exact only when nothing can change per-group row counts: no
+        # predicate, no row-drop, no ngram, and no transform func — a
+        # TransformSpec func may filter rows, invalidating the metadata
+        # counts.  Anything else falls back to consume-and-discard replay.
         self._item_rows = [self._pieces[it['piece_index']].num_rows
                            for it in items]
         self._fast_skip_ok = (predicate is None and
                               (shuffle_row_drop_partitions or 1) == 1 and
-                              self.ngram is None)
+                              self.ngram is None and
+                              (transform_spec is None or
+                               transform_spec.func is None))
 
     # ------------------------------------------------------------------
     def _push_down_partition_predicate(self, predicate, selected):

@@ -78,7 +78,10 @@ class RowReaderWorker(WorkerBase):
                                       shuffle_row_drop_partition)
         rows = [decode_row(r, self._a.view_schema) for r in rows]
         if self._a.transform_spec is not None and self._a.transform_spec.func:
-            rows = [self._a.transform_spec.func(r) for r in rows]
+            # a func may return None to drop the row (the row-path analog
+            # of the batch path's row-filtering transforms)
+            rows = [t for t in (self._a.transform_spec.func(r)
+                                for r in rows) if t is not None]
         if self._a.transform_spec is not None:
             keep = set(self._a.transformed_schema.fields.keys())
             rows = [{k: v for k, v in r.items() if k in keep} for r in rows]
