@@ -116,10 +116,36 @@ def _spark_type_shim(name):
     return _spark_type_shim_cache[name]
 
 
-# Modules whose symbols load for real (reference etl/legacy.py:22-31 minus
-# petastorm/pyspark, which resolve to shims here instead).
-_SAFE_REAL_MODULES = {'collections', 'numpy', 'decimal', 'builtins',
-                      'copy_reg', 'copyreg', '__builtin__'}
+# Symbols that load for real, by package (reference etl/legacy.py:22-31
+# minus petastorm/pyspark, which resolve to shims here instead).  Per-NAME
+# allow-lists: admitting a whole module would admit e.g. ``builtins.eval``
+# and make the "restricted" unpickler a code-execution vector.
+_SAFE_BUILTINS = {'object', 'set', 'frozenset', 'list', 'dict', 'tuple',
+                  'bytes', 'bytearray', 'str', 'int', 'float', 'complex',
+                  'bool', 'slice', 'range', 'type', 'NoneType'}
+# numpy surface a pickled Unischema can reference: the dtype machinery
+# (numpy.dtype / numpy.core.multiarray reconstruction helpers) and the
+# scalar-type CLASSES fields carry as numpy_dtype.
+_SAFE_NUMPY = {'dtype', 'ndarray', 'generic', 'number', '_reconstruct',
+               'scalar', '_frombuffer', '_DType_reconstruct',
+               'bool_', 'object_', 'bytes_', 'str_', 'string_', 'unicode_',
+               'void',
+               'int8', 'int16', 'int32', 'int64',
+               'uint8', 'uint16', 'uint32', 'uint64',
+               'float16', 'float32', 'float64', 'complex64', 'complex128',
+               'datetime64', 'timedelta64',
+               'byte', 'ubyte', 'short', 'ushort', 'intc', 'uintc',
+               'intp', 'uintp', 'longlong', 'ulonglong',
+               'half', 'single', 'double', 'longdouble'}
+_SAFE_GLOBALS = {
+    'collections': {'OrderedDict', 'defaultdict'},
+    'decimal': {'Decimal'},
+    'copyreg': {'_reconstructor', '__newobj__'},
+    'copy_reg': {'_reconstructor', '__newobj__'},
+    'builtins': _SAFE_BUILTINS,
+    '__builtin__': _SAFE_BUILTINS,
+    'numpy': _SAFE_NUMPY,
+}
 
 
 class RestrictedInteropUnpickler(pickle.Unpickler):
@@ -136,7 +162,7 @@ class RestrictedInteropUnpickler(pickle.Unpickler):
                 % (module, name))
         if package == 'pyspark':
             return _spark_type_shim(name)
-        if package in _SAFE_REAL_MODULES:
+        if name in _SAFE_GLOBALS.get(package, ()):
             return super(RestrictedInteropUnpickler, self).find_class(
                 module, name)
         raise pickle.UnpicklingError(

@@ -141,3 +141,18 @@ def test_gpu_batch_reader_on_reference_store(tmp_path):
         np.testing.assert_array_equal(got_img[rid], src['image_png'])
         np.testing.assert_array_equal(got_emb[rid], src['embedding'])
         np.testing.assert_array_equal(got_mat[rid], src['matrix_z'])
+
+def test_dangerous_builtins_rejected():
+    """builtins/os symbols must not resolve even though a few inert
+    builtins (object, set, ...) are allow-listed by name."""
+    for blob in (b'cbuiltins\neval\n.', b'c__builtin__\nexec\n.',
+                 b'cos\nsystem\n.', b'cbuiltins\ngetattr\n.',
+                 b'cnumpy.testing\nassert_equal\n.'):
+        with pytest.raises(pickle.UnpicklingError):
+            interop.restricted_loads(blob)
+    # the allow-listed inert ones still work
+    assert interop.restricted_loads(pickle.dumps({1, 2})) == {1, 2}
+    import numpy as _np
+    assert interop.restricted_loads(pickle.dumps(_np.int32)) is _np.int32
+    assert interop.restricted_loads(
+        pickle.dumps(_np.dtype('float32'))) == _np.dtype('float32')
