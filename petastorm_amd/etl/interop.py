@@ -146,6 +146,12 @@ _SAFE_GLOBALS = {
     '__builtin__': _SAFE_BUILTINS,
     'numpy': _SAFE_NUMPY,
 }
+# numpy-1.x scalar-type aliases removed by numpy 2.0 — upstream stores
+# pickled years ago reference these classes by their old names.
+_NUMPY_RENAMES = {'unicode_': 'str_', 'string_': 'bytes_',
+                  'float_': 'float64', 'bool8': 'bool_',
+                  'object0': 'object_', 'str0': 'str_', 'bytes0': 'bytes_',
+                  'void0': 'void', 'int0': 'intp', 'uint0': 'uintp'}
 
 
 class RestrictedInteropUnpickler(pickle.Unpickler):
@@ -162,6 +168,9 @@ class RestrictedInteropUnpickler(pickle.Unpickler):
                 % (module, name))
         if package == 'pyspark':
             return _spark_type_shim(name)
+        if package == 'numpy' and name in _NUMPY_RENAMES \
+                and not hasattr(np, name):
+            return getattr(np, _NUMPY_RENAMES[name])
         if name in _SAFE_GLOBALS.get(package, ()):
             return super(RestrictedInteropUnpickler, self).find_class(
                 module, name)

@@ -156,3 +156,11 @@ def test_dangerous_builtins_rejected():
     assert interop.restricted_loads(pickle.dumps(_np.int32)) is _np.int32
     assert interop.restricted_loads(
         pickle.dumps(_np.dtype('float32'))) == _np.dtype('float32')
+
+def test_numpy1_legacy_scalar_names():
+    """Stores pickled under numpy 1.x reference removed aliases
+    (numpy.unicode_ / numpy.string_); they must map to modern classes."""
+    import numpy as _np
+    assert interop.restricted_loads(b'cnumpy\nunicode_\n.') is _np.str_
+    assert interop.restricted_loads(b'cnumpy\nstring_\n.') is _np.bytes_
+    assert interop.restricted_loads(b'cnumpy\nfloat_\n.') is _np.float64
