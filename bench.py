@@ -362,28 +362,40 @@ def _run_timed(args, step_fn, device, dist, world):
                 t = t.to(device)
             dist.all_reduce(t, op=dist.ReduceOp.MAX)
             batches_per_step = int(t.item())
-    _barrier(dist)
-    _sync(device)
-    t0 = time.perf_counter()
-    samples = 0
-    for _ in range(args.steps):
-        for _ in range(batches_per_step):
-            samples += step_fn()
-    _sync(device)
-    _barrier(dist)
-    elapsed = time.perf_counter() - t0
-    # max elapsed over ranks (the slowest rank defines job time)
-    if dist is not None:
-        t = torch.tensor([elapsed], dtype=torch.float64)
-        if torch.cuda.is_available():
-            t = t.to(device)
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t.item())
-        s = torch.tensor([samples], dtype=torch.float64)
-        if torch.cuda.is_available():
-            s = s.to(device)
-        dist.all_reduce(s, op=dist.ReduceOp.SUM)
-        samples = int(s.item())
+    # The calibration probe can overestimate per-batch cost under transient
+    # contention, which would undershoot the region; if that happens, scale
+    # batches_per_step up and re-time (every rank sees the same MAX-reduced
+    # elapsed, so they rescale in lock-step).
+    attempts = 0
+    while True:
+        _barrier(dist)
+        _sync(device)
+        t0 = time.perf_counter()
+        samples = 0
+        for _ in range(args.steps):
+            for _ in range(batches_per_step):
+                samples += step_fn()
+        _sync(device)
+        _barrier(dist)
+        elapsed = time.perf_counter() - t0
+        # max elapsed over ranks (the slowest rank defines job time)
+        if dist is not None:
+            t = torch.tensor([elapsed], dtype=torch.float64)
+            if torch.cuda.is_available():
+                t = t.to(device)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            elapsed = float(t.item())
+            s = torch.tensor([samples], dtype=torch.float64)
+            if torch.cuda.is_available():
+                s = s.to(device)
+            dist.all_reduce(s, op=dist.ReduceOp.SUM)
+            samples = int(s.item())
+        attempts += 1
+        if args.min_region <= 0 or elapsed >= args.min_region \
+                or attempts >= 3:
+            break
+        scale = max(2.0, 1.5 * args.min_region / max(elapsed, 1e-9))
+        batches_per_step = int(np.ceil(batches_per_step * scale))
     return {'elapsed_s': elapsed, 'samples': samples,
             'ms_per_step': elapsed * 1000.0 / args.steps,
             'batches_per_step': batches_per_step}
