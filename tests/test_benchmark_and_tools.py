@@ -188,6 +188,7 @@ def test_examples_run(tmp_path):
              ['file://' + str(tmp_path / 'hw_ex')]),
             ('examples/hello_world/external_dataset.py', []),
             ('examples/mnist/main.py', []),
+            ('examples/spark_converter/main.py', ['--rows', '96']),
     ):
         out = subprocess.run([sys.executable, os.path.join(root, script)]
                              + args, capture_output=True, text=True,
