@@ -85,3 +85,25 @@ def test_process_pool_stop_with_unconsumed_results():
     assert time.time() - t0 < 20
     with pytest.raises(EmptyResultError):
         pool.get_results()
+
+
+def test_serializers_roundtrip():
+    """PickleSerializer and ArrowTableSerializer round-trip worker payloads
+    (reference pickle_serializer.py:17-23, arrow_table_serializer.py:22-33)."""
+    import numpy as np
+    import pyarrow as pa
+    from petastorm_amd.reader_impl.serializers import (ArrowTableSerializer,
+                                                       PickleSerializer)
+    rows = [{'a': np.arange(4), 'b': 'text'}, {'a': np.zeros(2), 'b': None}]
+    ps = PickleSerializer()
+    back = ps.deserialize(ps.serialize(rows))
+    assert back[1]['b'] is None
+    np.testing.assert_array_equal(back[0]['a'], rows[0]['a'])
+
+    table = pa.table({'x': pa.array([1, 2, 3], type=pa.int32()),
+                      's': pa.array(['p', 'q', None])})
+    ats = ArrowTableSerializer()
+    data = ats.serialize(table)
+    assert isinstance(data, bytes)
+    t2 = ats.deserialize(data)
+    assert t2.equals(table)
