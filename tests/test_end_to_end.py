@@ -366,3 +366,36 @@ def test_moved_dataset_reads_from_new_location(tmp_path):
                            shuffle_row_groups=False) as r:
         n = sum(len(b.id) for b in r)
     assert n == len(rows)
+
+
+def test_batch_transform_returning_torch_tensors(scalar_dataset):
+    """A batch-path TransformSpec may return torch tensors for columns
+    (reference tests/test_parquet_reader.py TransformSpec-returning-tensors);
+    they flow through the pool, the namedtuple, and BatchedDataLoader."""
+    import torch
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.pytorch import BatchedDataLoader
+    from petastorm_amd.transform import TransformSpec
+
+    def to_tensors(columns):
+        return {'id': torch.as_tensor(np.asarray(columns['id'])),
+                'f0': torch.as_tensor(np.asarray(columns['f0'])) * 2}
+
+    ts = TransformSpec(
+        to_tensors,
+        edit_fields=[UnischemaField('f0', np.float64, (), None, False)],
+        selected_fields=['id', 'f0'])
+    with make_batch_reader(scalar_dataset['url'], reader_pool_type='thread',
+                           shuffle_row_groups=False, transform_spec=ts) as r:
+        loader = BatchedDataLoader(r, batch_size=64)
+        got_ids, got_f0 = [], []
+        for batch in loader:
+            assert isinstance(batch['id'], torch.Tensor)
+            got_ids.append(batch['id'])
+            got_f0.append(batch['f0'])
+    ids = torch.cat(got_ids).numpy()
+    f0 = torch.cat(got_f0).numpy()
+    order = np.argsort(ids)
+    np.testing.assert_array_equal(ids[order], scalar_dataset['cols']['id'])
+    np.testing.assert_array_almost_equal(
+        f0[order], scalar_dataset['cols']['f0'] * 2)
