@@ -44,10 +44,13 @@ struct ImgInfo {
   int64_t scan_start = 0, scan_end = 0;
 };
 
-int build_huff(const uint8_t* bits, const uint8_t* vals, int nvals,
-               HuffTable& t) {
+// Returns false for a non-canonical table (corrupt bits[] would otherwise
+// drive the LUT fill out of bounds: 255 claimed 1-bit codes -> base up to
+// 255<<7 in an int32[256]).
+bool build_huff(const uint8_t* bits, const uint8_t* vals, int nvals,
+                HuffTable& t) {
   // canonical code assignment (ITU T.81 Annex C)
-  uint16_t code = 0;
+  uint32_t code = 0;
   int k = 0;
   int codes_of_len[17];
   for (int l = 1; l <= 16; ++l) codes_of_len[l] = bits[l - 1];
@@ -55,6 +58,7 @@ int build_huff(const uint8_t* bits, const uint8_t* vals, int nvals,
   for (int i = 0; i < 256; ++i) t.lut[i] = -1;
   for (int l = 1; l <= 16; ++l) {
     if (codes_of_len[l]) {
+      if (code + (uint32_t)codes_of_len[l] > (1u << l)) return false;
       t.valptr[l] = k;
       t.mincode[l] = code;
       for (int i = 0; i < codes_of_len[l]; ++i) {
@@ -77,7 +81,7 @@ int build_huff(const uint8_t* bits, const uint8_t* vals, int nvals,
     code <<= 1;
   }
   t.used = true;
-  return k;
+  return k == nvals;
 }
 
 inline uint16_t be16(const uint8_t* p) {
@@ -149,6 +153,10 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
         continue;
       }
       uint16_t seglen = be16(p + pos + 2);
+      // every marker reaching here carries a length field; the payload
+      // must fit in the buffer or the handlers below would read past it
+      TORCH_CHECK(seglen >= 2 && pos + 2 + (int64_t)seglen <= len,
+                  "image ", i, ": segment overruns buffer at ", pos);
       const uint8_t* s = p + pos + 4;
       switch (m) {
         case 0xDB: {  // DQT
@@ -157,6 +165,7 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
             int prec = s[q] >> 4, id = s[q] & 15;
             TORCH_CHECK(prec == 0, "image ", i, ": 16-bit quant tables "
                         "unsupported");
+            TORCH_CHECK(id < 4, "image ", i, ": quant table id ", id);
             std::array<uint16_t, 64> tab;
             for (int k2 = 0; k2 < 64; ++k2) tab[k2] = s[q + 1 + k2];
             auto it = qdedup.find(tab);
@@ -173,15 +182,21 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
           int64_t q = 0;
           while (q + 17 <= seglen - 2) {
             int cls = s[q] >> 4, id = s[q] & 15;
+            TORCH_CHECK(cls < 2 && id < 4,
+                        "image ", i, ": huffman table class/id ", cls,
+                        "/", id);
             const uint8_t* bits = s + q + 1;
             int nvals = 0;
             for (int l = 0; l < 16; ++l) nvals += bits[l];
             TORCH_CHECK(nvals <= 256, "image ", i, ": bad DHT");
+            TORCH_CHECK(q + 17 + nvals <= (int64_t)seglen - 2,
+                        "image ", i, ": DHT values overrun segment");
             std::vector<uint8_t> key(s + q + 1, s + q + 17 + nvals);
             auto it = hdedup.find(key);
             if (it == hdedup.end()) {
               HuffTable t{};
-              build_huff(bits, s + q + 17, nvals, t);
+              TORCH_CHECK(build_huff(bits, s + q + 17, nvals, t),
+                          "image ", i, ": non-canonical huffman table");
               hpool.push_back(t);
               it = hdedup.emplace(std::move(key),
                                   (int)hpool.size() - 1).first;
@@ -195,12 +210,17 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
           break;
         }
         case 0xC0: {  // SOF0 baseline
+          TORCH_CHECK(seglen - 2 >= 6, "image ", i, ": truncated SOF0");
           TORCH_CHECK(s[0] == 8, "image ", i, ": only 8-bit precision");
           im.h = be16(s + 1);
           im.w = be16(s + 3);
           im.ncomp = s[5];
           TORCH_CHECK(im.ncomp == 1 || im.ncomp == 3,
                       "image ", i, ": ", im.ncomp, " components unsupported");
+          TORCH_CHECK(seglen - 2 >= 6 + 3 * im.ncomp,
+                      "image ", i, ": truncated SOF0 components");
+          TORCH_CHECK(im.w > 0 && im.h > 0,
+                      "image ", i, ": zero-sized frame");
           for (int c = 0; c < im.ncomp; ++c) {
             im.comp_h[c] = s[7 + 3 * c] >> 4;
             im.comp_v[c] = s[7 + 3 * c] & 15;
@@ -226,21 +246,27 @@ py::dict jpeg_parse_batch(torch::Tensor buf, torch::Tensor val_off,
                       (int)(m - 0xC0), ") — CPU fallback required");
           break;
         case 0xDD:  // DRI
+          TORCH_CHECK(seglen - 2 >= 2, "image ", i, ": truncated DRI");
           im.restart_interval = be16(s);
           break;
         case 0xDA: {  // SOS
           TORCH_CHECK(got_sof, "image ", i, ": SOS before SOF");
+          TORCH_CHECK(seglen - 2 >= 1, "image ", i, ": truncated SOS");
           int ns = s[0];
           TORCH_CHECK(ns == im.ncomp, "image ", i,
                       ": multi-scan JPEG unsupported");
+          TORCH_CHECK(seglen - 2 >= 1 + 2 * ns,
+                      "image ", i, ": truncated SOS components");
           for (int c = 0; c < ns; ++c) {
             int tid = s[2 + 2 * c];
             int dc_id = tid >> 4, ac_id = tid & 15;
+            TORCH_CHECK(dc_id < 4 && ac_id < 4,
+                        "image ", i, ": scan table id out of range");
             TORCH_CHECK(cur_dc[dc_id] >= 0 && cur_ac[ac_id] >= 0,
                         "image ", i, ": missing huffman table");
             im.comp_dc[c] = cur_dc[dc_id];
             im.comp_ac[c] = cur_ac[ac_id];
-            TORCH_CHECK(cur_q[im.comp_q[c]] >= 0,
+            TORCH_CHECK(im.comp_q[c] < 4 && cur_q[im.comp_q[c]] >= 0,
                         "image ", i, ": missing quant table");
           }
           for (int c = 0; c < im.ncomp; ++c)

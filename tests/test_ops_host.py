@@ -167,3 +167,32 @@ def test_walk_pages_robust_to_garbage(scalar_dataset):
                                  torch.tensor([col.total_compressed_size]))
         except RuntimeError:
             pass
+
+
+def test_image_parsers_robust_to_corruption():
+    """jpeg/png host parsers must raise or return on bit-flipped streams,
+    never crash the process."""
+    from PIL import Image
+    e = _ext()
+    rng = np.random.RandomState(11)
+    img = Image.fromarray(rng.randint(0, 255, (32, 48, 3)).astype(np.uint8))
+    bj = io.BytesIO(); img.save(bj, format='JPEG', quality=85)
+    bp = io.BytesIO(); img.save(bp, format='PNG')
+    for parse, blob in ((e.jpeg_parse_batch, bj.getvalue()),
+                        (e.png_parse_batch, bp.getvalue())):
+        for trial in range(150):
+            buf = bytearray(blob)
+            for _ in range(rng.randint(1, 5)):
+                buf[rng.randint(0, len(buf))] ^= 1 << rng.randint(0, 8)
+            t = torch.frombuffer(bytes(buf), dtype=torch.uint8)
+            try:
+                parse(t, torch.tensor([0]), torch.tensor([len(buf)]))
+            except RuntimeError:
+                pass
+        # truncations at every prefix of the header region
+        for cut in range(1, min(64, len(blob))):
+            t = torch.frombuffer(bytes(blob[:cut]), dtype=torch.uint8)
+            try:
+                parse(t, torch.tensor([0]), torch.tensor([cut]))
+            except RuntimeError:
+                pass
