@@ -276,15 +276,27 @@ __global__ void npy_payload_offsets_kernel(
     pay_len[i] = 0;
     return;
   }
-  int hlen;
-  int hdr;
+  int64_t hlen;
+  int64_t hdr;
   if (p[6] == 1) {           // version 1.0: u16 header length
-    hlen = (int)p[8] | ((int)p[9] << 8);
+    hlen = (int64_t)p[8] | ((int64_t)p[9] << 8);
     hdr = 10 + hlen;
   } else {                   // version 2.0+: u32 header length
-    hlen = (int)p[8] | ((int)p[9] << 8) | ((int)p[10] << 16) |
-           ((int)p[11] << 24);
+    if (val_len[i] < 12) {
+      status[0] = 7;
+      pay_off[i] = val_off[i];
+      pay_len[i] = 0;
+      return;
+    }
+    hlen = (int64_t)p[8] | ((int64_t)p[9] << 8) | ((int64_t)p[10] << 16) |
+           ((int64_t)p[11] << 24);
     hdr = 12 + hlen;
+  }
+  if (hdr > (int64_t)val_len[i]) {  // corrupt header claims > payload
+    status[0] = 7;
+    pay_off[i] = val_off[i];
+    pay_len[i] = 0;
+    return;
   }
   pay_off[i] = val_off[i] + hdr;
   pay_len[i] = (int64_t)val_len[i] - hdr;
