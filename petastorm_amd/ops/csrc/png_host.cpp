@@ -62,7 +62,11 @@ py::dict png_parse_batch(torch::Tensor buf, torch::Tensor val_off,
       uint32_t clen = be32(p + pos);
       const uint8_t* ctag = p + pos + 4;
       const uint8_t* cdata = p + pos + 8;
+      // chunk payloads we CONSUME must lie inside the buffer; a corrupt
+      // length would otherwise send the GPU inflate past the upload
       if (!memcmp(ctag, "IHDR", 4)) {
+        TORCH_CHECK(clen >= 13 && pos + 8 + 13 <= len,
+                    "image ", i, ": truncated IHDR");
         w = (int)be32(cdata);
         h = (int)be32(cdata + 4);
         depth = cdata[8];
@@ -74,15 +78,20 @@ py::dict png_parse_batch(torch::Tensor buf, torch::Tensor val_off,
         TORCH_CHECK(depth == 8 || depth == 16, "image ", i,
                     ": bit depth ", depth, " unsupported");
       } else if (!memcmp(ctag, "IDAT", 4)) {
+        TORCH_CHECK(pos + 8 + (int64_t)clen <= len,
+                    "image ", i, ": IDAT overruns buffer");
         seg_off.push_back(off[i] + pos + 8);
         seg_len.push_back((int64_t)clen);
         ++nsegs;
       } else if (!memcmp(ctag, "IEND", 4)) {
         break;
       }
-      pos += 12 + clen;  // len + tag + data + crc
+      pos += 12 + (int64_t)clen;  // len + tag + data + crc
     }
-    TORCH_CHECK(w > 0 && nsegs > 0, "image ", i, ": missing IHDR/IDAT");
+    TORCH_CHECK(w > 0 && h > 0 && nsegs > 0,
+                "image ", i, ": missing IHDR/IDAT");
+    TORCH_CHECK(w <= (1 << 24) && h <= (1 << 24),
+                "image ", i, ": implausible dimensions ", w, "x", h);
     int ch = (ctype == 2) ? 3 : (ctype == 6) ? 4 : (ctype == 4) ? 2 : 1;
     int fu = ch * (depth / 8);
     width[i] = w;
@@ -91,7 +100,7 @@ py::dict png_parse_batch(torch::Tensor buf, torch::Tensor val_off,
     bit_depth[i] = depth;
     row_bytes[i] = w * fu;
     bpp[i] = fu;
-    raw_size[i] = (int64_t)(w * fu + 1) * h;
+    raw_size[i] = ((int64_t)w * fu + 1) * h;
     seg_count[i] = nsegs;
   }
 
@@ -134,18 +143,25 @@ py::dict byte_array_host_offsets(torch::Tensor buf, torch::Tensor val_start,
   int64_t* o = off.data_ptr<int64_t>();
   int64_t* l = len.data_ptr<int64_t>();
   int64_t k = 0;
+  const int64_t bufsz = buf.numel();
   {
     // pure C++ scan: release the GIL so IO-thread parsing overlaps python
     py::gil_scoped_release nogil;
     for (int64_t j = 0; j < n_pages; ++j) {
       int64_t pos = starts[j];
       for (int64_t v = 0; v < cnt[j]; ++v) {
+        TORCH_CHECK(pos >= 0 && pos + 4 <= bufsz,
+                    "byte-array page ", j, ": length prefix overruns "
+                    "buffer at value ", v);
         uint32_t ln;
         std::memcpy(&ln, base + pos, 4);
+        TORCH_CHECK((int64_t)ln <= bufsz - pos - 4,
+                    "byte-array page ", j, ": value ", v,
+                    " overruns buffer");
         o[k] = pos + 4;
         l[k] = ln;
         ++k;
-        pos += 4 + ln;
+        pos += 4 + (int64_t)ln;
       }
     }
   }

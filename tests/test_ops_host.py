@@ -196,3 +196,23 @@ def test_image_parsers_robust_to_corruption():
                 parse(t, torch.tensor([0]), torch.tensor([cut]))
             except RuntimeError:
                 pass
+
+
+def test_byte_array_host_offsets_rejects_overrun():
+    """Corrupt varlen length prefixes must raise, not walk off the buffer."""
+    e = _ext()
+    # two values: 3 bytes, then a length claiming 2^31
+    import struct
+    payload = struct.pack('<I', 3) + b'abc' + struct.pack('<I', 0x7fffffff)
+    t = torch.frombuffer(bytearray(payload + b'xy'), dtype=torch.uint8)
+    with pytest.raises(RuntimeError, match='overruns'):
+        e.byte_array_host_offsets(t, torch.tensor([0]), torch.tensor([2]))
+    # truncated length prefix
+    t2 = torch.frombuffer(bytearray(b'\x03\x00'), dtype=torch.uint8)
+    with pytest.raises(RuntimeError, match='overruns'):
+        e.byte_array_host_offsets(t2, torch.tensor([0]), torch.tensor([1]))
+    # valid input still works
+    ok = struct.pack('<I', 3) + b'abc' + struct.pack('<I', 1) + b'z'
+    t3 = torch.frombuffer(bytearray(ok), dtype=torch.uint8)
+    res = e.byte_array_host_offsets(t3, torch.tensor([0]), torch.tensor([2]))
+    assert res['off'].tolist() == [4, 11] and res['len'].tolist() == [3, 1]
