@@ -399,3 +399,49 @@ def test_batch_transform_returning_torch_tensors(scalar_dataset):
     np.testing.assert_array_equal(ids[order], scalar_dataset['cols']['id'])
     np.testing.assert_array_almost_equal(
         f0[order], scalar_dataset['cols']['f0'] * 2)
+
+
+def test_empty_dataset_raises_cleanly(tmp_path):
+    """A materialized store with zero rows raises NoDataAvailableError at
+    reader construction (loud, like the shards>rowgroups rule)."""
+    from petastorm_amd import make_reader
+    from petastorm_amd.codecs import ScalarCodec
+    from petastorm_amd.errors import NoDataAvailableError
+    from petastorm_amd.etl.dataset_metadata import materialize_dataset
+    from petastorm_amd.unischema import Unischema, UnischemaField
+    schema = Unischema('E', [UnischemaField('id', np.int64, (),
+                                            ScalarCodec(), False)])
+    url = 'file://' + str(tmp_path / 'empty')
+    with materialize_dataset(url, schema, rowgroup_size_mb=1):
+        pass
+    with pytest.raises(NoDataAvailableError):
+        make_reader(url, reader_pool_type='dummy')
+
+
+def test_edge_values_roundtrip(tmp_path):
+    """Zero-length ndarrays, NUL/astral strings, and 100k-char strings
+    survive write->read exactly."""
+    from petastorm_amd import make_reader
+    from petastorm_amd.codecs import NdarrayCodec, ScalarCodec
+    from petastorm_amd.etl.dataset_metadata import materialize_dataset
+    from petastorm_amd.unischema import Unischema, UnischemaField
+    schema = Unischema('Edge', [
+        UnischemaField('id', np.int64, (), ScalarCodec(), False),
+        UnischemaField('arr', np.float32, (None,), NdarrayCodec(), False),
+        UnischemaField('s', np.str_, (), ScalarCodec(), False),
+    ])
+    url = 'file://' + str(tmp_path / 'edge')
+    odd = 'null\x00byte and é漢\U0001f600'
+    with materialize_dataset(url, schema, rowgroup_size_mb=1) as w:
+        w.write_row({'id': np.int64(0),
+                     'arr': np.zeros(0, dtype=np.float32), 's': odd})
+        w.write_row({'id': np.int64(1),
+                     'arr': np.arange(3, dtype=np.float32),
+                     's': 'x' * 100000})
+    with make_reader(url, reader_pool_type='dummy',
+                     shuffle_row_groups=False) as r:
+        rows = list(r)
+    assert rows[0].arr.shape == (0,)
+    assert rows[0].s == odd
+    assert len(rows[1].s) == 100000
+    np.testing.assert_array_equal(rows[1].arr, [0.0, 1.0, 2.0])
