@@ -78,7 +78,10 @@ class LocalDiskCache(CacheBase):
         try:
             with open(path, 'rb') as f:
                 return pickle.load(f)
-        except (OSError, pickle.UnpicklingError, EOFError):
+        except (OSError, pickle.UnpicklingError, EOFError, ValueError,
+                AttributeError, IndexError, ImportError):
+            # a corrupt/truncated entry (partial write, version skew) is a
+            # miss: refill and overwrite
             pass
         value = fill_cache_func()
         data = pickle.dumps(value, protocol=pickle.HIGHEST_PROTOCOL)

@@ -74,3 +74,18 @@ def test_hbm_cache_lru_eviction_and_budget():
     assert cache.hits == 1 and cache.misses == 5
     cache.cleanup()
     assert cache.size_bytes == 0
+
+
+def test_corrupt_cache_entry_is_a_miss(tmp_path):
+    """A truncated/garbage entry must refill, not raise (partial writes
+    happen on crash)."""
+    from petastorm_amd.cache import LocalDiskCache
+    c = LocalDiskCache(str(tmp_path / 'c'), 10 << 20)
+    assert c.get('k', lambda: [1, 2, 3]) == [1, 2, 3]
+    # find the entry file and stomp it
+    import glob
+    (entry,) = glob.glob(str(tmp_path / 'c' / 'shard-*' / '*.pkl'))
+    for garbage in (b'', b'\x80', b'\x80\x04garbage', b'not a pickle'):
+        with open(entry, 'wb') as f:
+            f.write(garbage)
+        assert c.get('k', lambda: ['refilled']) == ['refilled']
