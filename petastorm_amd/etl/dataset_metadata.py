@@ -191,7 +191,13 @@ def materialize_dataset(dataset_url, schema,
     fs.makedirs(path, exist_ok=True)
     writer = DatasetWriter(fs, path, schema, rowgroup_size_mb, compression,
                            rows_per_rowgroup=rows_per_rowgroup)
-    yield writer
+    try:
+        yield writer
+    except BaseException:
+        # close file handles, but do NOT stamp schema metadata: a
+        # half-written store must not look complete to get_schema
+        writer.close()
+        raise
     writer.close()
     _write_dataset_metadata(fs, path, schema)
 

@@ -445,3 +445,26 @@ def test_edge_values_roundtrip(tmp_path):
     assert rows[0].s == odd
     assert len(rows[1].s) == 100000
     np.testing.assert_array_equal(rows[1].arr, [0.0, 1.0, 2.0])
+
+
+def test_materialize_exception_closes_writer_no_sidecar(tmp_path):
+    """When the write body raises, file handles still close (rows flushed
+    so far stay readable through the per-file embedded schema) but the
+    completion sidecar is NOT stamped."""
+    import os
+    from petastorm_amd import make_reader
+    from petastorm_amd.codecs import ScalarCodec
+    from petastorm_amd.etl.dataset_metadata import (METADATA_FILENAME,
+                                                    materialize_dataset)
+    from petastorm_amd.unischema import Unischema, UnischemaField
+    schema = Unischema('P', [UnischemaField('id', np.int64, (),
+                                            ScalarCodec(), False)])
+    url = 'file://' + str(tmp_path / 'partial')
+    with pytest.raises(RuntimeError, match='boom'):
+        with materialize_dataset(url, schema, rowgroup_size_mb=1) as w:
+            w.write_row({'id': np.int64(1)})
+            raise RuntimeError('boom')
+    assert not os.path.exists(str(tmp_path / 'partial' / METADATA_FILENAME))
+    # the flushed portion is readable (file was closed properly)
+    with make_reader(url, reader_pool_type='dummy') as r:
+        assert [int(row.id) for row in r] == [1]
