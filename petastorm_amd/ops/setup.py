@@ -1,19 +1,16 @@
 """Build the petastorm_amd HIP extension in-tree:
 
-    cd petastorm_amd/ops && PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+    PYTORCH_ROCM_ARCH=gfx950 python petastorm_amd/ops/setup.py build_ext --inplace
 
 The resulting .so lands next to this file and is loaded by
 petastorm_amd.ops (package __init__).  gfx950 (MI355X) is the only target.
+Works from any cwd (sources resolve relative to this file).
 """
 import os
 
-from setuptools import setup
+_HERE = os.path.dirname(os.path.abspath(__file__))
 
-os.environ.setdefault('PYTORCH_ROCM_ARCH', 'gfx950')
-
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
-
-SRC = [
+_SRC_NAMES = [
     'csrc/ext.cpp',
     'csrc/jpeg_host.cpp',
     'csrc/thrift_pages.cpp',
@@ -27,19 +24,33 @@ SRC = [
     'csrc/transforms.hip',
     'csrc/inflate.hip',
 ]
-SRC = [s for s in SRC if os.path.exists(os.path.join(os.path.dirname(__file__) or '.', s))]
 
-setup(
-    name='petastorm_amd_hip',
-    ext_modules=[
-        CUDAExtension(
-            name='_petastorm_amd_hip',
-            sources=SRC,
-            extra_compile_args={
-                'cxx': ['-O3', '-std=c++17'],
-                'nvcc': ['-O3', '-std=c++17'],
-            },
-        )
-    ],
-    cmdclass={'build_ext': BuildExtension},
-)
+
+def main():
+    from setuptools import setup
+
+    os.environ.setdefault('PYTORCH_ROCM_ARCH', 'gfx950')
+    from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+    # torch's hipify stage resolves sources against the cwd: run the build
+    # from this directory so the command works from anywhere
+    os.chdir(_HERE)
+    src = [s for s in _SRC_NAMES if os.path.exists(os.path.join(_HERE, s))]
+    setup(
+        name='petastorm_amd_hip',
+        ext_modules=[
+            CUDAExtension(
+                name='_petastorm_amd_hip',
+                sources=src,
+                extra_compile_args={
+                    'cxx': ['-O3', '-std=c++17'],
+                    'nvcc': ['-O3', '-std=c++17'],
+                },
+            )
+        ],
+        cmdclass={'build_ext': BuildExtension},
+    )
+
+
+if __name__ == '__main__':
+    main()
