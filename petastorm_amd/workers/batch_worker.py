@@ -162,6 +162,16 @@ class BatchReaderWorker(WorkerBase):
                 self._a.fs.open(path, 'rb'))
         return self._parquet_files[path]
 
+    def _partition_column(self, piece, name, n):
+        """Materialize a hive-partition value as a length-``n`` column
+        (the reference gets this from partitions= in piece.read,
+        arrow_reader_worker.py:358)."""
+        field = self._a.schema.fields.get(name)
+        if field is not None and field.numpy_dtype not in (np.str_,):
+            v = np.dtype(field.numpy_dtype).type(piece.partitions[name])
+            return np.full(n, v)
+        return np.full(n, str(piece.partitions[name]), dtype=object)
+
     def _load_columns(self, piece, column_names):
         pf = self._parquet_file(piece.path)
         available = set(pf.schema_arrow.names)
@@ -171,34 +181,41 @@ class BatchReaderWorker(WorkerBase):
                                         self._a.decode_codecs)
         for c in column_names:
             if c not in available and c in piece.partitions:
-                field = self._a.schema.fields.get(c)
-                if field is not None and field.numpy_dtype not in (np.str_,):
-                    v = np.dtype(field.numpy_dtype).type(piece.partitions[c])
-                    out[c] = np.full(table.num_rows, v)
-                else:
-                    out[c] = np.full(table.num_rows, str(piece.partitions[c]),
-                                     dtype=object)
+                out[c] = self._partition_column(piece, c, table.num_rows)
         return out
 
     def _load_with_predicate(self, piece, predicate, needed):
         """Vectorized predicate: mask on predicate columns first, early exit,
-        then gather remaining columns (reference :286-352)."""
+        then gather remaining columns (reference :286-352).  Predicate or
+        needed fields that are hive-partition keys materialize from the
+        piece path."""
         predicate_fields = list(predicate.get_fields())
         other = [f for f in needed if f not in predicate_fields]
         pf = self._parquet_file(piece.path)
-        pred_tab = pf.read_row_group(piece.row_group, columns=predicate_fields)
+        available = set(pf.schema_arrow.names)
+        pred_file_cols = [f for f in predicate_fields if f in available]
+        pred_tab = pf.read_row_group(piece.row_group, columns=pred_file_cols)
         pred_cols = arrow_table_to_numpy_dict(pred_tab, self._a.schema,
                                               self._a.decode_codecs)
+        for f in predicate_fields:
+            if f not in available and f in piece.partitions:
+                pred_cols[f] = self._partition_column(piece, f,
+                                                      pred_tab.num_rows)
         mask = np.asarray(predicate.do_include_vectorized(pred_cols),
                           dtype=bool)
         if not mask.any():
             return {}
         idx = np.nonzero(mask)[0]
         out = {f: pred_cols[f][idx] for f in predicate_fields if f in needed}
-        if other:
-            rest = pf.read_row_group(piece.row_group, columns=other).take(idx)
+        other_file = [f for f in other if f in available]
+        if other_file:
+            rest = pf.read_row_group(piece.row_group,
+                                     columns=other_file).take(idx)
             out.update(arrow_table_to_numpy_dict(rest, self._a.schema,
                                                  self._a.decode_codecs))
+        for f in other:
+            if f not in available and f in piece.partitions:
+                out[f] = self._partition_column(piece, f, len(idx))
         return out
 
     def _shuffle_and_drop(self, columns, piece, shuffle_row_drop_partition, n):

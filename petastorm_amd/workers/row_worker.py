@@ -132,12 +132,12 @@ class RowReaderWorker(WorkerBase):
         part_pred = [f for f in predicate_fields
                      if f not in available and f in piece.partitions]
         predicate_fields = [f for f in predicate_fields if f in available]
+        part_vals = {f: _partition_value(self._a.schema.fields.get(f),
+                                         piece.partitions[f])
+                     for f in part_pred}
         if part_pred and not predicate_fields:
             # all predicate fields are partition keys: evaluate once
-            vals = {f: _partition_value(self._a.schema.fields.get(f),
-                                        piece.partitions[f])
-                    for f in part_pred}
-            if not predicate.do_include(vals):
+            if not predicate.do_include(part_vals):
                 return []
             return self._load_rows(piece, all_fields)
         pred_table = pf.read_row_group(piece.row_group,
@@ -146,27 +146,35 @@ class RowReaderWorker(WorkerBase):
                      for name in predicate_fields}
         n = pred_table.num_rows
         # predicate operates on DECODED values (reference decodes predicate
-        # columns before evaluating, :232)
+        # columns before evaluating, :232); mixed predicates see their
+        # partition-key fields too (constant within the row group)
         decoded_pred_rows = [
-            decode_row({f: pred_cols[f][i] for f in predicate_fields},
-                       self._a.schema)
+            dict(decode_row({f: pred_cols[f][i]
+                             for f in predicate_fields}, self._a.schema),
+                 **part_vals)
             for i in range(n)]
         match_idx = [i for i, r in enumerate(decoded_pred_rows)
                      if predicate.do_include(r)]
         if not match_idx:
             return []
-        if not other_fields:
-            return [dict(zip(predicate_fields,
-                             (pred_cols[f][i] for f in predicate_fields)))
-                    for i in match_idx]
-        rest = pf.read_row_group(piece.row_group, columns=other_fields)
-        rest = rest.take(match_idx)
-        rest_cols = {name: rest.column(name).to_pylist()
-                     for name in other_fields}
+        other_file = [f for f in other_fields if f in available]
+        other_part = [f for f in other_fields
+                      if f not in available and f in piece.partitions]
+        rest_cols = {}
+        if other_file:
+            rest = pf.read_row_group(piece.row_group, columns=other_file)
+            rest = rest.take(match_idx)
+            rest_cols = {name: rest.column(name).to_pylist()
+                         for name in other_file}
+        other_part_vals = {
+            f: _partition_value(self._a.schema.fields.get(f),
+                                piece.partitions[f]) for f in other_part}
         rows = []
         for out_i, i in enumerate(match_idx):
             row = {f: pred_cols[f][i] for f in predicate_fields}
-            row.update({f: rest_cols[f][out_i] for f in other_fields})
+            row.update({f: rest_cols[f][out_i] for f in other_file})
+            row.update(part_vals)
+            row.update(other_part_vals)
             rows.append(row)
         return rows
 

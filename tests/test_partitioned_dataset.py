@@ -66,3 +66,40 @@ def test_partition_predicate_in_worker_for_batch(partitioned_dataset):
                      shuffle_row_groups=False, predicate=pred) as r:
         rows = list(r)
     assert len(rows) == 50 and all(r_.color == 'blue' for r_ in rows)
+
+
+def test_mixed_file_and_partition_predicate(tmp_path):
+    """A predicate over one parquet column AND one hive-partition key works
+    on both routes (reference piece.read(partitions=...) supports this,
+    arrow_reader_worker.py:358)."""
+    import os
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+    from petastorm_amd import make_batch_reader, make_reader
+    from petastorm_amd.predicates import in_lambda
+
+    d = str(tmp_path / 'mixed')
+    for color in ('red', 'blue'):
+        sub = os.path.join(d, 'color=%s' % color)
+        os.makedirs(sub)
+        t = pa.table({'id': pa.array(np.arange(100, dtype=np.int64))})
+        pq.write_table(t, os.path.join(sub, 'f.parquet'), row_group_size=50)
+    url = 'file://' + d
+
+    pred_vec = in_lambda(['id', 'color'],
+                         lambda v: (np.asarray(v['id']) % 2 == 0) &
+                                   (np.asarray(v['color']) == 'red'))
+    with make_batch_reader(url, predicate=pred_vec,
+                           shuffle_row_groups=False) as r:
+        rows = [(int(i), str(c)) for b in r
+                for i, c in zip(b.id, b.color)]
+    assert len(rows) == 50
+    assert all(i % 2 == 0 and c == 'red' for i, c in rows)
+
+    pred_row = in_lambda(['id', 'color'],
+                         lambda v: v['id'] % 2 == 0 and v['color'] == 'red')
+    with make_reader(url, predicate=pred_row,
+                     shuffle_row_groups=False) as r:
+        got = [(int(row.id), str(row.color)) for row in r]
+    assert sorted(got) == sorted(rows)
