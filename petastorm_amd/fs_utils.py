@@ -45,6 +45,9 @@ class RetryingFilesystem(object):
             for i in range(self._attempts):
                 try:
                     return attr(*args, **kwargs)
+                except (FileNotFoundError, PermissionError,
+                        IsADirectoryError, NotADirectoryError):
+                    raise  # deterministic: retrying cannot succeed
                 except Exception as e:  # noqa: BLE001 - deliberate catch-all retry
                     last = e
                     if i + 1 < self._attempts:
