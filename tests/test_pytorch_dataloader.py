@@ -129,3 +129,28 @@ def test_loader_guards_concurrent_iteration(test_dataset):
         next(iter(loader))
     loader.stop()
     loader.join()
+
+
+def test_dataloader_ngram_with_shuffling_queue(tmp_path):
+    """NGram windows collate through DataLoader with a shuffling queue
+    (reference tests/test_pytorch_dataloader ngram coverage)."""
+    from petastorm_amd import make_reader
+    from petastorm_amd.ngram import NGram
+    from petastorm_amd.pytorch import DataLoader
+    from petastorm_amd.test_util.dataset_gen import create_sequence_dataset
+    url = 'file://' + str(tmp_path / 'seq')
+    create_sequence_dataset(url, num_rows=80)
+    ng = NGram({0: ['timestamp', 'tokens'], 1: ['timestamp', 'tokens']},
+               delta_threshold=10, timestamp_field='timestamp')
+    with make_reader(url, schema_fields=ng, shuffle_row_groups=False) as r:
+        loader = DataLoader(r, batch_size=4, shuffling_queue_capacity=16,
+                            seed=3)
+        windows = 0
+        for batch in loader:
+            assert set(batch.keys()) == {0, 1}
+            n = batch[0]['timestamp'].shape[0]
+            assert batch[1]['timestamp'].shape[0] == n
+            # consecutive timesteps obey the delta rule
+            assert (batch[1]['timestamp'] - batch[0]['timestamp'] <= 10).all()
+            windows += n
+    assert windows > 0
