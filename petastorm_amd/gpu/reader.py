@@ -119,6 +119,10 @@ class GpuBatchReader(object):
         self._fs = fs
         self._paths = path_or_paths
         self.device = torch.device(device)
+        if self.device.type == 'cuda' and not torch.cuda.is_available():
+            raise RuntimeError(
+                "make_batch_reader(device='cuda') requires a GPU: "
+                'torch.cuda.is_available() is False on this machine')
         self._decoder = GpuRowGroupDecoder(self.device)
         self._pin_pool = _PinnedPool()
 

@@ -458,3 +458,17 @@ def test_dryrun_flba(stub_decoder, tmp_path):
     got = dec.decode_string_column(out['fb'], sch.fields['fb'])
     assert got.tolist() == [b'ab12'] * n
     assert not dec.cpu_assist_columns
+
+
+def test_cuda_route_requires_gpu_at_construction(tmp_path):
+    """device='cuda' on a GPU-less machine fails at CONSTRUCTION with a
+    clear message, not at first batch."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip('machine has a GPU')
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'sc')
+    create_scalar_dataset(url, num_rows=50, rowgroup_size=25)
+    with pytest.raises(RuntimeError, match='requires a GPU'):
+        make_batch_reader(url, device='cuda')

@@ -76,8 +76,9 @@ def test_gpu_reader_accepts_rowgroup_selector(indexed_dataset):
     check: the constructor's piece filtering needs no GPU)."""
     from petastorm_amd.gpu.reader import GpuBatchReader
     fs, path = get_filesystem_and_path_or_paths(indexed_dataset['url'])
-    all_r = GpuBatchReader(fs, path, shuffle_row_groups=False)
-    sel_r = GpuBatchReader(fs, path, shuffle_row_groups=False,
+    all_r = GpuBatchReader(fs, path, shuffle_row_groups=False,
+                           device='cpu')
+    sel_r = GpuBatchReader(fs, path, shuffle_row_groups=False, device='cpu',
                            rowgroup_selector=SingleIndexSelector(
                                'id2_index', [1]))
     assert 0 < len(sel_r._pieces) < len(all_r._pieces)
