@@ -357,6 +357,13 @@ class CompressedImageCodec(DataframeColumnCodec):
             expected = np.dtype(unischema_field.numpy_dtype)
             if arr.dtype != expected:
                 arr = arr.astype(expected)
+        # single-channel fields declared (H, W, 1) encode as 2-D images;
+        # give back the declared shape
+        shape = unischema_field.shape
+        if shape and all(d is not None for d in shape) and \
+                tuple(arr.shape) != tuple(shape) and \
+                arr.size == int(np.prod(shape)):
+            arr = arr.reshape(shape)
         return arr
 
     def arrow_type(self, unischema_field):

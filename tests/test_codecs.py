@@ -139,3 +139,20 @@ def test_compressed_ndarray_npz_container_upstream_compatible():
     assert back.container == 'npz'
     np.testing.assert_array_equal(back.decode(field, back.encode(field, arr)),
                                   arr)
+
+
+def test_image_codec_single_channel_shape_roundtrip():
+    """(H, W, 1) image fields come back with the DECLARED shape even though
+    the container stores a 2-D grayscale image."""
+    from petastorm_amd.codecs import CompressedImageCodec
+    from petastorm_amd.unischema import UnischemaField
+    rng = np.random.RandomState(0)
+    for codec_name in ('png', 'jpeg'):
+        f = UnischemaField('g', np.uint8, (28, 28, 1),
+                           CompressedImageCodec(codec_name, quality=95),
+                           False)
+        v = rng.randint(0, 255, (28, 28, 1)).astype(np.uint8)
+        out = f.codec.decode(f, f.codec.encode(f, v))
+        assert out.shape == (28, 28, 1)
+        if codec_name == 'png':
+            np.testing.assert_array_equal(out, v)
