@@ -34,7 +34,11 @@ from petastorm_amd.reader_impl.serializers import PickleSerializer
 
 _STOP = '__petastorm_amd_stop__'
 _STARTED = '__petastorm_amd_worker_started__'
-_WORKER_START_TIMEOUT_S = 60
+# Spawned workers each import the full torch stack; on a contended host a
+# wide pool can need minutes.  A dead worker fails the handshake
+# immediately, so a generous cap only delays reporting genuine hangs.
+_WORKER_START_TIMEOUT_S = int(os.environ.get(
+    'PSA_WORKER_START_TIMEOUT_S', '180'))
 
 
 def _orphan_monitor(parent_pid):
@@ -104,11 +108,17 @@ class ProcessPool(object):
         deadline = time.time() + _WORKER_START_TIMEOUT_S
         pending = []
         while started < self.workers_count:
+            dead = [p for p in self._procs if p.exitcode is not None]
+            if dead:
+                raise RuntimeError(
+                    'Worker process(es) died during startup (exitcodes '
+                    '{})'.format([p.exitcode for p in dead]))
             remaining = deadline - time.time()
             if remaining <= 0:
                 raise TimeoutWaitingForResultError(
-                    'Timed out waiting for {} worker processes to start'
-                    .format(self.workers_count - started))
+                    'Timed out waiting for {} worker processes to start '
+                    '(raise PSA_WORKER_START_TIMEOUT_S for contended '
+                    'hosts)'.format(self.workers_count - started))
             try:
                 kind, msg = self._results_q.get(timeout=min(remaining, 0.5))
             except Exception:
