@@ -37,3 +37,12 @@ class SlowWorker(WorkerBase):
         import time
         time.sleep(0.2)
         self.publish_func(item)
+
+
+class DiesOnInitWorker(WorkerBase):
+    """Exits the process during construction (spawn-bootstrap failure
+    stand-in for the startup fail-fast test)."""
+
+    def __init__(self, worker_id, publish_func, args):
+        import os
+        os._exit(3)

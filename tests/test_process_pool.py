@@ -107,3 +107,18 @@ def test_serializers_roundtrip():
     assert isinstance(data, bytes)
     t2 = ats.deserialize(data)
     assert t2.equals(table)
+
+
+def test_startup_dead_worker_fails_fast():
+    """A worker that dies during construction fails start() immediately
+    with its exitcode — not after the full startup timeout."""
+    import time
+    from petastorm_amd.test_util.stub_workers import DiesOnInitWorker
+    from petastorm_amd.workers_pool.process_pool import ProcessPool
+    pool = ProcessPool(2)
+    t0 = time.monotonic()
+    with pytest.raises(RuntimeError, match='died during startup'):
+        pool.start(DiesOnInitWorker)
+    assert time.monotonic() - t0 < 60  # far below the 180 s timeout
+    pool.stop()
+    pool.join()
