@@ -91,6 +91,16 @@ class RowReaderWorker(WorkerBase):
             self.publish_func(rows)
 
     # ------------------------------------------------------------------
+    def shutdown(self):
+        # close cached footer/file handles (held per worker for the
+        # pool's lifetime); without this they linger until GC
+        for pf in self._parquet_files.values():
+            try:
+                pf.close()
+            except Exception:  # noqa: BLE001 - best-effort teardown
+                pass
+        self._parquet_files.clear()
+
     def _parquet_file(self, path):
         if path not in self._parquet_files:
             import pyarrow.parquet as pq
