@@ -79,6 +79,43 @@ def _make_pool(reader_pool_type, workers_count, results_queue_size,
     raise ValueError('Unknown reader_pool_type: {!r}'.format(reader_pool_type))
 
 
+def _upstream_compat(seed, shard_seed, ignored):
+    """Accept upstream-petastorm kwargs that have no effect here: map
+    ``shard_seed`` onto the unified ``seed`` (reference deprecates it the
+    same way, reader.py:475-479) and warn about transport/driver knobs this
+    framework replaced (zmq/libhdfs3/pyarrow-serialization)."""
+    if shard_seed is not None:
+        warnings.warn('shard_seed is deprecated; it now sets the unified '
+                      'seed (row-group shuffle + sharding + row shuffle)',
+                      DeprecationWarning)
+        if seed is None:
+            seed = shard_seed
+    for name, value in ignored.items():
+        if value is not None:
+            warnings.warn('{} is accepted for upstream-petastorm '
+                          'compatibility but has no effect here'
+                          .format(name), DeprecationWarning)
+    return seed
+
+
+def _resolve_fs(dataset_url_or_urls, storage_options, filesystem):
+    """URL(s) -> (fs, path-or-paths); an explicit ``filesystem`` (upstream
+    reader.py:60-77 ``filesystem=``) bypasses scheme resolution."""
+    if filesystem is None:
+        return get_filesystem_and_path_or_paths(dataset_url_or_urls,
+                                                storage_options)
+    from urllib.parse import urlparse
+
+    def path_of(u):
+        p = urlparse(u)
+        return p.path if p.scheme in ('file', '') \
+            else (p.netloc + p.path)
+
+    if isinstance(dataset_url_or_urls, list):
+        return filesystem, [path_of(u) for u in dataset_url_or_urls]
+    return filesystem, path_of(dataset_url_or_urls)
+
+
 def make_reader(dataset_url,
                 schema_fields=None,
                 reader_pool_type='thread', workers_count=10,
@@ -92,14 +129,24 @@ def make_reader(dataset_url,
                 cache_type='null', cache_location=None, cache_size_limit=None,
                 cache_row_size_estimate=None, cache_extra_settings=None,
                 transform_spec=None, filters=None,
-                storage_options=None):
+                storage_options=None,
+                shard_seed=None, filesystem=None,
+                pyarrow_serialize=None, hdfs_driver=None,
+                zmq_copy_buffers=None):
     """Row-oriented reader over a petastorm_amd dataset (reference :60-206).
 
     Each ``next(reader)`` yields one row as a schema-named namedtuple (or a
     ``{timestep: namedtuple}`` dict when ``schema_fields`` is an NGram).
+    ``shard_seed``/``pyarrow_serialize``/``hdfs_driver``/``zmq_copy_buffers``
+    are accepted for upstream call-site compatibility (deprecation warning);
+    ``filesystem`` passes an explicit fsspec/pyarrow filesystem.
     """
+    seed = _upstream_compat(seed, shard_seed,
+                            dict(pyarrow_serialize=pyarrow_serialize,
+                                 hdfs_driver=hdfs_driver,
+                                 zmq_copy_buffers=zmq_copy_buffers))
     dataset_url = normalize_dataset_url_or_urls(dataset_url)
-    fs, path = get_filesystem_and_path_or_paths(dataset_url, storage_options)
+    fs, path = _resolve_fs(dataset_url, storage_options, filesystem)
     try:
         dsm.get_schema(fs, path)
     except ValueError:
@@ -141,7 +188,10 @@ def make_batch_reader(dataset_url_or_urls,
                       transform_spec=None, filters=None,
                       decode_codecs=True,
                       storage_options=None,
-                      device=None, gpu_options=None):
+                      device=None, gpu_options=None,
+                      shard_seed=None, filesystem=None,
+                      hdfs_driver=None, zmq_copy_buffers=None,
+                      convert_early_to_numpy=None):
     """Columnar batch reader (reference :209-352): each ``next(reader)``
     yields one row-group-sized batch as a namedtuple of column arrays.
 
@@ -152,9 +202,17 @@ def make_batch_reader(dataset_url_or_urls,
     :param device: ``None``/'cpu' for the worker-pool path, 'cuda' for the
         MI355X HIP pipeline (on-GPU page decode; batches are torch tensors).
     """
+    seed = _upstream_compat(seed, shard_seed,
+                            dict(hdfs_driver=hdfs_driver,
+                                 zmq_copy_buffers=zmq_copy_buffers))
+    if convert_early_to_numpy is False:
+        warnings.warn('convert_early_to_numpy=False (arrow-table payloads) '
+                      'is not supported: this framework always converts in '
+                      'the worker (the upstream option True path)',
+                      DeprecationWarning)
     dataset_url_or_urls = normalize_dataset_url_or_urls(dataset_url_or_urls)
-    fs, path_or_paths = get_filesystem_and_path_or_paths(
-        dataset_url_or_urls, storage_options)
+    fs, path_or_paths = _resolve_fs(dataset_url_or_urls, storage_options,
+                                    filesystem)
     if device is not None and str(device).startswith('cuda'):
         from petastorm_amd.gpu.reader import GpuBatchReader
         # loud rejection of options the GPU pipeline does not implement —

@@ -203,3 +203,26 @@ def test_weighted_sampling_over_batch_readers(scalar_dataset, tmp_path):
     batches = [next(mixed) for _ in range(20)]
     assert all(hasattr(b, 'id') for b in batches)
     r1.stop(); r1.join(); r2.stop(); r2.join()
+
+
+def test_upstream_compat_kwargs(scalar_dataset):
+    """Upstream call sites pass shard_seed/hdfs_driver/zmq_copy_buffers/
+    pyarrow_serialize/filesystem — accepted (deprecation warnings for
+    no-effect knobs; filesystem is honored)."""
+    import warnings
+    import fsspec
+    from petastorm_amd import make_batch_reader
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter('always')
+        with make_batch_reader(scalar_dataset['url'], shard_seed=7,
+                               hdfs_driver='libhdfs3',
+                               zmq_copy_buffers=True,
+                               shuffle_row_groups=True) as r:
+            ids = _collect(list(r))['id']
+        dep = [x for x in w if issubclass(x.category, DeprecationWarning)]
+    assert len(ids) == 500
+    assert len(dep) == 3  # shard_seed + hdfs_driver + zmq_copy_buffers
+    fs = fsspec.filesystem('file')
+    with make_batch_reader(scalar_dataset['url'], filesystem=fs,
+                           shuffle_row_groups=False) as r:
+        assert len(_collect(list(r))['id']) == 500
