@@ -156,3 +156,31 @@ def test_image_codec_single_channel_shape_roundtrip():
         assert out.shape == (28, 28, 1)
         if codec_name == 'png':
             np.testing.assert_array_equal(out, v)
+
+
+def test_float16_scalar_roundtrip_both_routes(tmp_path):
+    """float16 scalars (stored as Parquet FLBA/half) survive write->read
+    exactly on the row and batch routes."""
+    from petastorm_amd import make_batch_reader, make_reader
+    from petastorm_amd.codecs import ScalarCodec
+    from petastorm_amd.etl.dataset_metadata import materialize_dataset
+    from petastorm_amd.unischema import Unischema, UnischemaField
+    S = Unischema('F16', [
+        UnischemaField('id', np.int64, (), ScalarCodec(), False),
+        UnischemaField('h', np.float16, (), ScalarCodec(), False),
+    ])
+    url = 'file://' + str(tmp_path / 'f16')
+    rng = np.random.RandomState(0)
+    vals = rng.rand(50).astype(np.float16)
+    with materialize_dataset(url, S, rowgroup_size_mb=1) as w:
+        for i in range(50):
+            w.write_row({'id': np.int64(i), 'h': vals[i]})
+    with make_reader(url, reader_pool_type='dummy',
+                     shuffle_row_groups=False) as r:
+        got = {int(row.id): row.h for row in r}
+    assert all(got[i] == vals[i] for i in range(50))
+    with make_batch_reader(url, shuffle_row_groups=False) as r:
+        b = next(iter(r))
+    assert b.h.dtype == np.float16
+    np.testing.assert_array_equal(np.asarray(b.h),
+                                  vals[np.asarray(b.id)])
