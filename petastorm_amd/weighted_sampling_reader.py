@@ -43,7 +43,10 @@ class WeightedSamplingReader(object):
         return self
 
     def __next__(self):
-        idx = int(np.searchsorted(self._cum, self._rng.uniform()))
+        # clamp: float rounding can leave cum[-1] a hair under 1.0, and a
+        # draw above it would index past the last reader
+        idx = min(int(np.searchsorted(self._cum, self._rng.uniform())),
+                  len(self._readers) - 1)
         return next(self._readers[idx])
 
     next = __next__
