@@ -8,7 +8,7 @@ Parity: /root/reference/petastorm/selectors.py:20-100.
   ``UnionIndexSelector`` (:78-100)
 
 Indexes themselves are built by petastorm_amd.etl.rowgroup_indexing and
-stored as a JSON+npy sidecar (not pickled parquet metadata).
+stored as a JSON sidecar (not pickled parquet metadata).
 """
 
 
