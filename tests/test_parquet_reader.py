@@ -226,3 +226,25 @@ def test_upstream_compat_kwargs(scalar_dataset):
     with make_batch_reader(scalar_dataset['url'], filesystem=fs,
                            shuffle_row_groups=False) as r:
         assert len(_collect(list(r))['id']) == 500
+
+
+def test_weighted_sampling_seeded_determinism(scalar_dataset):
+    """Same seed -> same source-reader sequence (reference uses np.random
+    without seeding; seeding is an extension worth pinning)."""
+    from petastorm_amd import make_batch_reader
+    from petastorm_amd.weighted_sampling_reader import WeightedSamplingReader
+
+    def run(seed):
+        r1 = make_batch_reader(scalar_dataset['url'], num_epochs=None,
+                               schema_fields=['id'],
+                               shuffle_row_groups=False)
+        r2 = make_batch_reader(scalar_dataset['url'], num_epochs=None,
+                               schema_fields=['f0'],
+                               shuffle_row_groups=False)
+        mixed = WeightedSamplingReader([r1, r2], [0.7, 0.3], seed=seed)
+        kinds = ['id' if hasattr(b, 'id') else 'f0'
+                 for b in (next(mixed) for _ in range(12))]
+        r1.stop(); r1.join(); r2.stop(); r2.join()
+        return kinds
+
+    assert run(5) == run(5)
