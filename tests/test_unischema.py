@@ -123,3 +123,22 @@ def test_many_fields_namedtuple():
     s = Unischema('big', fields)
     nt = s.make_namedtuple(**{'f{:04d}'.format(i): i for i in range(300)})
     assert nt.f0299 == 299
+
+
+def test_namedtuple_cache_shared_across_instances():
+    """Two Unischema instances with the same name+fields share one
+    namedtuple class, so rows from different readers compare equal
+    (reference unischema.py:88-111 _NamedtupleCache)."""
+    from petastorm_amd.codecs import ScalarCodec
+    from petastorm_amd.unischema import Unischema, UnischemaField
+
+    def build():
+        return Unischema('CacheT', [
+            UnischemaField('a', np.int32, (), ScalarCodec(), False),
+            UnischemaField('b', np.float32, (), ScalarCodec(), False)])
+
+    s1, s2 = build(), build()
+    r1 = s1.make_namedtuple(a=1, b=2.0)
+    r2 = s2.make_namedtuple(a=1, b=2.0)
+    assert type(r1) is type(r2)
+    assert r1 == r2
