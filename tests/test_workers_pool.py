@@ -142,3 +142,18 @@ def test_ventilator_reset():
 def test_ventilator_invalid_iterations():
     with pytest.raises(ValueError):
         ConcurrentVentilator(lambda v: None, [], iterations=0)
+
+
+def test_reader_diagnostics_contract(tmp_path):
+    """reader.diagnostics exposes the pool counters the benchmark harness
+    reads (reference thread_pool.py:258-263, process_pool.py:303-312)."""
+    from petastorm_amd import make_reader
+    from petastorm_amd.test_util.dataset_gen import create_scalar_dataset
+    url = 'file://' + str(tmp_path / 'diag')
+    create_scalar_dataset(url, num_rows=100, rowgroup_size=50)
+    for pool in ('dummy', 'thread'):
+        with make_reader(url, reader_pool_type=pool, num_epochs=1) as r:
+            list(r)
+            d = r.diagnostics
+            assert d['items_ventilated'] == 2
+            assert d['items_processed'] == 2
