@@ -264,3 +264,18 @@ def test_weighted_sampling_seeded_determinism():
     assert run(5) == run(5)
     assert run(5) != run(6)
     assert set(run(5)) == {'a', 'b'}
+
+
+def test_seeded_thread_pool_order_matches_dummy(scalar_dataset):
+    """With a seed, the thread pool's round-robin readout reproduces the
+    dummy pool's (inline) batch order exactly — the determinism guarantee
+    state_dict's fast-forward relies on (reference thread_pool.py:181-199
+    round-robin rationale)."""
+    def ids(pool):
+        from petastorm_amd import make_batch_reader
+        with make_batch_reader(scalar_dataset['url'], reader_pool_type=pool,
+                               workers_count=3, shuffle_row_groups=True,
+                               seed=31, num_epochs=2) as r:
+            return [int(b.id[0]) for b in r]  # first id of each batch
+
+    assert ids('thread') == ids('dummy')
